@@ -1,0 +1,690 @@
+// dllama_amd — hand-written CDNA4 (gfx950) kernels for Q40xQ80 LLM inference.
+//
+// Design notes (MI355X-first, not a port):
+//  - decode is HBM-bandwidth bound on the Q40 weight stream; the GEMV keeps
+//    one wave per output row streaming 16B nibble payloads per lane and does
+//    the int8 math with v_dot4 (sdot4), correcting the Q40 "-8" offset with
+//    precomputed per-block activation sums (so nibbles never get unpacked
+//    to signed values).
+//  - every runtime scalar the decode loop needs (position) is read from
+//    device memory so whole-token hipGraph capture works.
+//  - role parity with the reference op set: src/nn/nn-cpu-ops.cpp and the
+//    Vulkan shaders in src/nn/vulkan/ (see SURVEY.md §2.3/§2.5).
+//
+// All f32 accumulation (parity with the reference numerics).
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_fp16.h>
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#define WAVE 64
+#define QB 32  // quant block size
+
+static inline int ceil_div(int64_t a, int64_t b) { return (int)((a + b - 1) / b); }
+
+#if defined(__HIPCC__)
+#if __has_builtin(__builtin_amdgcn_sdot4)
+__device__ __forceinline__ int dot4(int a, int b, int c) {
+    return __builtin_amdgcn_sdot4(a, b, c, false);
+}
+#else
+__device__ __forceinline__ int dot4(int a, int b, int c) {
+    const char4 va = *reinterpret_cast<const char4 *>(&a);
+    const char4 vb = *reinterpret_cast<const char4 *>(&b);
+    return c + va.x * vb.x + va.y * vb.y + va.z * vb.z + va.w * vb.w;
+}
+#endif
+
+__device__ __forceinline__ float wave_reduce_sum(float v) {
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        v += __shfl_xor(v, off, WAVE);
+    return v;
+}
+
+__device__ __forceinline__ float group32_reduce_max(float v) {
+    #pragma unroll
+    for (int off = 16; off > 0; off >>= 1)
+        v = fmaxf(v, __shfl_xor(v, off, 32));
+    return v;
+}
+
+__device__ __forceinline__ float group32_reduce_sum(float v) {
+    #pragma unroll
+    for (int off = 16; off > 0; off >>= 1)
+        v += __shfl_xor(v, off, 32);
+    return v;
+}
+
+// ------------------------------------------------------------------ q80 quantize
+// f32 [rows, n] -> int8 q [rows, n], f32 scale [rows, n/32], f32 bsum [rows, n/32]
+// (role of reference quantizeF32toQ80, nn-quants.cpp:67 and the
+//  cast-forward-f32-q80 Vulkan shader). bsum = sum of the int8 values,
+// used by the GEMV to fold the Q40 nibble offset.
+__global__ void k_q80_quantize(const float *__restrict__ x,
+                               int8_t *__restrict__ q,
+                               float *__restrict__ s,
+                               float *__restrict__ bs,
+                               int n_blocks_total) {
+    // one 32-lane group per block
+    int gid = (blockIdx.x * blockDim.x + threadIdx.x) / 32;
+    int lane = threadIdx.x & 31;
+    if (gid >= n_blocks_total) return;
+    float v = x[gid * QB + lane];
+    float amax = group32_reduce_max(fabsf(v));
+    float d = amax / 127.0f;
+    float inv = d > 0.0f ? 1.0f / d : 0.0f;
+    float qf = rintf(v * inv);
+    int8_t qi = (int8_t)qf;
+    q[gid * QB + lane] = qi;
+    float bsum = group32_reduce_sum(qf);
+    if (lane == 0) {
+        s[gid] = d;
+        bs[gid] = bsum;
+    }
+}
+
+// ------------------------------------------------------------------ rmsnorm
+// y = x * w / sqrt(mean(x^2)+eps) (reference invRms_F32 + rmsNorm_F32,
+// nn-cpu-ops.cpp:114-175, fused). One workgroup per row.
+__global__ void k_rmsnorm(const float *__restrict__ x,
+                          const float *__restrict__ w,
+                          float *__restrict__ y,
+                          int n, float eps) {
+    const float *row = x + (int64_t)blockIdx.x * n;
+    float *out = y + (int64_t)blockIdx.x * n;
+    float acc = 0.0f;
+    for (int i = threadIdx.x; i < n; i += blockDim.x) {
+        float v = row[i];
+        acc += v * v;
+    }
+    __shared__ float red[16];
+    acc = wave_reduce_sum(acc);
+    int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    if (lane == 0) red[wid] = acc;
+    __syncthreads();
+    int nw = blockDim.x / WAVE;
+    float total = 0.0f;
+    for (int i = 0; i < nw; i++) total += red[i];
+    float inv = rsqrtf(total / n + eps);
+    for (int i = threadIdx.x; i < n; i += blockDim.x)
+        out[i] = row[i] * inv * w[i];
+}
+
+// fused rmsnorm + q80 quantize: avoids a full extra activation pass
+// (reference runs inv_rms -> rms_norm -> cast as three ops).
+__global__ void k_rmsnorm_q80(const float *__restrict__ x,
+                              const float *__restrict__ w,
+                              int8_t *__restrict__ q,
+                              float *__restrict__ s,
+                              float *__restrict__ bs,
+                              int n, float eps) {
+    const float *row = x + (int64_t)blockIdx.x * n;
+    float acc = 0.0f;
+    for (int i = threadIdx.x; i < n; i += blockDim.x) {
+        float v = row[i];
+        acc += v * v;
+    }
+    __shared__ float red[16];
+    acc = wave_reduce_sum(acc);
+    int wid = threadIdx.x / WAVE, lane64 = threadIdx.x % WAVE;
+    if (lane64 == 0) red[wid] = acc;
+    __syncthreads();
+    int nw = blockDim.x / WAVE;
+    float total = 0.0f;
+    for (int i = 0; i < nw; i++) total += red[i];
+    float inv = rsqrtf(total / n + eps);
+
+    int nb = n / QB;
+    int8_t *qrow = q + (int64_t)blockIdx.x * n;
+    float *srow = s + (int64_t)blockIdx.x * nb;
+    float *bsrow = bs + (int64_t)blockIdx.x * nb;
+    int lane32 = threadIdx.x & 31;
+    for (int blk = threadIdx.x / 32; blk < nb; blk += blockDim.x / 32) {
+        float v = row[blk * QB + lane32] * inv * w[blk * QB + lane32];
+        float amax = group32_reduce_max(fabsf(v));
+        float d = amax / 127.0f;
+        float qinv = d > 0.0f ? 1.0f / d : 0.0f;
+        float qf = rintf(v * qinv);
+        qrow[blk * QB + lane32] = (int8_t)qf;
+        float bsum = group32_reduce_sum(qf);
+        if (lane32 == 0) { srow[blk] = d; bsrow[blk] = bsum; }
+    }
+}
+
+// per-head rmsnorm (Qwen3 q/k-norm; reference multi-column OP_INV_RMS +
+// OP_RMS_NORM, llm.cpp:178-187). x [rows, hd], w [hd]; hd <= 512.
+__global__ void k_rmsnorm_rows(const float *__restrict__ x,
+                               const float *__restrict__ w,
+                               float *__restrict__ y,
+                               int hd, float eps) {
+    const float *row = x + (int64_t)blockIdx.x * hd;
+    float *out = y + (int64_t)blockIdx.x * hd;
+    float acc = 0.0f;
+    for (int i = threadIdx.x; i < hd; i += blockDim.x) {
+        float v = row[i];
+        acc += v * v;
+    }
+    acc = wave_reduce_sum(acc);  // blockDim == 64 (one wave)
+    float inv = rsqrtf(acc / hd + eps);
+    for (int i = threadIdx.x; i < hd; i += blockDim.x)
+        out[i] = row[i] * inv * w[i];
+}
+
+// ------------------------------------------------------------------ Q40 GEMV
+// y[b, row] = sum_j w[row, j] * x[b, j]  with W in Q40 planes and x in Q80.
+// One wave per output row; lane l streams block l, l+64, ... of the row.
+// Weight layout: qs uint8 [d, n/2] (16B per block, byte j = elem j | elem
+// j+16 << 4), scales f16 [d, n/32]. Per block:
+//   true = sw*sx*(sum_i q_i*x_i - 8*sum_i x_i)  with q in 0..15
+// so the nibble unpack is 2 VALU ops per 8 elems and sdot4 does the MAC
+// (role of reference matmul_Q80_Q40_F32, nn-cpu-ops.cpp:231-449, and the
+//  matmul-forward-q80-q40-f32.comp Vulkan shader).
+template <int NB>
+__global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
+                           const __half *__restrict__ scales,
+                           const int8_t *__restrict__ xq,
+                           const float *__restrict__ xs,
+                           const float *__restrict__ xbs,
+                           float *__restrict__ y,
+                           int d, int n) {
+    const int row = blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
+    if (row >= d) return;
+    const int lane = threadIdx.x % WAVE;
+    const int nb = n / QB;
+    const uint4 *wrow = reinterpret_cast<const uint4 *>(qs + (int64_t)row * (n >> 1));
+    const __half *srow = scales + (int64_t)row * nb;
+
+    float acc[NB];
+    #pragma unroll
+    for (int b = 0; b < NB; b++) acc[b] = 0.0f;
+
+    for (int j = lane; j < nb; j += WAVE) {
+        const uint4 wq = wrow[j];
+        const float sw = __half2float(srow[j]);
+        const uint32_t wv[4] = {wq.x, wq.y, wq.z, wq.w};
+        #pragma unroll
+        for (int b = 0; b < NB; b++) {
+            const int4 x0 = reinterpret_cast<const int4 *>(xq + (int64_t)b * n + j * QB)[0];
+            const int4 x1 = reinterpret_cast<const int4 *>(xq + (int64_t)b * n + j * QB)[1];
+            const int32_t xv[8] = {x0.x, x0.y, x0.z, x0.w, x1.x, x1.y, x1.z, x1.w};
+            int idot = 0;
+            #pragma unroll
+            for (int wi = 0; wi < 4; wi++) {
+                idot = dot4((int)(wv[wi] & 0x0F0F0F0Fu), xv[wi], idot);          // elems 4wi..4wi+3
+                idot = dot4((int)((wv[wi] >> 4) & 0x0F0F0F0Fu), xv[4 + wi], idot); // elems 16+4wi..
+            }
+            const float sx = xs[b * nb + j];
+            const float bsum = xbs[b * nb + j];
+            acc[b] = fmaf(sw * sx, (float)idot - 8.0f * bsum, acc[b]);
+        }
+    }
+    #pragma unroll
+    for (int b = 0; b < NB; b++) {
+        float r = wave_reduce_sum(acc[b]);
+        if (lane == 0) y[(int64_t)b * d + row] = r;
+    }
+}
+
+// Grouped (MoE) variant: weights [n_experts, d, ...]; slot s uses expert
+// expert_idx[s] and input row slot_batch[s]; y [S, d]
+// (reference 3-D expert matmul with index indirection, nn-core.hpp:209-213).
+__global__ void k_q40_gemv_grouped(const uint8_t *__restrict__ qs,
+                                   const __half *__restrict__ scales,
+                                   const int8_t *__restrict__ xq,
+                                   const float *__restrict__ xs,
+                                   const float *__restrict__ xbs,
+                                   const int *__restrict__ expert_idx,
+                                   float *__restrict__ y,
+                                   int d, int n, int k_slots) {
+    const int row = blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
+    const int slot = blockIdx.y;
+    if (row >= d) return;
+    const int lane = threadIdx.x % WAVE;
+    const int nb = n / QB;
+    const int e = expert_idx[slot];
+    const int b = slot / k_slots;
+    const uint4 *wrow = reinterpret_cast<const uint4 *>(
+        qs + ((int64_t)e * d + row) * (n >> 1));
+    const __half *srow = scales + ((int64_t)e * d + row) * nb;
+    float acc = 0.0f;
+    for (int j = lane; j < nb; j += WAVE) {
+        const uint4 wq = wrow[j];
+        const float sw = __half2float(srow[j]);
+        const uint32_t wv[4] = {wq.x, wq.y, wq.z, wq.w};
+        const int4 x0 = reinterpret_cast<const int4 *>(xq + (int64_t)b * n + j * QB)[0];
+        const int4 x1 = reinterpret_cast<const int4 *>(xq + (int64_t)b * n + j * QB)[1];
+        const int32_t xv[8] = {x0.x, x0.y, x0.z, x0.w, x1.x, x1.y, x1.z, x1.w};
+        int idot = 0;
+        #pragma unroll
+        for (int wi = 0; wi < 4; wi++) {
+            idot = dot4((int)(wv[wi] & 0x0F0F0F0Fu), xv[wi], idot);
+            idot = dot4((int)((wv[wi] >> 4) & 0x0F0F0F0Fu), xv[4 + wi], idot);
+        }
+        acc = fmaf(sw * xs[b * nb + j], (float)idot - 8.0f * xbs[b * nb + j], acc);
+    }
+    float r = wave_reduce_sum(acc);
+    if (lane == 0) y[(int64_t)slot * d + row] = r;
+}
+
+// ------------------------------------------------------------------ rope
+// style 0 = llama interleaved pairs (reference ropeLlama_F32,
+// nn-cpu-ops.cpp:843-863), style 1 = falcon/neox half-rotated
+// (ropeFalcon_F32, :865-885). cache [seq, hd/2, 2] = (cos, sin).
+// x [B, dim0]; row b uses position pos[0]+b.
+template <int STYLE>
+__global__ void k_rope(float *__restrict__ x,
+                       const float *__restrict__ cache,
+                       const int *__restrict__ pos,
+                       int dim0, int hd) {
+    const int b = blockIdx.y;
+    const int p = pos[0] + b;
+    const int half = hd >> 1;
+    float *row = x + (int64_t)b * dim0;
+    const float *pc = cache + (int64_t)p * hd;  // hd floats = hd/2 pairs
+    for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < dim0 / 2;
+         i += gridDim.x * blockDim.x) {
+        const int head = i / half;
+        const int j = i % half;  // cache pair index
+        int i0, i1;
+        if (STYLE == 0) {          // pair (2j, 2j+1) within each head
+            i0 = head * hd + 2 * j;
+            i1 = i0 + 1;
+        } else {                   // pair (j, j+hd/2) within each head
+            i0 = head * hd + j;
+            i1 = i0 + half;
+        }
+        const float cr = pc[2 * j];
+        const float ci = pc[2 * j + 1];
+        const float v0 = row[i0];
+        const float v1 = row[i1];
+        row[i0] = v0 * cr - v1 * ci;
+        row[i1] = v0 * ci + v1 * cr;
+    }
+}
+
+// ------------------------------------------------------------------ kv append
+// copy k,v batch rows into the caches at row pos[0]+b (reference OP_SHIFT,
+// nn-cpu-ops.cpp:1419-1441).
+__global__ void k_kv_append(const float *__restrict__ k,
+                            const float *__restrict__ v,
+                            float *__restrict__ kc,
+                            float *__restrict__ vc,
+                            const int *__restrict__ pos,
+                            int kv_dim0) {
+    const int b = blockIdx.y;
+    const int p = pos[0] + b;
+    for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < kv_dim0;
+         i += gridDim.x * blockDim.x) {
+        kc[(int64_t)p * kv_dim0 + i] = k[(int64_t)b * kv_dim0 + i];
+        vc[(int64_t)p * kv_dim0 + i] = v[(int64_t)b * kv_dim0 + i];
+    }
+}
+
+// ------------------------------------------------------------------ attention
+// Flash-style decode/prefill attention over the f32 KV cache.
+// One workgroup (4 waves) per (head, batch); waves split the t range with
+// online softmax, combined through LDS (replaces the reference's
+// materialized att buffer, multiheadAtt_F32 nn-cpu-ops.cpp:753-788).
+// VEC = head_dim / 64 elements per lane.
+template <int VEC>
+__global__ void k_attn(const float *__restrict__ q,
+                       const float *__restrict__ kc,
+                       const float *__restrict__ vc,
+                       float *__restrict__ y,
+                       const int *__restrict__ pos,
+                       int n_heads0, int kv_mul, int kv_dim0, float scale) {
+    const int h0 = blockIdx.x;
+    const int b = blockIdx.y;
+    const int hd = VEC * WAVE;
+    const int plen = pos[0] + b + 1;  // row b attends to cache rows 0..pos+b
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int kv_off = (h0 / kv_mul) * hd;
+
+    float qreg[VEC];
+    #pragma unroll
+    for (int v = 0; v < VEC; v++)
+        qreg[v] = q[((int64_t)b * n_heads0 + h0) * hd + lane * VEC + v] * scale;
+
+    float m = -1e30f, l = 0.0f, o[VEC];
+    #pragma unroll
+    for (int v = 0; v < VEC; v++) o[v] = 0.0f;
+
+    for (int t = wave; t < plen; t += 4) {
+        const float *krow = kc + (int64_t)t * kv_dim0 + kv_off + lane * VEC;
+        float partial = 0.0f;
+        #pragma unroll
+        for (int v = 0; v < VEC; v++) partial = fmaf(qreg[v], krow[v], partial);
+        const float s = wave_reduce_sum(partial);
+        const float mn = fmaxf(m, s);
+        const float f = __expf(m - mn);
+        const float w = __expf(s - mn);
+        const float *vrow = vc + (int64_t)t * kv_dim0 + kv_off + lane * VEC;
+        l = l * f + w;
+        #pragma unroll
+        for (int v = 0; v < VEC; v++) o[v] = fmaf(o[v], f, w * vrow[v]);
+        m = mn;
+    }
+
+    __shared__ float sm[4], sl[4];
+    __shared__ float so[4][VEC * WAVE];
+    if (lane == 0) { sm[wave] = m; sl[wave] = l; }
+    __syncthreads();
+    const float M = fmaxf(fmaxf(sm[0], sm[1]), fmaxf(sm[2], sm[3]));
+    const float fw = __expf(m - M);
+    #pragma unroll
+    for (int v = 0; v < VEC; v++) so[wave][lane * VEC + v] = o[v] * fw;
+    __syncthreads();
+    if (wave == 0) {
+        const float L = sl[0] * __expf(sm[0] - M) + sl[1] * __expf(sm[1] - M)
+                      + sl[2] * __expf(sm[2] - M) + sl[3] * __expf(sm[3] - M);
+        const float invL = 1.0f / L;
+        #pragma unroll
+        for (int v = 0; v < VEC; v++) {
+            const int i = lane * VEC + v;
+            y[((int64_t)b * n_heads0 + h0) * hd + i] =
+                (so[0][i] + so[1][i] + so[2][i] + so[3][i]) * invL;
+        }
+    }
+}
+
+// ------------------------------------------------------------------ swiglu (+q80)
+// d = silu(a) * g, quantized straight to Q80 (reference OP_SILU + OP_MUL +
+// OP_CAST fused; silu nn-cpu-ops.cpp:462-500).
+__global__ void k_swiglu_q80(const float *__restrict__ a,
+                             const float *__restrict__ g,
+                             int8_t *__restrict__ q,
+                             float *__restrict__ s,
+                             float *__restrict__ bs,
+                             int n_blocks_total) {
+    int gid = (blockIdx.x * blockDim.x + threadIdx.x) / 32;
+    int lane = threadIdx.x & 31;
+    if (gid >= n_blocks_total) return;
+    float av = a[gid * QB + lane];
+    float gv = g[gid * QB + lane];
+    float v = av / (1.0f + __expf(-av)) * gv;
+    float amax = group32_reduce_max(fabsf(v));
+    float d = amax / 127.0f;
+    float inv = d > 0.0f ? 1.0f / d : 0.0f;
+    float qf = rintf(v * inv);
+    q[gid * QB + lane] = (int8_t)qf;
+    float bsum = group32_reduce_sum(qf);
+    if (lane == 0) { s[gid] = d; bs[gid] = bsum; }
+}
+
+__global__ void k_silu_mul(const float *__restrict__ a,
+                           const float *__restrict__ g,
+                           float *__restrict__ out,
+                           int64_t n) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        float av = a[i];
+        out[i] = av / (1.0f + __expf(-av)) * g[i];
+    }
+}
+
+// ------------------------------------------------------------------ q80 sync pack / merge-add
+// wire layout per row: n int8 payload then n/32 f16 scales
+// (role of cast-forward-f32-q80.comp; consumed after RCCL all-gather by
+//  k_merge_add, the reference merge-add-forward-q80-f32.comp equivalent).
+__global__ void k_sync_pack(const int8_t *__restrict__ q,
+                            const float *__restrict__ s,
+                            uint8_t *__restrict__ buf,
+                            int n, int rows) {
+    const int nb = n / QB;
+    const int row_bytes = n + 2 * nb;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         i < (int64_t)rows * n; i += (int64_t)gridDim.x * blockDim.x) {
+        const int r = i / n, c = i % n;
+        buf[(int64_t)r * row_bytes + c] = (uint8_t)q[i];
+        if (c < nb) {
+            const __half h = __float2half(s[(int64_t)r * nb + c]);
+            const uint16_t u = *reinterpret_cast<const uint16_t *>(&h);
+            buf[(int64_t)r * row_bytes + n + 2 * c] = (uint8_t)(u & 0xFF);
+            buf[(int64_t)r * row_bytes + n + 2 * c + 1] = (uint8_t)(u >> 8);
+        }
+    }
+}
+
+// x[r, i] += sum_w dequant(bufs[w, r, i]) — the all-reduce completion
+// (reference OP_MERGE_ADD, nn-cpu-ops.cpp:920-957).
+__global__ void k_merge_add(float *__restrict__ x,
+                            const uint8_t *__restrict__ bufs,
+                            int world, int n, int rows) {
+    const int nb = n / QB;
+    const int row_bytes = n + 2 * nb;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         i < (int64_t)rows * n; i += (int64_t)gridDim.x * blockDim.x) {
+        const int r = i / n, c = i % n;
+        float acc = x[i];
+        for (int w = 0; w < world; w++) {
+            const uint8_t *row = bufs + ((int64_t)w * rows + r) * row_bytes;
+            const int8_t qv = (int8_t)row[c];
+            const uint16_t u = (uint16_t)row[n + 2 * (c / QB)]
+                             | ((uint16_t)row[n + 2 * (c / QB) + 1] << 8);
+            const __half h = *reinterpret_cast<const __half *>(&u);
+            acc = fmaf((float)qv, __half2float(h), acc);
+        }
+        x[i] = acc;
+    }
+}
+
+// x += y (residual merge for the f32/TP=1 path)
+__global__ void k_add(float *__restrict__ x, const float *__restrict__ y, int64_t n) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x)
+        x[i] += y[i];
+}
+
+__global__ void k_inc(int *__restrict__ pos, int delta) {
+    if (threadIdx.x == 0 && blockIdx.x == 0) pos[0] += delta;
+}
+
+#endif  // __HIPCC__
+
+// ============================================================== host bindings
+
+#define CHECK_CUDA(x) TORCH_CHECK((x).is_cuda(), #x " must be a GPU tensor")
+#define CHECK_CONT(x) TORCH_CHECK((x).is_contiguous(), #x " must be contiguous")
+
+static hipStream_t cur_stream() {
+    return at::hip::getCurrentHIPStream().stream();
+}
+
+void q80_quantize(torch::Tensor x, torch::Tensor q, torch::Tensor s, torch::Tensor bs) {
+    CHECK_CUDA(x); CHECK_CONT(x);
+    const int64_t blocks = x.numel() / QB;
+    const int threads = 256;
+    hipLaunchKernelGGL(k_q80_quantize, dim3(ceil_div(blocks * 32, threads)), dim3(threads),
+                       0, cur_stream(), x.data_ptr<float>(), q.data_ptr<int8_t>(),
+                       s.data_ptr<float>(), bs.data_ptr<float>(), (int)blocks);
+}
+
+void rmsnorm(torch::Tensor x, torch::Tensor w, torch::Tensor y, double eps) {
+    CHECK_CUDA(x);
+    const int n = x.size(-1);
+    const int rows = x.numel() / n;
+    hipLaunchKernelGGL(k_rmsnorm, dim3(rows), dim3(256), 0, cur_stream(),
+                       x.data_ptr<float>(), w.data_ptr<float>(), y.data_ptr<float>(),
+                       n, (float)eps);
+}
+
+void rmsnorm_q80(torch::Tensor x, torch::Tensor w, torch::Tensor q,
+                 torch::Tensor s, torch::Tensor bs, double eps) {
+    CHECK_CUDA(x);
+    const int n = x.size(-1);
+    const int rows = x.numel() / n;
+    hipLaunchKernelGGL(k_rmsnorm_q80, dim3(rows), dim3(256), 0, cur_stream(),
+                       x.data_ptr<float>(), w.data_ptr<float>(), q.data_ptr<int8_t>(),
+                       s.data_ptr<float>(), bs.data_ptr<float>(), n, (float)eps);
+}
+
+void rmsnorm_rows(torch::Tensor x, torch::Tensor w, torch::Tensor y, double eps) {
+    CHECK_CUDA(x);
+    const int hd = x.size(-1);
+    const int rows = x.numel() / hd;
+    hipLaunchKernelGGL(k_rmsnorm_rows, dim3(rows), dim3(64), 0, cur_stream(),
+                       x.data_ptr<float>(), w.data_ptr<float>(), y.data_ptr<float>(),
+                       hd, (float)eps);
+}
+
+void q40_gemv(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
+              torch::Tensor xs, torch::Tensor xbs, torch::Tensor y, int64_t batch) {
+    CHECK_CUDA(qs); CHECK_CONT(qs); CHECK_CONT(xq);
+    const int d = qs.size(0);
+    const int n = qs.size(1) * 2;
+    TORCH_CHECK(xq.size(-1) == n, "x width mismatch");
+    const int waves_per_block = 4;
+    const dim3 grid(ceil_div(d, waves_per_block));
+    const dim3 block(waves_per_block * WAVE);
+    auto launch = [&](auto nb_const) {
+        hipLaunchKernelGGL((k_q40_gemv<decltype(nb_const)::value>), grid, block, 0,
+                           cur_stream(),
+                           qs.data_ptr<uint8_t>(),
+                           reinterpret_cast<const __half *>(scales.data_ptr<at::Half>()),
+                           xq.data_ptr<int8_t>(), xs.data_ptr<float>(),
+                           xbs.data_ptr<float>(), y.data_ptr<float>(), d, n);
+    };
+    switch (batch) {
+        case 1: launch(std::integral_constant<int, 1>{}); break;
+        case 2: launch(std::integral_constant<int, 2>{}); break;
+        case 4: launch(std::integral_constant<int, 4>{}); break;
+        case 8: launch(std::integral_constant<int, 8>{}); break;
+        case 16: launch(std::integral_constant<int, 16>{}); break;
+        case 32: launch(std::integral_constant<int, 32>{}); break;
+        default: TORCH_CHECK(false, "unsupported batch ", batch);
+    }
+}
+
+void q40_gemv_grouped(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
+                      torch::Tensor xs, torch::Tensor xbs, torch::Tensor expert_idx,
+                      torch::Tensor y, int64_t k_slots) {
+    CHECK_CUDA(qs); CHECK_CONT(qs);
+    const int d = qs.size(1);
+    const int n = qs.size(2) * 2;
+    const int n_slots = expert_idx.numel();
+    const int waves_per_block = 4;
+    const dim3 grid(ceil_div(d, waves_per_block), n_slots);
+    hipLaunchKernelGGL(k_q40_gemv_grouped, grid, dim3(waves_per_block * WAVE), 0,
+                       cur_stream(), qs.data_ptr<uint8_t>(),
+                       reinterpret_cast<const __half *>(scales.data_ptr<at::Half>()),
+                       xq.data_ptr<int8_t>(), xs.data_ptr<float>(),
+                       xbs.data_ptr<float>(), expert_idx.data_ptr<int>(),
+                       y.data_ptr<float>(), d, n, (int)k_slots);
+}
+
+void rope(torch::Tensor x, torch::Tensor cache, torch::Tensor pos,
+          int64_t head_dim, int64_t style) {
+    CHECK_CUDA(x);
+    const int B = x.size(0);
+    const int dim0 = x.size(1);
+    const dim3 grid(ceil_div(dim0 / 2, 256), B);
+    if (style == 0)
+        hipLaunchKernelGGL(k_rope<0>, grid, dim3(256), 0, cur_stream(),
+                           x.data_ptr<float>(), cache.data_ptr<float>(),
+                           pos.data_ptr<int>(), dim0, (int)head_dim);
+    else
+        hipLaunchKernelGGL(k_rope<1>, grid, dim3(256), 0, cur_stream(),
+                           x.data_ptr<float>(), cache.data_ptr<float>(),
+                           pos.data_ptr<int>(), dim0, (int)head_dim);
+}
+
+void kv_append(torch::Tensor k, torch::Tensor v, torch::Tensor kc,
+               torch::Tensor vc, torch::Tensor pos) {
+    CHECK_CUDA(k);
+    const int B = k.size(0);
+    const int kv_dim0 = k.size(1);
+    const dim3 grid(ceil_div(kv_dim0, 256), B);
+    hipLaunchKernelGGL(k_kv_append, grid, dim3(256), 0, cur_stream(),
+                       k.data_ptr<float>(), v.data_ptr<float>(),
+                       kc.data_ptr<float>(), vc.data_ptr<float>(),
+                       pos.data_ptr<int>(), kv_dim0);
+}
+
+void attn(torch::Tensor q, torch::Tensor kc, torch::Tensor vc, torch::Tensor y,
+          torch::Tensor pos, int64_t batch, int64_t n_heads0, int64_t kv_mul,
+          int64_t head_dim) {
+    CHECK_CUDA(q);
+    const int kv_dim0 = kc.size(1);
+    const float scale = 1.0f / sqrtf((float)head_dim);
+    const dim3 grid(n_heads0, batch);
+    if (head_dim == 128)
+        hipLaunchKernelGGL(k_attn<2>, grid, dim3(256), 0, cur_stream(),
+                           q.data_ptr<float>(), kc.data_ptr<float>(), vc.data_ptr<float>(),
+                           y.data_ptr<float>(), pos.data_ptr<int>(),
+                           (int)n_heads0, (int)kv_mul, kv_dim0, scale);
+    else if (head_dim == 64)
+        hipLaunchKernelGGL(k_attn<1>, grid, dim3(256), 0, cur_stream(),
+                           q.data_ptr<float>(), kc.data_ptr<float>(), vc.data_ptr<float>(),
+                           y.data_ptr<float>(), pos.data_ptr<int>(),
+                           (int)n_heads0, (int)kv_mul, kv_dim0, scale);
+    else
+        TORCH_CHECK(false, "unsupported head_dim ", head_dim);
+}
+
+void swiglu_q80(torch::Tensor a, torch::Tensor g, torch::Tensor q,
+                torch::Tensor s, torch::Tensor bs) {
+    CHECK_CUDA(a);
+    const int64_t blocks = a.numel() / QB;
+    hipLaunchKernelGGL(k_swiglu_q80, dim3(ceil_div(blocks * 32, 256)), dim3(256), 0,
+                       cur_stream(), a.data_ptr<float>(), g.data_ptr<float>(),
+                       q.data_ptr<int8_t>(), s.data_ptr<float>(), bs.data_ptr<float>(),
+                       (int)blocks);
+}
+
+void silu_mul(torch::Tensor a, torch::Tensor g, torch::Tensor out) {
+    CHECK_CUDA(a);
+    const int64_t n = a.numel();
+    hipLaunchKernelGGL(k_silu_mul, dim3(ceil_div(n, 256)), dim3(256), 0, cur_stream(),
+                       a.data_ptr<float>(), g.data_ptr<float>(), out.data_ptr<float>(), n);
+}
+
+void sync_pack(torch::Tensor q, torch::Tensor s, torch::Tensor buf) {
+    CHECK_CUDA(q);
+    const int n = q.size(-1);
+    const int rows = q.numel() / n;
+    hipLaunchKernelGGL(k_sync_pack, dim3(ceil_div((int64_t)rows * n, 256)), dim3(256),
+                       0, cur_stream(), q.data_ptr<int8_t>(), s.data_ptr<float>(),
+                       buf.data_ptr<uint8_t>(), n, rows);
+}
+
+void merge_add(torch::Tensor x, torch::Tensor bufs) {
+    CHECK_CUDA(x);
+    const int n = x.size(-1);
+    const int rows = x.numel() / n;
+    const int world = bufs.size(0);
+    hipLaunchKernelGGL(k_merge_add, dim3(ceil_div((int64_t)rows * n, 256)), dim3(256),
+                       0, cur_stream(), x.data_ptr<float>(), bufs.data_ptr<uint8_t>(),
+                       world, n, rows);
+}
+
+void add_(torch::Tensor x, torch::Tensor y) {
+    CHECK_CUDA(x);
+    hipLaunchKernelGGL(k_add, dim3(ceil_div(x.numel(), 256)), dim3(256), 0,
+                       cur_stream(), x.data_ptr<float>(), y.data_ptr<float>(), x.numel());
+}
+
+void pos_inc(torch::Tensor pos, int64_t delta) {
+    hipLaunchKernelGGL(k_inc, dim3(1), dim3(64), 0, cur_stream(),
+                       pos.data_ptr<int>(), (int)delta);
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("q80_quantize", &q80_quantize);
+    m.def("rmsnorm", &rmsnorm);
+    m.def("rmsnorm_q80", &rmsnorm_q80);
+    m.def("rmsnorm_rows", &rmsnorm_rows);
+    m.def("q40_gemv", &q40_gemv);
+    m.def("q40_gemv_grouped", &q40_gemv_grouped);
+    m.def("rope", &rope);
+    m.def("kv_append", &kv_append);
+    m.def("attn", &attn);
+    m.def("swiglu_q80", &swiglu_q80);
+    m.def("silu_mul", &silu_mul);
+    m.def("sync_pack", &sync_pack);
+    m.def("merge_add", &merge_add);
+    m.def("add_", &add_);
+    m.def("pos_inc", &pos_inc);
+}
