@@ -34,7 +34,7 @@ def make_byte_tokenizer(path: str, chat_template: str | None = None) -> None:
 def make_tiny_llama(path: str, seed: int = 7, vocab_size: int = 512,
                     seq_len: int = 128) -> mf.LlmHeader:
     h = mf.LlmHeader(arch_type=mf.ARCH_LLAMA, dim=64, hidden_dim=128, n_layers=2,
-                     n_heads=4, n_kv_heads=2, head_dim=16, vocab_size=vocab_size,
+                     n_heads=4, n_kv_heads=2, head_dim=64, vocab_size=vocab_size,
                      seq_len=seq_len, rope_theta=10000, rope_type=mf.ROPE_LLAMA)
     h.finalize()
     mf.write_synthetic_model(path, h, seed=seed)
@@ -44,13 +44,13 @@ def make_tiny_llama(path: str, seed: int = 7, vocab_size: int = 512,
 def make_tiny_qwen3(path: str, seed: int = 9, moe: bool = False) -> mf.LlmHeader:
     if moe:
         h = mf.LlmHeader(arch_type=mf.ARCH_QWEN3_MOE, dim=64, hidden_dim=96,
-                         n_layers=2, n_heads=4, n_kv_heads=2, head_dim=16,
+                         n_layers=2, n_heads=4, n_kv_heads=2, head_dim=64,
                          n_experts=4, n_active_experts=2, moe_hidden_dim=32,
                          vocab_size=256, seq_len=64, rope_theta=10000,
                          norm_epsilon=1e-6)
     else:
         h = mf.LlmHeader(arch_type=mf.ARCH_QWEN3, dim=64, hidden_dim=96,
-                         n_layers=2, n_heads=4, n_kv_heads=2, head_dim=16,
+                         n_layers=2, n_heads=4, n_kv_heads=2, head_dim=64,
                          vocab_size=256, seq_len=64, rope_theta=10000,
                          norm_epsilon=1e-6)
     h.finalize()
