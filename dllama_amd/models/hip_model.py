@@ -110,26 +110,38 @@ class HipTransformer:
         self.final_norm = f32("final_norm")
         self.wcls = lin("final_matmul_logits", -1, c.vocab0, c.dim)
         for l in range(c.n_layers):
+            # fuse Q|K|V (and W1|W3) into single GEMV weights: one launch,
+            # one pass over x, full-chip grids even for the small K/V shards
+            q = lin("block_matmul_q", l, c.q_dim0, c.dim)
+            kk = lin("block_matmul_k", l, c.kv_dim0, c.dim)
+            v = lin("block_matmul_v", l, c.kv_dim0, c.dim)
             lw = {
-                "q": lin("block_matmul_q", l, c.q_dim0, c.dim),
-                "k": lin("block_matmul_k", l, c.kv_dim0, c.dim),
-                "v": lin("block_matmul_v", l, c.kv_dim0, c.dim),
+                "qkv": Linear(torch.cat([q.qs, kk.qs, v.qs]),
+                              torch.cat([q.scales, kk.scales, v.scales])),
                 "wo": lin("block_matmul_wo", l, c.dim, c.q_dim0),
                 "norm0": f32("block_norm_0", l),
                 "norm1": f32("block_norm_1", l),
             }
             if c.is_moe:
                 lw["gate"] = f32("block_moe_gate", l)
-                for wn, dd, nn in (("w1", c.ff_dim0, c.dim), ("w2", c.dim, c.ff_dim0),
-                                   ("w3", c.ff_dim0, c.dim)):
-                    ls = [lin(f"block_matmul_{wn}", l, dd, nn, e)
-                          for e in range(c.n_experts)]
-                    lw[wn] = Linear(torch.stack([x.qs for x in ls]),
-                                    torch.stack([x.scales for x in ls]))
+                w13 = []
+                for e in range(c.n_experts):
+                    w1 = lin("block_matmul_w1", l, c.ff_dim0, c.dim, e)
+                    w3 = lin("block_matmul_w3", l, c.ff_dim0, c.dim, e)
+                    w13.append(Linear(torch.cat([w1.qs, w3.qs]),
+                                      torch.cat([w1.scales, w3.scales])))
+                lw["w13"] = Linear(torch.stack([x.qs for x in w13]),
+                                   torch.stack([x.scales for x in w13]))
+                w2 = [lin("block_matmul_w2", l, c.dim, c.ff_dim0, e)
+                      for e in range(c.n_experts)]
+                lw["w2"] = Linear(torch.stack([x.qs for x in w2]),
+                                  torch.stack([x.scales for x in w2]))
             else:
-                lw["w1"] = lin("block_matmul_w1", l, c.ff_dim0, c.dim)
+                w1 = lin("block_matmul_w1", l, c.ff_dim0, c.dim)
+                w3 = lin("block_matmul_w3", l, c.ff_dim0, c.dim)
+                lw["w13"] = Linear(torch.cat([w1.qs, w3.qs]),
+                                   torch.cat([w1.scales, w3.scales]))
                 lw["w2"] = lin("block_matmul_w2", l, c.dim, c.ff_dim0)
-                lw["w3"] = lin("block_matmul_w3", l, c.ff_dim0, c.dim)
             if c.is_qwen3:
                 lw["q_norm"] = f32("block_norm_q", l)
                 lw["k_norm"] = f32("block_norm_k", l)
@@ -152,9 +164,7 @@ class HipTransformer:
         self.wcls = Linear.synthetic(c.vocab0, c.dim, dev, gen)
         for l in range(c.n_layers):
             lw = {
-                "q": Linear.synthetic(c.q_dim0, c.dim, dev, gen),
-                "k": Linear.synthetic(c.kv_dim0, c.dim, dev, gen),
-                "v": Linear.synthetic(c.kv_dim0, c.dim, dev, gen),
+                "qkv": Linear.synthetic(c.q_dim0 + 2 * c.kv_dim0, c.dim, dev, gen),
                 "wo": Linear.synthetic(c.dim, c.q_dim0, dev, gen),
                 "norm0": torch.ones(c.dim, device=dev),
                 "norm1": torch.ones(c.dim, device=dev),
@@ -162,17 +172,16 @@ class HipTransformer:
             if c.is_moe:
                 lw["gate"] = torch.randn(c.n_experts, c.dim, device=dev,
                                          generator=gen) * 0.02
-                for wn, dd, nn in (("w1", c.ff_dim0, c.dim), ("w2", c.dim, c.ff_dim0),
-                                   ("w3", c.ff_dim0, c.dim)):
+                for wn, dd, nn in (("w13", 2 * c.ff_dim0, c.dim),
+                                   ("w2", c.dim, c.ff_dim0)):
                     qs = torch.randint(0, 256, (c.n_experts, dd, nn // 2),
                                        dtype=torch.uint8, device=dev, generator=gen)
                     sc = (torch.rand((c.n_experts, dd, nn // QB), device=dev,
                                      generator=gen) * 0.02 / 8).to(torch.float16)
                     lw[wn] = Linear(qs, sc)
             else:
-                lw["w1"] = Linear.synthetic(c.ff_dim0, c.dim, dev, gen)
+                lw["w13"] = Linear.synthetic(2 * c.ff_dim0, c.dim, dev, gen)
                 lw["w2"] = Linear.synthetic(c.dim, c.ff_dim0, dev, gen)
-                lw["w3"] = Linear.synthetic(c.ff_dim0, c.dim, dev, gen)
             if c.is_qwen3:
                 lw["q_norm"] = torch.ones(c.head_dim, device=dev)
                 lw["k_norm"] = torch.ones(c.head_dim, device=dev)
@@ -187,17 +196,15 @@ class HipTransformer:
         self.x = torch.zeros(NB, c.dim, device=dev)
         self.t_norm = torch.zeros(NB, c.dim, device=dev)
         self.xq = QuantBuf(NB, c.dim, dev)
-        self.qbuf = torch.zeros(NB, c.q_dim0, device=dev)
-        self.kbuf = torch.zeros(NB, c.kv_dim0, device=dev)
-        self.vbuf = torch.zeros(NB, c.kv_dim0, device=dev)
+        self.qkv_ld = c.q_dim0 + 2 * c.kv_dim0
+        self.qkv_out = torch.zeros(NB, self.qkv_ld, device=dev)
         self.zbuf = torch.zeros(NB, c.q_dim0, device=dev)
         self.zq = QuantBuf(NB, c.q_dim0, dev)
         self.partial = torch.zeros(NB, c.dim, device=dev)
-        ffw = c.ff_dim0
-        self.abuf = torch.zeros(NB, ffw, device=dev)
-        self.gbuf = torch.zeros(NB, ffw, device=dev)
-        self.dq = QuantBuf(NB, ffw, dev)
+        self.ff_out = torch.zeros(NB, 2 * c.ff_dim0, device=dev)
+        self.dq = QuantBuf(NB, c.ff_dim0, dev)
         self.logits0 = torch.zeros(NB, c.vocab0, device=dev)
+        self.amax_slot = torch.zeros(1, dtype=torch.int64, device=dev)
         if c.world > 1:
             self.logits_gather = torch.zeros(c.world, NB, c.vocab0, device=dev)
             if c.sync_type == Q80:
@@ -208,9 +215,7 @@ class HipTransformer:
         if c.is_moe:
             S = NB * c.n_active_experts
             self.moe_idx = torch.zeros(S, dtype=torch.int32, device=dev)
-            self.moe_w = torch.zeros(NB, c.n_active_experts, device=dev)
-            self.moe_a = torch.zeros(S, c.ff_dim0, device=dev)
-            self.moe_g = torch.zeros(S, c.ff_dim0, device=dev)
+            self.moe_out13 = torch.zeros(S, 2 * c.ff_dim0, device=dev)
             self.moe_dq = QuantBuf(S, c.ff_dim0, dev)
             self.moe_y = torch.zeros(S, c.dim, device=dev)
 
@@ -227,11 +232,8 @@ class HipTransformer:
     # ------------------------------------------------------------ forward
 
     def _sync_partial(self, B: int, NB: int):
-        """All-reduce self.partial[:NB] across ranks into x += sum(partials)."""
+        """TP>1: all-reduce self.partial[:NB] into x (+= sum of partials)."""
         c = self.cfg
-        if c.world == 1:
-            self.k.add_(self.x[:NB], self.partial[:NB])
-            return
         if c.sync_type == Q80:
             nb_dim = c.dim // QB
             q = self.xq  # reuse dim-sized quant buffer
@@ -246,6 +248,20 @@ class HipTransformer:
             self.comm.allreduce_(self.partial[:NB])
             self.k.add_(self.x[:NB], self.partial[:NB])
 
+    def _merge_norm(self, pending, w, NB: int, quant: bool):
+        """Fold the pending partial into x and norm: TP=1 fuses
+        add+rmsnorm(+q80) into one kernel; TP>1 syncs first."""
+        k = self.k
+        if pending is not None and self.cfg.world > 1:
+            self._sync_partial(NB, NB)
+            pending = None
+        if quant:
+            k.add_rmsnorm_q80(self.x[:NB], pending, w, self.xq.q[:NB],
+                              self.xq.s[:NB], self.xq.bs[:NB], self.cfg.norm_eps)
+        else:
+            k.add_rmsnorm(self.x[:NB], pending, w, self.t_norm[:NB],
+                          self.cfg.norm_eps)
+
     def forward_buffers(self, B: int):
         """Run one step over tokens[:B] at positions pos..pos+B-1, writing
         logits into logits0 (and the gather buffer under TP). Everything
@@ -254,69 +270,67 @@ class HipTransformer:
         NB = _pow2_batch(B)
         x = self.x
         torch.index_select(self.embedding, 0, self.tokens[:NB], out=x[:NB])
+        pending = None  # partial output not yet folded into the residual
 
         kv_mul = c.n_heads0 // max(1, c.kv_dim0 // c.head_dim)
         for l, lw in enumerate(self.layers):
             # attention block
-            k.rmsnorm_q80(x[:NB], lw["norm0"], self.xq.q[:NB], self.xq.s[:NB],
-                          self.xq.bs[:NB], c.norm_eps)
-            k.q40_gemv(lw["q"].qs, lw["q"].scales, self.xq.q, self.xq.s,
-                       self.xq.bs, self.qbuf, NB)
-            k.q40_gemv(lw["k"].qs, lw["k"].scales, self.xq.q, self.xq.s,
-                       self.xq.bs, self.kbuf, NB)
-            k.q40_gemv(lw["v"].qs, lw["v"].scales, self.xq.q, self.xq.s,
-                       self.xq.bs, self.vbuf, NB)
+            self._merge_norm(pending, lw["norm0"], NB, quant=True)
+            k.q40_gemv(lw["qkv"].qs, lw["qkv"].scales, self.xq.q, self.xq.s,
+                       self.xq.bs, self.qkv_out, NB)
             if c.is_qwen3:
-                k.rmsnorm_rows(self.qbuf[:B].view(-1, c.head_dim), lw["q_norm"],
-                               self.qbuf[:B].view(-1, c.head_dim), c.norm_eps)
-                k.rmsnorm_rows(self.kbuf[:B].view(-1, c.head_dim), lw["k_norm"],
-                               self.kbuf[:B].view(-1, c.head_dim), c.norm_eps)
-            k.rope(self.qbuf[:B], self.rope_cache, self.pos, c.head_dim, self.rope_style)
-            k.rope(self.kbuf[:B], self.rope_cache, self.pos, c.head_dim, self.rope_style)
-            k.kv_append(self.kbuf[:B], self.vbuf[:B], self.k_cache[l],
-                        self.v_cache[l], self.pos)
-            k.attn(self.qbuf[:B], self.k_cache[l], self.v_cache[l], self.zbuf[:B],
-                   self.pos, B, c.n_heads0, kv_mul, c.head_dim)
+                k.rmsnorm_rows_s(self.qkv_out, self.qkv_ld, 0,
+                                 c.q_dim0 // c.head_dim, B, lw["q_norm"],
+                                 c.head_dim, c.norm_eps)
+                k.rmsnorm_rows_s(self.qkv_out, self.qkv_ld, c.q_dim0,
+                                 c.kv_dim0 // c.head_dim, B, lw["k_norm"],
+                                 c.head_dim, c.norm_eps)
+            k.rope_kv(self.qkv_out, self.qkv_ld, c.q_dim0, c.kv_dim0,
+                      self.rope_cache, self.pos, self.k_cache[l], self.v_cache[l],
+                      c.head_dim, self.rope_style, B)
+            k.attn(self.qkv_out, self.qkv_ld, self.k_cache[l], self.v_cache[l],
+                   self.zbuf[:B], self.pos, B, c.n_heads0, kv_mul, c.head_dim)
             k.q80_quantize(self.zbuf[:NB], self.zq.q[:NB], self.zq.s[:NB],
                            self.zq.bs[:NB])
             k.q40_gemv(lw["wo"].qs, lw["wo"].scales, self.zq.q, self.zq.s,
                        self.zq.bs, self.partial, NB)
-            self._sync_partial(B, NB)
+            pending = self.partial
 
             # ffn block
-            k.rmsnorm(x[:NB], lw["norm1"], self.t_norm[:NB], c.norm_eps)
             if c.is_moe:
+                self._merge_norm(pending, lw["norm1"], NB, quant=False)
+                pending = None
                 self._moe_ffn(B, NB, lw)
             else:
-                k.q80_quantize(self.t_norm[:NB], self.xq.q[:NB], self.xq.s[:NB],
-                               self.xq.bs[:NB])
-                k.q40_gemv(lw["w1"].qs, lw["w1"].scales, self.xq.q, self.xq.s,
-                           self.xq.bs, self.abuf, NB)
-                k.q40_gemv(lw["w3"].qs, lw["w3"].scales, self.xq.q, self.xq.s,
-                           self.xq.bs, self.gbuf, NB)
-                k.swiglu_q80(self.abuf[:NB], self.gbuf[:NB], self.dq.q[:NB],
+                self._merge_norm(pending, lw["norm1"], NB, quant=True)
+                k.q40_gemv(lw["w13"].qs, lw["w13"].scales, self.xq.q, self.xq.s,
+                           self.xq.bs, self.ff_out, NB)
+                k.swiglu_q80(self.ff_out, self.ff_out[:, c.ff_dim0:],
+                             2 * c.ff_dim0, c.ff_dim0, NB, self.dq.q[:NB],
                              self.dq.s[:NB], self.dq.bs[:NB])
                 k.q40_gemv(lw["w2"].qs, lw["w2"].scales, self.dq.q, self.dq.s,
                            self.dq.bs, self.partial, NB)
-            self._sync_partial(B, NB)
+            pending = self.partial
 
-        k.rmsnorm_q80(x[:NB], self.final_norm, self.xq.q[:NB], self.xq.s[:NB],
-                      self.xq.bs[:NB], c.norm_eps)
+        self._merge_norm(pending, self.final_norm, NB, quant=True)
+        use_amax = (self.greedy_feedback and B == 1 and c.world == 1)
         k.q40_gemv(self.wcls.qs, self.wcls.scales, self.xq.q, self.xq.s,
-                   self.xq.bs, self.logits0, NB)
+                   self.xq.bs, self.logits0, NB,
+                   self.amax_slot if use_amax else None)
         if c.world > 1:
             self.comm.all_gather(self.logits_gather[:, :NB], self.logits0[:NB])
         if self.greedy_feedback and B == 1:
             # on-device greedy sampling feeding the next decode step (used by
             # the fully graph-captured bench loop; real serving samples on host)
-            if c.world > 1:
+            if use_amax:
+                k.token_from_argmax(self.tokens, self.amax_slot)
+            else:
                 full = self.logits_gather[:, 0].reshape(-1)
                 self.tokens[0].copy_(torch.argmax(full))
-            else:
-                self.tokens[0].copy_(torch.argmax(self.logits0[0]))
 
     def _moe_ffn(self, B: int, NB: int, lw: dict):
-        """Router (torch) + grouped expert GEMVs (reference llm.cpp:450-487)."""
+        """Router (torch) + grouped expert GEMVs (reference llm.cpp:450-487);
+        t_norm holds the f32 normed activations (router input)."""
         c, k = self.cfg, self.k
         ka = c.n_active_experts
         router = self.t_norm[:NB] @ lw["gate"].t()
@@ -327,11 +341,10 @@ class HipTransformer:
         S = NB * ka
         k.q80_quantize(self.t_norm[:NB], self.xq.q[:NB], self.xq.s[:NB],
                        self.xq.bs[:NB])
-        k.q40_gemv_grouped(lw["w1"].qs, lw["w1"].scales, self.xq.q, self.xq.s,
-                           self.xq.bs, self.moe_idx[:S], self.moe_a, ka)
-        k.q40_gemv_grouped(lw["w3"].qs, lw["w3"].scales, self.xq.q, self.xq.s,
-                           self.xq.bs, self.moe_idx[:S], self.moe_g, ka)
-        k.swiglu_q80(self.moe_a[:S], self.moe_g[:S], self.moe_dq.q[:S],
+        k.q40_gemv_grouped(lw["w13"].qs, lw["w13"].scales, self.xq.q, self.xq.s,
+                           self.xq.bs, self.moe_idx[:S], self.moe_out13, ka)
+        k.swiglu_q80(self.moe_out13, self.moe_out13[:, c.ff_dim0:],
+                     2 * c.ff_dim0, c.ff_dim0, S, self.moe_dq.q[:S],
                      self.moe_dq.s[:S], self.moe_dq.bs[:S])
         k.q40_gemv_grouped(lw["w2"].qs, lw["w2"].scales, self.moe_dq.q,
                            self.moe_dq.s, self.moe_dq.bs, self.moe_idx[:S],

@@ -172,6 +172,106 @@ __global__ void k_rmsnorm_rows(const float *__restrict__ x,
         out[i] = row[i] * inv * w[i];
 }
 
+// per-head rmsnorm on a slice of the fused QKV buffer (Qwen3 q/k-norm on
+// strided rows): batch b, head h -> row at buf + b*ld + off + h*hd.
+__global__ void k_rmsnorm_rows_s(float *__restrict__ buf, int ld, int off,
+                                 int heads, const float *__restrict__ w,
+                                 int hd, float eps) {
+    const int b = blockIdx.x / heads;
+    const int h = blockIdx.x % heads;
+    float *row = buf + (int64_t)b * ld + off + h * hd;
+    float acc = 0.0f;
+    for (int i = threadIdx.x; i < hd; i += blockDim.x) {
+        const float v = row[i];
+        acc += v * v;
+    }
+    acc = wave_reduce_sum(acc);  // blockDim == 64
+    const float inv = rsqrtf(acc / hd + eps);
+    for (int i = threadIdx.x; i < hd; i += blockDim.x)
+        row[i] = row[i] * inv * w[i];
+}
+
+// ---------------------------------------------- fused residual+rmsnorm(+q80)
+// x[row] += partial[row] (ADD), then rmsnorm with w; output either f32 y
+// (!QUANT) or the Q80 triple (QUANT). Fuses the reference's
+// merge_add -> inv_rms -> rms_norm -> cast chain (llm.cpp:263-270) into one
+// kernel: one extra pass saved per layer half, and the B=1 row gets a full
+// 1024-thread workgroup with float4 traffic.
+template <bool ADD, bool QUANT>
+__global__ void k_add_rmsnorm(float *__restrict__ x,
+                              const float *__restrict__ partial,
+                              const float *__restrict__ w,
+                              float *__restrict__ y,
+                              int8_t *__restrict__ q,
+                              float *__restrict__ s,
+                              float *__restrict__ bs,
+                              int n, float eps) {
+    const int64_t base = (int64_t)blockIdx.x * n;
+    float acc = 0.0f;
+    for (int i = threadIdx.x * 4; i < n; i += blockDim.x * 4) {
+        float4 v = *reinterpret_cast<const float4 *>(x + base + i);
+        if (ADD) {
+            const float4 p = *reinterpret_cast<const float4 *>(partial + base + i);
+            v.x += p.x; v.y += p.y; v.z += p.z; v.w += p.w;
+            *reinterpret_cast<float4 *>(x + base + i) = v;
+        }
+        acc += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+    }
+    __shared__ float red[16];
+    acc = wave_reduce_sum(acc);
+    const int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    if (lane == 0) red[wid] = acc;
+    __syncthreads();
+    const int nw = blockDim.x / WAVE;
+    float total = 0.0f;
+    for (int i = 0; i < nw; i++) total += red[i];
+    const float inv = rsqrtf(total / n + eps);
+
+    if (!QUANT) {
+        for (int i = threadIdx.x * 4; i < n; i += blockDim.x * 4) {
+            const float4 v = *reinterpret_cast<const float4 *>(x + base + i);
+            const float4 wv = *reinterpret_cast<const float4 *>(w + i);
+            float4 o;
+            o.x = v.x * inv * wv.x; o.y = v.y * inv * wv.y;
+            o.z = v.z * inv * wv.z; o.w = v.w * inv * wv.w;
+            *reinterpret_cast<float4 *>(y + base + i) = o;
+        }
+    } else {
+        const int nb = n / QB;
+        const int lane32 = threadIdx.x & 31;
+        for (int blk = threadIdx.x / 32; blk < nb; blk += blockDim.x / 32) {
+            const int i = blk * QB + lane32;
+            const float v = x[base + i] * inv * w[i];
+            const float amax = group32_reduce_max(fabsf(v));
+            const float d = amax / 127.0f;
+            const float qinv = d > 0.0f ? 1.0f / d : 0.0f;
+            const float qf = rintf(v * qinv);
+            q[base + i] = (int8_t)qf;
+            const float bsum = group32_reduce_sum(qf);
+            if (lane32 == 0) {
+                s[(int64_t)blockIdx.x * nb + blk] = d;
+                bs[(int64_t)blockIdx.x * nb + blk] = bsum;
+            }
+        }
+    }
+}
+
+// argmax pack: monotonic unsigned ordering of (float value, smallest index
+// wins ties) for a single global atomicMax — on-device greedy sampling.
+__device__ __forceinline__ unsigned long long argmax_pack(float v, int idx) {
+    unsigned u = __float_as_uint(v);
+    u = (u & 0x80000000u) ? ~u : (u | 0x80000000u);
+    return ((unsigned long long)u << 32) | (unsigned)(0x7FFFFFFF - idx);
+}
+
+__global__ void k_token_from_argmax(long *__restrict__ token,
+                                    unsigned long long *__restrict__ slot) {
+    if (threadIdx.x == 0 && blockIdx.x == 0) {
+        token[0] = (long)(0x7FFFFFFF - (int)(slot[0] & 0xFFFFFFFFu));
+        slot[0] = 0ull;  // reset for the next decode step
+    }
+}
+
 // ------------------------------------------------------------------ Q40 GEMV
 // y[b, row] = sum_j w[row, j] * x[b, j]  with W in Q40 planes and x in Q80.
 // One wave per output row; lane l streams block l, l+64, ... of the row.
@@ -181,6 +281,19 @@ __global__ void k_rmsnorm_rows(const float *__restrict__ x,
 // so the nibble unpack is 2 VALU ops per 8 elems and sdot4 does the MAC
 // (role of reference matmul_Q80_Q40_F32, nn-cpu-ops.cpp:231-449, and the
 //  matmul-forward-q80-q40-f32.comp Vulkan shader).
+__device__ __forceinline__ int q40_block_dot(const uint4 wq, const int4 x0,
+                                             const int4 x1) {
+    const uint32_t wv[4] = {wq.x, wq.y, wq.z, wq.w};
+    const int32_t xv[8] = {x0.x, x0.y, x0.z, x0.w, x1.x, x1.y, x1.z, x1.w};
+    int idot = 0;
+    #pragma unroll
+    for (int wi = 0; wi < 4; wi++) {
+        idot = dot4((int)(wv[wi] & 0x0F0F0F0Fu), xv[wi], idot);            // elems 4wi..4wi+3
+        idot = dot4((int)((wv[wi] >> 4) & 0x0F0F0F0Fu), xv[4 + wi], idot); // elems 16+4wi..
+    }
+    return idot;
+}
+
 template <int NB>
 __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
                            const __half *__restrict__ scales,
@@ -188,11 +301,13 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
                            const float *__restrict__ xs,
                            const float *__restrict__ xbs,
                            float *__restrict__ y,
-                           int d, int n) {
+                           int d, int n,
+                           unsigned long long *__restrict__ amax_slot) {
     const int row = blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
     if (row >= d) return;
     const int lane = threadIdx.x % WAVE;
     const int nb = n / QB;
+    const int nbp = nb >> 1;  // block pairs (n % 64 == 0 asserted host-side)
     const uint4 *wrow = reinterpret_cast<const uint4 *>(qs + (int64_t)row * (n >> 1));
     const __half *srow = scales + (int64_t)row * nb;
 
@@ -200,30 +315,31 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
     #pragma unroll
     for (int b = 0; b < NB; b++) acc[b] = 0.0f;
 
-    for (int j = lane; j < nb; j += WAVE) {
-        const uint4 wq = wrow[j];
-        const float sw = __half2float(srow[j]);
-        const uint32_t wv[4] = {wq.x, wq.y, wq.z, wq.w};
+    for (int jp = lane; jp < nbp; jp += WAVE) {
+        const int j = jp << 1;
+        const uint4 wq0 = wrow[j];
+        const uint4 wq1 = wrow[j + 1];
+        const float2 sw = __half22float2(
+            *reinterpret_cast<const __half2 *>(srow + j));
         #pragma unroll
         for (int b = 0; b < NB; b++) {
-            const int4 x0 = reinterpret_cast<const int4 *>(xq + (int64_t)b * n + j * QB)[0];
-            const int4 x1 = reinterpret_cast<const int4 *>(xq + (int64_t)b * n + j * QB)[1];
-            const int32_t xv[8] = {x0.x, x0.y, x0.z, x0.w, x1.x, x1.y, x1.z, x1.w};
-            int idot = 0;
-            #pragma unroll
-            for (int wi = 0; wi < 4; wi++) {
-                idot = dot4((int)(wv[wi] & 0x0F0F0F0Fu), xv[wi], idot);          // elems 4wi..4wi+3
-                idot = dot4((int)((wv[wi] >> 4) & 0x0F0F0F0Fu), xv[4 + wi], idot); // elems 16+4wi..
-            }
-            const float sx = xs[b * nb + j];
-            const float bsum = xbs[b * nb + j];
-            acc[b] = fmaf(sw * sx, (float)idot - 8.0f * bsum, acc[b]);
+            const int4 *xrow = reinterpret_cast<const int4 *>(xq + (int64_t)b * n) + j * 2;
+            const int idot0 = q40_block_dot(wq0, xrow[0], xrow[1]);
+            const int idot1 = q40_block_dot(wq1, xrow[2], xrow[3]);
+            const float2 sx = *reinterpret_cast<const float2 *>(xs + (int64_t)b * nb + j);
+            const float2 bsum = *reinterpret_cast<const float2 *>(xbs + (int64_t)b * nb + j);
+            acc[b] = fmaf(sw.x * sx.x, (float)idot0 - 8.0f * bsum.x, acc[b]);
+            acc[b] = fmaf(sw.y * sx.y, (float)idot1 - 8.0f * bsum.y, acc[b]);
         }
     }
     #pragma unroll
     for (int b = 0; b < NB; b++) {
         float r = wave_reduce_sum(acc[b]);
-        if (lane == 0) y[(int64_t)b * d + row] = r;
+        if (lane == 0) {
+            y[(int64_t)b * d + row] = r;
+            if (NB == 1 && amax_slot != nullptr)
+                atomicMax(amax_slot, argmax_pack(r, row));
+        }
     }
 }
 
@@ -304,6 +420,59 @@ __global__ void k_rope(float *__restrict__ x,
     }
 }
 
+// ------------------------------------------- fused rope(q,k) + kv append
+// Operates directly on the fused QKV GEMV output buffer ([B, ld] with
+// q | k | v packed per row): rotates q in place, rotates k and writes it to
+// the cache row pos+b, and copies v to the cache — one launch replacing
+// rope(q), rope(k), kv_append (reference runs 4 separate ops here,
+// llm.cpp:300-330).
+template <int STYLE>
+__global__ void k_rope_kv(float *__restrict__ qkv, int ld,
+                          int q_dim0, int kv_dim0,
+                          const float *__restrict__ cache,
+                          const int *__restrict__ pos,
+                          float *__restrict__ kc,
+                          float *__restrict__ vc,
+                          int hd) {
+    const int b = blockIdx.y;
+    const int p = pos[0] + b;
+    const int half = hd >> 1;
+    const float *pc = cache + (int64_t)p * hd;
+    float *qrow = qkv + (int64_t)b * ld;
+    float *krow = qrow + q_dim0;
+    const float *vrow = krow + kv_dim0;
+    const int qp = q_dim0 >> 1, kp = kv_dim0 >> 1, vq = kv_dim0 >> 2;
+    const int total = qp + kp + vq;
+    for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+         i += gridDim.x * blockDim.x) {
+        if (i < qp + kp) {
+            const bool is_q = i < qp;
+            const int ii = is_q ? i : i - qp;
+            const int head = ii / half;
+            const int j = ii % half;
+            int i0, i1;
+            if (STYLE == 0) { i0 = head * hd + 2 * j; i1 = i0 + 1; }
+            else            { i0 = head * hd + j;     i1 = i0 + half; }
+            const float cr = pc[2 * j];
+            const float ci = pc[2 * j + 1];
+            float *src = is_q ? qrow : krow;
+            const float v0 = src[i0];
+            const float v1 = src[i1];
+            const float o0 = v0 * cr - v1 * ci;
+            const float o1 = v0 * ci + v1 * cr;
+            if (is_q) { src[i0] = o0; src[i1] = o1; }
+            else {
+                float *dst = kc + (int64_t)p * kv_dim0;
+                dst[i0] = o0; dst[i1] = o1;
+            }
+        } else {
+            const int j = (i - qp - kp) * 4;
+            *reinterpret_cast<float4 *>(vc + (int64_t)p * kv_dim0 + j) =
+                *reinterpret_cast<const float4 *>(vrow + j);
+        }
+    }
+}
+
 // ------------------------------------------------------------------ kv append
 // copy k,v batch rows into the caches at row pos[0]+b (reference OP_SHIFT,
 // nn-cpu-ops.cpp:1419-1441).
@@ -329,7 +498,7 @@ __global__ void k_kv_append(const float *__restrict__ k,
 // materialized att buffer, multiheadAtt_F32 nn-cpu-ops.cpp:753-788).
 // VEC = head_dim / 64 elements per lane.
 template <int VEC>
-__global__ void k_attn(const float *__restrict__ q,
+__global__ void k_attn(const float *__restrict__ q, int q_ld,
                        const float *__restrict__ kc,
                        const float *__restrict__ vc,
                        float *__restrict__ y,
@@ -346,7 +515,7 @@ __global__ void k_attn(const float *__restrict__ q,
     float qreg[VEC];
     #pragma unroll
     for (int v = 0; v < VEC; v++)
-        qreg[v] = q[((int64_t)b * n_heads0 + h0) * hd + lane * VEC + v] * scale;
+        qreg[v] = q[(int64_t)b * q_ld + h0 * hd + lane * VEC + v] * scale;
 
     float m = -1e30f, l = 0.0f, o[VEC];
     #pragma unroll
@@ -395,6 +564,7 @@ __global__ void k_attn(const float *__restrict__ q,
 // OP_CAST fused; silu nn-cpu-ops.cpp:462-500).
 __global__ void k_swiglu_q80(const float *__restrict__ a,
                              const float *__restrict__ g,
+                             int lda, int n,
                              int8_t *__restrict__ q,
                              float *__restrict__ s,
                              float *__restrict__ bs,
@@ -402,8 +572,11 @@ __global__ void k_swiglu_q80(const float *__restrict__ a,
     int gid = (blockIdx.x * blockDim.x + threadIdx.x) / 32;
     int lane = threadIdx.x & 31;
     if (gid >= n_blocks_total) return;
-    float av = a[gid * QB + lane];
-    float gv = g[gid * QB + lane];
+    const int nb = n / QB;
+    const int r = gid / nb;
+    const int i = (gid % nb) * QB + lane;
+    float av = a[(int64_t)r * lda + i];
+    float gv = g[(int64_t)r * lda + i];
     float v = av / (1.0f + __expf(-av)) * gv;
     float amax = group32_reduce_max(fabsf(v));
     float d = amax / 127.0f;
@@ -531,11 +704,16 @@ void rmsnorm_rows(torch::Tensor x, torch::Tensor w, torch::Tensor y, double eps)
 }
 
 void q40_gemv(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
-              torch::Tensor xs, torch::Tensor xbs, torch::Tensor y, int64_t batch) {
+              torch::Tensor xs, torch::Tensor xbs, torch::Tensor y, int64_t batch,
+              c10::optional<torch::Tensor> amax_slot = c10::nullopt) {
     CHECK_CUDA(qs); CHECK_CONT(qs); CHECK_CONT(xq);
     const int d = qs.size(0);
     const int n = qs.size(1) * 2;
     TORCH_CHECK(xq.size(-1) == n, "x width mismatch");
+    TORCH_CHECK(n % 64 == 0, "n must be a multiple of 64");
+    unsigned long long *slot = nullptr;
+    if (amax_slot.has_value())
+        slot = reinterpret_cast<unsigned long long *>(amax_slot->data_ptr<int64_t>());
     const int waves_per_block = 4;
     const dim3 grid(ceil_div(d, waves_per_block));
     const dim3 block(waves_per_block * WAVE);
@@ -545,7 +723,7 @@ void q40_gemv(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
                            qs.data_ptr<uint8_t>(),
                            reinterpret_cast<const __half *>(scales.data_ptr<at::Half>()),
                            xq.data_ptr<int8_t>(), xs.data_ptr<float>(),
-                           xbs.data_ptr<float>(), y.data_ptr<float>(), d, n);
+                           xbs.data_ptr<float>(), y.data_ptr<float>(), d, n, slot);
     };
     switch (batch) {
         case 1: launch(std::integral_constant<int, 1>{}); break;
@@ -603,35 +781,107 @@ void kv_append(torch::Tensor k, torch::Tensor v, torch::Tensor kc,
                        pos.data_ptr<int>(), kv_dim0);
 }
 
-void attn(torch::Tensor q, torch::Tensor kc, torch::Tensor vc, torch::Tensor y,
-          torch::Tensor pos, int64_t batch, int64_t n_heads0, int64_t kv_mul,
-          int64_t head_dim) {
+void attn(torch::Tensor q, int64_t q_ld, torch::Tensor kc, torch::Tensor vc,
+          torch::Tensor y, torch::Tensor pos, int64_t batch, int64_t n_heads0,
+          int64_t kv_mul, int64_t head_dim) {
     CHECK_CUDA(q);
     const int kv_dim0 = kc.size(1);
     const float scale = 1.0f / sqrtf((float)head_dim);
     const dim3 grid(n_heads0, batch);
     if (head_dim == 128)
         hipLaunchKernelGGL(k_attn<2>, grid, dim3(256), 0, cur_stream(),
-                           q.data_ptr<float>(), kc.data_ptr<float>(), vc.data_ptr<float>(),
+                           q.data_ptr<float>(), (int)q_ld, kc.data_ptr<float>(),
+                           vc.data_ptr<float>(),
                            y.data_ptr<float>(), pos.data_ptr<int>(),
                            (int)n_heads0, (int)kv_mul, kv_dim0, scale);
     else if (head_dim == 64)
         hipLaunchKernelGGL(k_attn<1>, grid, dim3(256), 0, cur_stream(),
-                           q.data_ptr<float>(), kc.data_ptr<float>(), vc.data_ptr<float>(),
+                           q.data_ptr<float>(), (int)q_ld, kc.data_ptr<float>(),
+                           vc.data_ptr<float>(),
                            y.data_ptr<float>(), pos.data_ptr<int>(),
                            (int)n_heads0, (int)kv_mul, kv_dim0, scale);
     else
         TORCH_CHECK(false, "unsupported head_dim ", head_dim);
 }
 
-void swiglu_q80(torch::Tensor a, torch::Tensor g, torch::Tensor q,
-                torch::Tensor s, torch::Tensor bs) {
+void swiglu_q80(torch::Tensor a, torch::Tensor g, int64_t lda, int64_t n,
+                int64_t rows, torch::Tensor q, torch::Tensor s, torch::Tensor bs) {
     CHECK_CUDA(a);
-    const int64_t blocks = a.numel() / QB;
+    const int64_t blocks = rows * (n / QB);
     hipLaunchKernelGGL(k_swiglu_q80, dim3(ceil_div(blocks * 32, 256)), dim3(256), 0,
                        cur_stream(), a.data_ptr<float>(), g.data_ptr<float>(),
+                       (int)lda, (int)n,
                        q.data_ptr<int8_t>(), s.data_ptr<float>(), bs.data_ptr<float>(),
                        (int)blocks);
+}
+
+void rmsnorm_rows_s(torch::Tensor buf, int64_t ld, int64_t off, int64_t heads,
+                    int64_t batch, torch::Tensor w, int64_t hd, double eps) {
+    CHECK_CUDA(buf);
+    hipLaunchKernelGGL(k_rmsnorm_rows_s, dim3(batch * heads), dim3(64), 0,
+                       cur_stream(), buf.data_ptr<float>(), (int)ld, (int)off,
+                       (int)heads, w.data_ptr<float>(), (int)hd, (float)eps);
+}
+
+void add_rmsnorm(torch::Tensor x, c10::optional<torch::Tensor> partial,
+                 torch::Tensor w, torch::Tensor y, double eps) {
+    CHECK_CUDA(x);
+    const int n = x.size(-1);
+    const int rows = x.numel() / n;
+    const bool add = partial.has_value();
+    const float *pp = add ? partial->data_ptr<float>() : nullptr;
+    if (add)
+        hipLaunchKernelGGL((k_add_rmsnorm<true, false>), dim3(rows), dim3(1024), 0,
+                           cur_stream(), x.data_ptr<float>(), pp, w.data_ptr<float>(),
+                           y.data_ptr<float>(), nullptr, nullptr, nullptr, n, (float)eps);
+    else
+        hipLaunchKernelGGL((k_add_rmsnorm<false, false>), dim3(rows), dim3(1024), 0,
+                           cur_stream(), x.data_ptr<float>(), pp, w.data_ptr<float>(),
+                           y.data_ptr<float>(), nullptr, nullptr, nullptr, n, (float)eps);
+}
+
+void add_rmsnorm_q80(torch::Tensor x, c10::optional<torch::Tensor> partial,
+                     torch::Tensor w, torch::Tensor q, torch::Tensor s,
+                     torch::Tensor bs, double eps) {
+    CHECK_CUDA(x);
+    const int n = x.size(-1);
+    const int rows = x.numel() / n;
+    const bool add = partial.has_value();
+    const float *pp = add ? partial->data_ptr<float>() : nullptr;
+    if (add)
+        hipLaunchKernelGGL((k_add_rmsnorm<true, true>), dim3(rows), dim3(1024), 0,
+                           cur_stream(), x.data_ptr<float>(), pp, w.data_ptr<float>(),
+                           nullptr, q.data_ptr<int8_t>(), s.data_ptr<float>(),
+                           bs.data_ptr<float>(), n, (float)eps);
+    else
+        hipLaunchKernelGGL((k_add_rmsnorm<false, true>), dim3(rows), dim3(1024), 0,
+                           cur_stream(), x.data_ptr<float>(), pp, w.data_ptr<float>(),
+                           nullptr, q.data_ptr<int8_t>(), s.data_ptr<float>(),
+                           bs.data_ptr<float>(), n, (float)eps);
+}
+
+void rope_kv(torch::Tensor qkv, int64_t ld, int64_t q_dim0, int64_t kv_dim0,
+             torch::Tensor cache, torch::Tensor pos, torch::Tensor kc,
+             torch::Tensor vc, int64_t head_dim, int64_t style, int64_t batch) {
+    CHECK_CUDA(qkv);
+    const int total = (int)(q_dim0 / 2 + kv_dim0 / 2 + kv_dim0 / 4);
+    const dim3 grid(ceil_div(total, 256), batch);
+    if (style == 0)
+        hipLaunchKernelGGL(k_rope_kv<0>, grid, dim3(256), 0, cur_stream(),
+                           qkv.data_ptr<float>(), (int)ld, (int)q_dim0, (int)kv_dim0,
+                           cache.data_ptr<float>(), pos.data_ptr<int>(),
+                           kc.data_ptr<float>(), vc.data_ptr<float>(), (int)head_dim);
+    else
+        hipLaunchKernelGGL(k_rope_kv<1>, grid, dim3(256), 0, cur_stream(),
+                           qkv.data_ptr<float>(), (int)ld, (int)q_dim0, (int)kv_dim0,
+                           cache.data_ptr<float>(), pos.data_ptr<int>(),
+                           kc.data_ptr<float>(), vc.data_ptr<float>(), (int)head_dim);
+}
+
+void token_from_argmax(torch::Tensor token, torch::Tensor slot) {
+    hipLaunchKernelGGL(k_token_from_argmax, dim3(1), dim3(64), 0, cur_stream(),
+                       (long *)token.data_ptr<int64_t>(),
+                       reinterpret_cast<unsigned long long *>(slot.data_ptr<int64_t>()));
 }
 
 void silu_mul(torch::Tensor a, torch::Tensor g, torch::Tensor out) {
@@ -676,9 +926,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rmsnorm", &rmsnorm);
     m.def("rmsnorm_q80", &rmsnorm_q80);
     m.def("rmsnorm_rows", &rmsnorm_rows);
-    m.def("q40_gemv", &q40_gemv);
+    m.def("rmsnorm_rows_s", &rmsnorm_rows_s);
+    m.def("add_rmsnorm", &add_rmsnorm);
+    m.def("add_rmsnorm_q80", &add_rmsnorm_q80);
+    m.def("q40_gemv", &q40_gemv, py::arg("qs"), py::arg("scales"), py::arg("xq"),
+          py::arg("xs"), py::arg("xbs"), py::arg("y"), py::arg("batch"),
+          py::arg("amax_slot") = py::none());
     m.def("q40_gemv_grouped", &q40_gemv_grouped);
     m.def("rope", &rope);
+    m.def("rope_kv", &rope_kv);
     m.def("kv_append", &kv_append);
     m.def("attn", &attn);
     m.def("swiglu_q80", &swiglu_q80);
@@ -687,4 +943,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("merge_add", &merge_add);
     m.def("add_", &add_);
     m.def("pos_inc", &pos_inc);
+    m.def("token_from_argmax", &token_from_argmax);
 }

@@ -151,7 +151,7 @@ def test_kv_append_and_attn(k):
     q = rand(B, H0 * hd, seed=62)
     y = torch.zeros(B, H0 * hd, device=DEV)
     pos = torch.tensor([4], dtype=torch.int32, device=DEV)
-    k.attn(q, kc, vc, y, pos, B, H0, H0 // n_kv0, hd)
+    k.attn(q, H0 * hd, kc, vc, y, pos, B, H0, H0 // n_kv0, hd)
     want = R.attention(q.cpu(), kc.cpu(), vc.cpu(), torch.tensor([4, 5]), H0, hd)
     assert torch.allclose(y.cpu(), want, atol=1e-4, rtol=1e-3), \
         (y.cpu() - want).abs().max().item()
@@ -166,7 +166,7 @@ def test_attn_long_context(k):
     q = rand(B, H0 * hd, seed=72)
     y = torch.zeros(B, H0 * hd, device=DEV)
     pos = torch.tensor([298], dtype=torch.int32, device=DEV)
-    k.attn(q, kc, vc, y, pos, B, H0, H0 // n_kv0, hd)
+    k.attn(q, H0 * hd, kc, vc, y, pos, B, H0, H0 // n_kv0, hd)
     want = R.attention(q.cpu(), kc.cpu(), vc.cpu(), torch.tensor([298]), H0, hd)
     assert torch.allclose(y.cpu(), want, atol=1e-4, rtol=1e-3)
 
@@ -177,10 +177,82 @@ def test_swiglu_q80(k):
     q = torch.zeros(2, 512, dtype=torch.int8, device=DEV)
     s = torch.zeros(2, 16, device=DEV)
     bs = torch.zeros(2, 16, device=DEV)
-    k.swiglu_q80(a, g, q, s, bs)
+    k.swiglu_q80(a, g, 512, 512, 2, q, s, bs)
     want = R.swiglu(a.cpu(), g.cpu())
     got = R.q80_dequantize(q.cpu(), s.cpu())
     assert torch.allclose(got, want, atol=want.abs().max().item() / 100)
+
+
+def test_swiglu_q80_fused_layout(k):
+    # a|g packed in one [rows, 2n] buffer (the fused W1|W3 GEMV output)
+    rows, n = 3, 256
+    buf = rand(rows, 2 * n, seed=82)
+    q = torch.zeros(rows, n, dtype=torch.int8, device=DEV)
+    s = torch.zeros(rows, n // 32, device=DEV)
+    bs = torch.zeros(rows, n // 32, device=DEV)
+    k.swiglu_q80(buf, buf[:, n:], 2 * n, n, rows, q, s, bs)
+    want = R.swiglu(buf[:, :n].cpu(), buf[:, n:].cpu())
+    got = R.q80_dequantize(q.cpu(), s.cpu())
+    assert torch.allclose(got, want, atol=want.abs().max().item() / 100)
+
+
+def test_add_rmsnorm_q80(k):
+    x = rand(2, 4096, seed=83)
+    p = rand(2, 4096, seed=84)
+    w = rand(4096, seed=85).abs()
+    q = torch.zeros(2, 4096, dtype=torch.int8, device=DEV)
+    s = torch.zeros(2, 128, device=DEV)
+    bs = torch.zeros(2, 128, device=DEV)
+    xs = x.clone()
+    k.add_rmsnorm_q80(xs, p, w, q, s, bs, 1e-5)
+    want_x = (x + p).cpu()
+    assert torch.allclose(xs.cpu(), want_x, atol=1e-6)
+    want = R.rms_norm(want_x, w.cpu(), 1e-5)
+    got = R.q80_dequantize(q.cpu(), s.cpu())
+    assert torch.allclose(got, want, atol=want.abs().max().item() / 100)
+    # no-partial variant, f32 out
+    y = torch.zeros(2, 4096, device=DEV)
+    k.add_rmsnorm(xs, None, w, y, 1e-5)
+    want2 = R.rms_norm(want_x, w.cpu(), 1e-5)
+    assert torch.allclose(y.cpu(), want2, atol=1e-4, rtol=1e-4)
+
+
+def test_rope_kv_fused(k):
+    B, hd, qh, kvh, seq = 2, 64, 4, 2, 32
+    q_dim0, kv_dim0 = qh * hd, kvh * hd
+    ld = q_dim0 + 2 * kv_dim0
+    buf = rand(B, ld, seed=86)
+    orig = buf.clone()
+    cache = R.rope_cache(seq, hd, 10000.0).to(DEV).reshape(seq, hd).contiguous()
+    kc = torch.zeros(seq, kv_dim0, device=DEV)
+    vc = torch.zeros(seq, kv_dim0, device=DEV)
+    pos = torch.tensor([3], dtype=torch.int32, device=DEV)
+    k.rope_kv(buf, ld, q_dim0, kv_dim0, cache, pos, kc, vc, hd, 0, B)
+    cache_cpu = R.rope_cache(seq, hd, 10000.0)
+    positions = torch.arange(3, 3 + B)
+    want_q = R.rope_llama(orig[:, :q_dim0].cpu(), cache_cpu, positions, hd)
+    want_k = R.rope_llama(orig[:, q_dim0:q_dim0 + kv_dim0].cpu(), cache_cpu,
+                          positions, hd)
+    assert torch.allclose(buf[:, :q_dim0].cpu(), want_q, atol=1e-5)
+    assert torch.allclose(kc[3:5].cpu(), want_k, atol=1e-5)
+    assert torch.allclose(vc[3:5].cpu(), orig[:, q_dim0 + kv_dim0:].cpu())
+
+
+def test_gemv_argmax(k):
+    d, n = 512, 256
+    qs, sc, wref = _mk_linear(d, n, 44)
+    x = rand(1, n, seed=45, scale=0.5)
+    q = torch.zeros(1, n, dtype=torch.int8, device=DEV)
+    s = torch.zeros(1, n // 32, device=DEV)
+    bs = torch.zeros(1, n // 32, device=DEV)
+    k.q80_quantize(x, q, s, bs)
+    y = torch.zeros(1, d, device=DEV)
+    slot = torch.zeros(1, dtype=torch.int64, device=DEV)
+    k.q40_gemv(qs, sc, q, s, bs, y, 1, slot)
+    tok = torch.zeros(1, dtype=torch.int64, device=DEV)
+    k.token_from_argmax(tok, slot)
+    assert int(tok.item()) == int(y[0].argmax().item())
+    assert int(slot.item()) == 0  # reset for next step
 
 
 def test_sync_pack_merge_add(k):
@@ -203,8 +275,11 @@ def test_sync_pack_merge_add(k):
     k.merge_add(got, gathered)
     want = x.cpu().clone()
     for p in partials:
-        want += R.q80_roundtrip(p.cpu())
-    assert torch.allclose(got.cpu(), want, atol=1e-3)
+        # wire scales are f16 (reference Q80 block format) — model that
+        q, s, _ = R.q80_quantize(p.cpu())
+        want += R.q80_dequantize(q, s.to(torch.float16).float())
+    assert torch.allclose(got.cpu(), want, atol=1e-3), \
+        (got.cpu() - want).abs().max().item()
 
 
 def test_add_and_pos_inc(k):
