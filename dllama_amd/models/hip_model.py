@@ -204,7 +204,12 @@ class HipTransformer:
         self.ff_out = torch.zeros(NB, 2 * c.ff_dim0, device=dev)
         self.dq = QuantBuf(NB, c.ff_dim0, dev)
         self.logits0 = torch.zeros(NB, c.vocab0, device=dev)
-        self.amax_slot = torch.zeros(1, dtype=torch.int64, device=dev)
+        self.amax_blocks = (c.vocab0 + 7) // 8  # waves_per_block(4) * RPW(2)
+        self.amax_scratch = torch.zeros(self.amax_blocks, dtype=torch.int64, device=dev)
+        self.attn_splits = 8
+        self.attn_ml = torch.zeros(NB * c.n_heads0 * self.attn_splits * 2, device=dev)
+        self.attn_o = torch.zeros(NB * c.n_heads0 * self.attn_splits * c.head_dim,
+                                  device=dev)
         if c.world > 1:
             self.logits_gather = torch.zeros(c.world, NB, c.vocab0, device=dev)
             if c.sync_type == Q80:
@@ -289,7 +294,8 @@ class HipTransformer:
                       self.rope_cache, self.pos, self.k_cache[l], self.v_cache[l],
                       c.head_dim, self.rope_style, B)
             k.attn(self.qkv_out, self.qkv_ld, self.k_cache[l], self.v_cache[l],
-                   self.zbuf[:B], self.pos, B, c.n_heads0, kv_mul, c.head_dim)
+                   self.zbuf[:B], self.pos, B, c.n_heads0, kv_mul, c.head_dim,
+                   self.attn_splits, self.attn_ml, self.attn_o)
             k.q80_quantize(self.zbuf[:NB], self.zq.q[:NB], self.zq.s[:NB],
                            self.zq.bs[:NB])
             k.q40_gemv(lw["wo"].qs, lw["wo"].scales, self.zq.q, self.zq.s,
@@ -316,14 +322,14 @@ class HipTransformer:
         use_amax = (self.greedy_feedback and B == 1 and c.world == 1)
         k.q40_gemv(self.wcls.qs, self.wcls.scales, self.xq.q, self.xq.s,
                    self.xq.bs, self.logits0, NB,
-                   self.amax_slot if use_amax else None)
+                   self.amax_scratch if use_amax else None)
         if c.world > 1:
             self.comm.all_gather(self.logits_gather[:, :NB], self.logits0[:NB])
         if self.greedy_feedback and B == 1:
             # on-device greedy sampling feeding the next decode step (used by
             # the fully graph-captured bench loop; real serving samples on host)
             if use_amax:
-                k.token_from_argmax(self.tokens, self.amax_slot)
+                k.token_from_argmax(self.tokens, self.amax_scratch, self.amax_blocks)
             else:
                 full = self.logits_gather[:, 0].reshape(-1)
                 self.tokens[0].copy_(torch.argmax(full))

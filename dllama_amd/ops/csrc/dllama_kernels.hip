@@ -264,11 +264,26 @@ __device__ __forceinline__ unsigned long long argmax_pack(float v, int idx) {
     return ((unsigned long long)u << 32) | (unsigned)(0x7FFFFFFF - idx);
 }
 
+// stage 2 of the on-device greedy argmax: reduce the per-workgroup bests the
+// logits GEMV wrote to scratch (no atomics anywhere — a single-slot atomicMax
+// across 32k workgroups serializes on one cacheline and costs >1 ms).
 __global__ void k_token_from_argmax(long *__restrict__ token,
-                                    unsigned long long *__restrict__ slot) {
-    if (threadIdx.x == 0 && blockIdx.x == 0) {
-        token[0] = (long)(0x7FFFFFFF - (int)(slot[0] & 0xFFFFFFFFu));
-        slot[0] = 0ull;  // reset for the next decode step
+                                    const unsigned long long *__restrict__ scratch,
+                                    int count) {
+    unsigned long long best = 0ull;
+    for (int i = threadIdx.x; i < count; i += blockDim.x)
+        best = max(best, scratch[i]);
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        best = max(best, (unsigned long long)__shfl_xor((long long)best, off, WAVE));
+    __shared__ unsigned long long red[16];
+    const int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    if (lane == 0) red[wid] = best;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        for (int i = 1; i < blockDim.x / WAVE; i++) best = max(best, red[i]);
+        best = max(best, red[0]);
+        token[0] = (long)(0x7FFFFFFF - (int)(best & 0xFFFFFFFFu));
     }
 }
 
@@ -294,7 +309,10 @@ __device__ __forceinline__ int q40_block_dot(const uint4 wq, const int4 x0,
     return idot;
 }
 
-template <int NB>
+// RPW = rows per wave: processing 2 rows per wave doubles the independent
+// 16B weight loads in flight per lane (the decode GEMV is HBM-latency
+// limited at 1 row/wave). NB = batch columns.
+template <int NB, int RPW>
 __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
                            const __half *__restrict__ scales,
                            const int8_t *__restrict__ xq,
@@ -302,44 +320,193 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
                            const float *__restrict__ xbs,
                            float *__restrict__ y,
                            int d, int n,
-                           unsigned long long *__restrict__ amax_slot) {
-    const int row = blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
-    if (row >= d) return;
+                           unsigned long long *__restrict__ amax_scratch) {
+    const int wpb = blockDim.x / WAVE;
+    const int wid = threadIdx.x / WAVE;
+    const int row0 = (blockIdx.x * wpb + wid) * RPW;
     const int lane = threadIdx.x % WAVE;
     const int nb = n / QB;
-    const int nbp = nb >> 1;  // block pairs (n % 64 == 0 asserted host-side)
-    const uint4 *wrow = reinterpret_cast<const uint4 *>(qs + (int64_t)row * (n >> 1));
-    const __half *srow = scales + (int64_t)row * nb;
+    const int nbp = nb >> 1;
+    const bool wave_valid = row0 < d;
+    const int rbase = wave_valid ? row0 : 0;
 
-    float acc[NB];
+    const uint4 *wrow[RPW];
+    const __half *srow[RPW];
     #pragma unroll
-    for (int b = 0; b < NB; b++) acc[b] = 0.0f;
+    for (int r = 0; r < RPW; r++) {
+        const int row = min(rbase + r, d - 1);
+        wrow[r] = reinterpret_cast<const uint4 *>(qs + (int64_t)row * (n >> 1));
+        srow[r] = scales + (int64_t)row * nb;
+    }
+
+    float acc[RPW][NB];
+    #pragma unroll
+    for (int r = 0; r < RPW; r++)
+        #pragma unroll
+        for (int b = 0; b < NB; b++) acc[r][b] = 0.0f;
 
     for (int jp = lane; jp < nbp; jp += WAVE) {
         const int j = jp << 1;
-        const uint4 wq0 = wrow[j];
-        const uint4 wq1 = wrow[j + 1];
-        const float2 sw = __half22float2(
-            *reinterpret_cast<const __half2 *>(srow + j));
+        uint4 wq0[RPW], wq1[RPW];
+        float2 sw[RPW];
+        #pragma unroll
+        for (int r = 0; r < RPW; r++) {
+            wq0[r] = wrow[r][j];
+            wq1[r] = wrow[r][j + 1];
+            sw[r] = __half22float2(*reinterpret_cast<const __half2 *>(srow[r] + j));
+        }
         #pragma unroll
         for (int b = 0; b < NB; b++) {
             const int4 *xrow = reinterpret_cast<const int4 *>(xq + (int64_t)b * n) + j * 2;
-            const int idot0 = q40_block_dot(wq0, xrow[0], xrow[1]);
-            const int idot1 = q40_block_dot(wq1, xrow[2], xrow[3]);
+            const int4 x0 = xrow[0], x1 = xrow[1], x2 = xrow[2], x3 = xrow[3];
             const float2 sx = *reinterpret_cast<const float2 *>(xs + (int64_t)b * nb + j);
             const float2 bsum = *reinterpret_cast<const float2 *>(xbs + (int64_t)b * nb + j);
-            acc[b] = fmaf(sw.x * sx.x, (float)idot0 - 8.0f * bsum.x, acc[b]);
-            acc[b] = fmaf(sw.y * sx.y, (float)idot1 - 8.0f * bsum.y, acc[b]);
+            #pragma unroll
+            for (int r = 0; r < RPW; r++) {
+                const int idot0 = q40_block_dot(wq0[r], x0, x1);
+                const int idot1 = q40_block_dot(wq1[r], x2, x3);
+                acc[r][b] = fmaf(sw[r].x * sx.x, (float)idot0 - 8.0f * bsum.x, acc[r][b]);
+                acc[r][b] = fmaf(sw[r].y * sx.y, (float)idot1 - 8.0f * bsum.y, acc[r][b]);
+            }
         }
     }
-    #pragma unroll
-    for (int b = 0; b < NB; b++) {
-        float r = wave_reduce_sum(acc[b]);
-        if (lane == 0) {
-            y[(int64_t)b * d + row] = r;
-            if (NB == 1 && amax_slot != nullptr)
-                atomicMax(amax_slot, argmax_pack(r, row));
+    if ((nb & 1) && lane == 0) {  // odd trailing block
+        const int j = nb - 1;
+        #pragma unroll
+        for (int r = 0; r < RPW; r++) {
+            const uint4 wq = wrow[r][j];
+            const float sw1 = __half2float(srow[r][j]);
+            #pragma unroll
+            for (int b = 0; b < NB; b++) {
+                const int4 *xb = reinterpret_cast<const int4 *>(xq + (int64_t)b * n) + j * 2;
+                const int idot = q40_block_dot(wq, xb[0], xb[1]);
+                acc[r][b] = fmaf(sw1 * xs[(int64_t)b * nb + j],
+                                 (float)idot - 8.0f * xbs[(int64_t)b * nb + j], acc[r][b]);
+            }
         }
+    }
+
+    unsigned long long wave_best = 0ull;
+    #pragma unroll
+    for (int r = 0; r < RPW; r++) {
+        const int row = rbase + r;
+        const bool ok = wave_valid && row < d;
+        #pragma unroll
+        for (int b = 0; b < NB; b++) {
+            const float v = wave_reduce_sum(acc[r][b]);
+            if (lane == 0 && ok) {
+                y[(int64_t)b * d + row] = v;
+                if (NB == 1) wave_best = max(wave_best, argmax_pack(v, row));
+            }
+        }
+    }
+    if (NB == 1 && amax_scratch != nullptr) {
+        // stage 1 of the greedy argmax: one packed best per workgroup
+        __shared__ unsigned long long wb[16];
+        if (lane == 0) wb[wid] = wave_best;
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            unsigned long long best = wb[0];
+            for (int i = 1; i < wpb; i++) best = max(best, wb[i]);
+            amax_scratch[blockIdx.x] = best;
+        }
+    }
+}
+
+// ------------------------------------------------- split-K flash decode
+// Stage 1: grid (H0, B, S); split sp covers t = (sp*4+wave) + 4*S*i — the
+// t-range is spread over S*4 waves so long contexts fill the chip.
+// Per-split (m, l, o) goes to scratch; stage 2 combines the S partials.
+template <int VEC>
+__global__ void k_attn_split(const float *__restrict__ q, int q_ld,
+                             const float *__restrict__ kc,
+                             const float *__restrict__ vc,
+                             const int *__restrict__ pos,
+                             int n_heads0, int kv_mul, int kv_dim0, float scale,
+                             float *__restrict__ ml_scratch,
+                             float *__restrict__ o_scratch) {
+    const int h0 = blockIdx.x;
+    const int b = blockIdx.y;
+    const int sp = blockIdx.z;
+    const int S = gridDim.z;
+    const int hd = VEC * WAVE;
+    const int plen = pos[0] + b + 1;
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int kv_off = (h0 / kv_mul) * hd;
+
+    float qreg[VEC];
+    #pragma unroll
+    for (int v = 0; v < VEC; v++)
+        qreg[v] = q[(int64_t)b * q_ld + h0 * hd + lane * VEC + v] * scale;
+
+    float m = -1e30f, l = 0.0f, o[VEC];
+    #pragma unroll
+    for (int v = 0; v < VEC; v++) o[v] = 0.0f;
+
+    for (int t = sp * 4 + wave; t < plen; t += 4 * S) {
+        const float *krow = kc + (int64_t)t * kv_dim0 + kv_off + lane * VEC;
+        float partial = 0.0f;
+        #pragma unroll
+        for (int v = 0; v < VEC; v++) partial = fmaf(qreg[v], krow[v], partial);
+        const float s = wave_reduce_sum(partial);
+        const float mn = fmaxf(m, s);
+        const float f = __expf(m - mn);
+        const float w = __expf(s - mn);
+        const float *vrow = vc + (int64_t)t * kv_dim0 + kv_off + lane * VEC;
+        l = l * f + w;
+        #pragma unroll
+        for (int v = 0; v < VEC; v++) o[v] = fmaf(o[v], f, w * vrow[v]);
+        m = mn;
+    }
+
+    __shared__ float sm[4], sl[4];
+    __shared__ float so[4][VEC * WAVE];
+    if (lane == 0) { sm[wave] = m; sl[wave] = l; }
+    __syncthreads();
+    const float M = fmaxf(fmaxf(sm[0], sm[1]), fmaxf(sm[2], sm[3]));
+    const float fw = __expf(m - M);
+    #pragma unroll
+    for (int v = 0; v < VEC; v++) so[wave][lane * VEC + v] = o[v] * fw;
+    __syncthreads();
+    const int64_t slot = ((int64_t)b * n_heads0 + h0) * S + sp;
+    if (wave == 0) {
+        const float L = sl[0] * __expf(sm[0] - M) + sl[1] * __expf(sm[1] - M)
+                      + sl[2] * __expf(sm[2] - M) + sl[3] * __expf(sm[3] - M);
+        if (lane == 0) {
+            ml_scratch[slot * 2] = M;
+            ml_scratch[slot * 2 + 1] = L;
+        }
+        #pragma unroll
+        for (int v = 0; v < VEC; v++) {
+            const int i = lane * VEC + v;
+            o_scratch[slot * hd + i] = so[0][i] + so[1][i] + so[2][i] + so[3][i];
+        }
+    }
+}
+
+template <int VEC>
+__global__ void k_attn_combine(const float *__restrict__ ml_scratch,
+                               const float *__restrict__ o_scratch,
+                               float *__restrict__ y,
+                               int n_heads0, int S) {
+    const int h0 = blockIdx.x;
+    const int b = blockIdx.y;
+    const int hd = VEC * WAVE;
+    const int64_t base = ((int64_t)b * n_heads0 + h0) * S;
+    float M = -1e30f;
+    for (int sp = 0; sp < S; sp++)
+        M = fmaxf(M, ml_scratch[(base + sp) * 2]);
+    float L = 0.0f;
+    for (int sp = 0; sp < S; sp++)
+        L += ml_scratch[(base + sp) * 2 + 1] * __expf(ml_scratch[(base + sp) * 2] - M);
+    const float invL = 1.0f / L;
+    for (int i = threadIdx.x; i < hd; i += blockDim.x) {
+        float acc = 0.0f;
+        for (int sp = 0; sp < S; sp++)
+            acc += o_scratch[(base + sp) * hd + i]
+                 * __expf(ml_scratch[(base + sp) * 2] - M);
+        y[((int64_t)b * n_heads0 + h0) * hd + i] = acc * invL;
     }
 }
 
@@ -488,74 +655,6 @@ __global__ void k_kv_append(const float *__restrict__ k,
          i += gridDim.x * blockDim.x) {
         kc[(int64_t)p * kv_dim0 + i] = k[(int64_t)b * kv_dim0 + i];
         vc[(int64_t)p * kv_dim0 + i] = v[(int64_t)b * kv_dim0 + i];
-    }
-}
-
-// ------------------------------------------------------------------ attention
-// Flash-style decode/prefill attention over the f32 KV cache.
-// One workgroup (4 waves) per (head, batch); waves split the t range with
-// online softmax, combined through LDS (replaces the reference's
-// materialized att buffer, multiheadAtt_F32 nn-cpu-ops.cpp:753-788).
-// VEC = head_dim / 64 elements per lane.
-template <int VEC>
-__global__ void k_attn(const float *__restrict__ q, int q_ld,
-                       const float *__restrict__ kc,
-                       const float *__restrict__ vc,
-                       float *__restrict__ y,
-                       const int *__restrict__ pos,
-                       int n_heads0, int kv_mul, int kv_dim0, float scale) {
-    const int h0 = blockIdx.x;
-    const int b = blockIdx.y;
-    const int hd = VEC * WAVE;
-    const int plen = pos[0] + b + 1;  // row b attends to cache rows 0..pos+b
-    const int wave = threadIdx.x / WAVE;
-    const int lane = threadIdx.x % WAVE;
-    const int kv_off = (h0 / kv_mul) * hd;
-
-    float qreg[VEC];
-    #pragma unroll
-    for (int v = 0; v < VEC; v++)
-        qreg[v] = q[(int64_t)b * q_ld + h0 * hd + lane * VEC + v] * scale;
-
-    float m = -1e30f, l = 0.0f, o[VEC];
-    #pragma unroll
-    for (int v = 0; v < VEC; v++) o[v] = 0.0f;
-
-    for (int t = wave; t < plen; t += 4) {
-        const float *krow = kc + (int64_t)t * kv_dim0 + kv_off + lane * VEC;
-        float partial = 0.0f;
-        #pragma unroll
-        for (int v = 0; v < VEC; v++) partial = fmaf(qreg[v], krow[v], partial);
-        const float s = wave_reduce_sum(partial);
-        const float mn = fmaxf(m, s);
-        const float f = __expf(m - mn);
-        const float w = __expf(s - mn);
-        const float *vrow = vc + (int64_t)t * kv_dim0 + kv_off + lane * VEC;
-        l = l * f + w;
-        #pragma unroll
-        for (int v = 0; v < VEC; v++) o[v] = fmaf(o[v], f, w * vrow[v]);
-        m = mn;
-    }
-
-    __shared__ float sm[4], sl[4];
-    __shared__ float so[4][VEC * WAVE];
-    if (lane == 0) { sm[wave] = m; sl[wave] = l; }
-    __syncthreads();
-    const float M = fmaxf(fmaxf(sm[0], sm[1]), fmaxf(sm[2], sm[3]));
-    const float fw = __expf(m - M);
-    #pragma unroll
-    for (int v = 0; v < VEC; v++) so[wave][lane * VEC + v] = o[v] * fw;
-    __syncthreads();
-    if (wave == 0) {
-        const float L = sl[0] * __expf(sm[0] - M) + sl[1] * __expf(sm[1] - M)
-                      + sl[2] * __expf(sm[2] - M) + sl[3] * __expf(sm[3] - M);
-        const float invL = 1.0f / L;
-        #pragma unroll
-        for (int v = 0; v < VEC; v++) {
-            const int i = lane * VEC + v;
-            y[((int64_t)b * n_heads0 + h0) * hd + i] =
-                (so[0][i] + so[1][i] + so[2][i] + so[3][i]) * invL;
-        }
     }
 }
 
@@ -710,30 +809,38 @@ void q40_gemv(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
     const int d = qs.size(0);
     const int n = qs.size(1) * 2;
     TORCH_CHECK(xq.size(-1) == n, "x width mismatch");
-    TORCH_CHECK(n % 64 == 0, "n must be a multiple of 64");
+    TORCH_CHECK(n % QB == 0, "n must be a multiple of 32");
     unsigned long long *slot = nullptr;
     if (amax_slot.has_value())
         slot = reinterpret_cast<unsigned long long *>(amax_slot->data_ptr<int64_t>());
     const int waves_per_block = 4;
-    const dim3 grid(ceil_div(d, waves_per_block));
     const dim3 block(waves_per_block * WAVE);
-    auto launch = [&](auto nb_const) {
-        hipLaunchKernelGGL((k_q40_gemv<decltype(nb_const)::value>), grid, block, 0,
-                           cur_stream(),
+    auto launch = [&](auto nb_const, auto rpw_const) {
+        constexpr int RPW = decltype(rpw_const)::value;
+        const dim3 grid(ceil_div(d, waves_per_block * RPW));
+        hipLaunchKernelGGL((k_q40_gemv<decltype(nb_const)::value, RPW>), grid, block,
+                           0, cur_stream(),
                            qs.data_ptr<uint8_t>(),
                            reinterpret_cast<const __half *>(scales.data_ptr<at::Half>()),
                            xq.data_ptr<int8_t>(), xs.data_ptr<float>(),
                            xbs.data_ptr<float>(), y.data_ptr<float>(), d, n, slot);
     };
+    std::integral_constant<int, 1> r1;
+    std::integral_constant<int, 2> r2;
     switch (batch) {
-        case 1: launch(std::integral_constant<int, 1>{}); break;
-        case 2: launch(std::integral_constant<int, 2>{}); break;
-        case 4: launch(std::integral_constant<int, 4>{}); break;
-        case 8: launch(std::integral_constant<int, 8>{}); break;
-        case 16: launch(std::integral_constant<int, 16>{}); break;
-        case 32: launch(std::integral_constant<int, 32>{}); break;
+        case 1: launch(std::integral_constant<int, 1>{}, r2); break;
+        case 2: launch(std::integral_constant<int, 2>{}, r2); break;
+        case 4: launch(std::integral_constant<int, 4>{}, r2); break;
+        case 8: launch(std::integral_constant<int, 8>{}, r1); break;
+        case 16: launch(std::integral_constant<int, 16>{}, r1); break;
+        case 32: launch(std::integral_constant<int, 32>{}, r1); break;
         default: TORCH_CHECK(false, "unsupported batch ", batch);
     }
+}
+
+int64_t q40_gemv_argmax_blocks(int64_t d) {
+    // number of stage-1 argmax scratch entries for a batch-1 GEMV over d rows
+    return ceil_div(d, 4 * 2);  // waves_per_block * RPW
 }
 
 void q40_gemv_grouped(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
@@ -783,25 +890,34 @@ void kv_append(torch::Tensor k, torch::Tensor v, torch::Tensor kc,
 
 void attn(torch::Tensor q, int64_t q_ld, torch::Tensor kc, torch::Tensor vc,
           torch::Tensor y, torch::Tensor pos, int64_t batch, int64_t n_heads0,
-          int64_t kv_mul, int64_t head_dim) {
+          int64_t kv_mul, int64_t head_dim, int64_t splits,
+          torch::Tensor ml_scratch, torch::Tensor o_scratch) {
     CHECK_CUDA(q);
     const int kv_dim0 = kc.size(1);
     const float scale = 1.0f / sqrtf((float)head_dim);
-    const dim3 grid(n_heads0, batch);
-    if (head_dim == 128)
-        hipLaunchKernelGGL(k_attn<2>, grid, dim3(256), 0, cur_stream(),
+    const dim3 grid(n_heads0, batch, splits);
+    const dim3 cgrid(n_heads0, batch);
+    if (head_dim == 128) {
+        hipLaunchKernelGGL(k_attn_split<2>, grid, dim3(256), 0, cur_stream(),
                            q.data_ptr<float>(), (int)q_ld, kc.data_ptr<float>(),
-                           vc.data_ptr<float>(),
-                           y.data_ptr<float>(), pos.data_ptr<int>(),
-                           (int)n_heads0, (int)kv_mul, kv_dim0, scale);
-    else if (head_dim == 64)
-        hipLaunchKernelGGL(k_attn<1>, grid, dim3(256), 0, cur_stream(),
+                           vc.data_ptr<float>(), pos.data_ptr<int>(),
+                           (int)n_heads0, (int)kv_mul, kv_dim0, scale,
+                           ml_scratch.data_ptr<float>(), o_scratch.data_ptr<float>());
+        hipLaunchKernelGGL(k_attn_combine<2>, cgrid, dim3(128), 0, cur_stream(),
+                           ml_scratch.data_ptr<float>(), o_scratch.data_ptr<float>(),
+                           y.data_ptr<float>(), (int)n_heads0, (int)splits);
+    } else if (head_dim == 64) {
+        hipLaunchKernelGGL(k_attn_split<1>, grid, dim3(256), 0, cur_stream(),
                            q.data_ptr<float>(), (int)q_ld, kc.data_ptr<float>(),
-                           vc.data_ptr<float>(),
-                           y.data_ptr<float>(), pos.data_ptr<int>(),
-                           (int)n_heads0, (int)kv_mul, kv_dim0, scale);
-    else
+                           vc.data_ptr<float>(), pos.data_ptr<int>(),
+                           (int)n_heads0, (int)kv_mul, kv_dim0, scale,
+                           ml_scratch.data_ptr<float>(), o_scratch.data_ptr<float>());
+        hipLaunchKernelGGL(k_attn_combine<1>, cgrid, dim3(64), 0, cur_stream(),
+                           ml_scratch.data_ptr<float>(), o_scratch.data_ptr<float>(),
+                           y.data_ptr<float>(), (int)n_heads0, (int)splits);
+    } else {
         TORCH_CHECK(false, "unsupported head_dim ", head_dim);
+    }
 }
 
 void swiglu_q80(torch::Tensor a, torch::Tensor g, int64_t lda, int64_t n,
@@ -878,10 +994,11 @@ void rope_kv(torch::Tensor qkv, int64_t ld, int64_t q_dim0, int64_t kv_dim0,
                            kc.data_ptr<float>(), vc.data_ptr<float>(), (int)head_dim);
 }
 
-void token_from_argmax(torch::Tensor token, torch::Tensor slot) {
-    hipLaunchKernelGGL(k_token_from_argmax, dim3(1), dim3(64), 0, cur_stream(),
+void token_from_argmax(torch::Tensor token, torch::Tensor scratch, int64_t count) {
+    hipLaunchKernelGGL(k_token_from_argmax, dim3(1), dim3(1024), 0, cur_stream(),
                        (long *)token.data_ptr<int64_t>(),
-                       reinterpret_cast<unsigned long long *>(slot.data_ptr<int64_t>()));
+                       reinterpret_cast<unsigned long long *>(scratch.data_ptr<int64_t>()),
+                       (int)count);
 }
 
 void silu_mul(torch::Tensor a, torch::Tensor g, torch::Tensor out) {
@@ -944,4 +1061,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("add_", &add_);
     m.def("pos_inc", &pos_inc);
     m.def("token_from_argmax", &token_from_argmax);
+    m.def("q40_gemv_argmax_blocks", &q40_gemv_argmax_blocks);
 }

@@ -151,7 +151,10 @@ def test_kv_append_and_attn(k):
     q = rand(B, H0 * hd, seed=62)
     y = torch.zeros(B, H0 * hd, device=DEV)
     pos = torch.tensor([4], dtype=torch.int32, device=DEV)
-    k.attn(q, H0 * hd, kc, vc, y, pos, B, H0, H0 // n_kv0, hd)
+    S = 8
+    ml = torch.zeros(B * H0 * S * 2, device=DEV)
+    osc = torch.zeros(B * H0 * S * hd, device=DEV)
+    k.attn(q, H0 * hd, kc, vc, y, pos, B, H0, H0 // n_kv0, hd, S, ml, osc)
     want = R.attention(q.cpu(), kc.cpu(), vc.cpu(), torch.tensor([4, 5]), H0, hd)
     assert torch.allclose(y.cpu(), want, atol=1e-4, rtol=1e-3), \
         (y.cpu() - want).abs().max().item()
@@ -166,7 +169,10 @@ def test_attn_long_context(k):
     q = rand(B, H0 * hd, seed=72)
     y = torch.zeros(B, H0 * hd, device=DEV)
     pos = torch.tensor([298], dtype=torch.int32, device=DEV)
-    k.attn(q, H0 * hd, kc, vc, y, pos, B, H0, H0 // n_kv0, hd)
+    S = 8
+    ml = torch.zeros(B * H0 * S * 2, device=DEV)
+    osc = torch.zeros(B * H0 * S * hd, device=DEV)
+    k.attn(q, H0 * hd, kc, vc, y, pos, B, H0, H0 // n_kv0, hd, S, ml, osc)
     want = R.attention(q.cpu(), kc.cpu(), vc.cpu(), torch.tensor([298]), H0, hd)
     assert torch.allclose(y.cpu(), want, atol=1e-4, rtol=1e-3)
 
@@ -247,12 +253,12 @@ def test_gemv_argmax(k):
     bs = torch.zeros(1, n // 32, device=DEV)
     k.q80_quantize(x, q, s, bs)
     y = torch.zeros(1, d, device=DEV)
-    slot = torch.zeros(1, dtype=torch.int64, device=DEV)
-    k.q40_gemv(qs, sc, q, s, bs, y, 1, slot)
+    nblocks = int(k.q40_gemv_argmax_blocks(d))
+    scratch = torch.zeros(nblocks, dtype=torch.int64, device=DEV)
+    k.q40_gemv(qs, sc, q, s, bs, y, 1, scratch)
     tok = torch.zeros(1, dtype=torch.int64, device=DEV)
-    k.token_from_argmax(tok, slot)
+    k.token_from_argmax(tok, scratch, nblocks)
     assert int(tok.item()) == int(y[0].argmax().item())
-    assert int(slot.item()) == 0  # reset for next step
 
 
 def test_sync_pack_merge_add(k):
