@@ -204,7 +204,7 @@ class HipTransformer:
         self.ff_out = torch.zeros(NB, 2 * c.ff_dim0, device=dev)
         self.dq = QuantBuf(NB, c.ff_dim0, dev)
         self.logits0 = torch.zeros(NB, c.vocab0, device=dev)
-        self.amax_blocks = (c.vocab0 + 7) // 8  # waves_per_block(4) * RPW(2)
+        self.amax_blocks = (c.vocab0 + 15) // 16  # waves_per_block(4) * RPW(4)
         self.amax_scratch = torch.zeros(self.amax_blocks, dtype=torch.int64, device=dev)
         self.attn_splits = 8
         self.attn_ml = torch.zeros(NB * c.n_heads0 * self.attn_splits * 2, device=dev)
@@ -274,7 +274,7 @@ class HipTransformer:
         c, k = self.cfg, self.k
         NB = _pow2_batch(B)
         x = self.x
-        torch.index_select(self.embedding, 0, self.tokens[:NB], out=x[:NB])
+        k.embed_gather(self.embedding, self.tokens, x, NB)
         pending = None  # partial output not yet folded into the residual
 
         kv_mul = c.n_heads0 // max(1, c.kv_dim0 // c.head_dim)
@@ -295,9 +295,8 @@ class HipTransformer:
                       c.head_dim, self.rope_style, B)
             k.attn(self.qkv_out, self.qkv_ld, self.k_cache[l], self.v_cache[l],
                    self.zbuf[:B], self.pos, B, c.n_heads0, kv_mul, c.head_dim,
-                   self.attn_splits, self.attn_ml, self.attn_o)
-            k.q80_quantize(self.zbuf[:NB], self.zq.q[:NB], self.zq.s[:NB],
-                           self.zq.bs[:NB])
+                   self.attn_splits, self.attn_ml, self.attn_o,
+                   self.zq.q, self.zq.s, self.zq.bs)
             k.q40_gemv(lw["wo"].qs, lw["wo"].scales, self.zq.q, self.zq.s,
                        self.zq.bs, self.partial, NB)
             pending = self.partial

@@ -206,6 +206,7 @@ __global__ void k_add_rmsnorm(float *__restrict__ x,
                               float *__restrict__ s,
                               float *__restrict__ bs,
                               int n, float eps) {
+    extern __shared__ float lds[];  // n floats: x staged once, re-read from LDS
     const int64_t base = (int64_t)blockIdx.x * n;
     float acc = 0.0f;
     for (int i = threadIdx.x * 4; i < n; i += blockDim.x * 4) {
@@ -215,6 +216,7 @@ __global__ void k_add_rmsnorm(float *__restrict__ x,
             v.x += p.x; v.y += p.y; v.z += p.z; v.w += p.w;
             *reinterpret_cast<float4 *>(x + base + i) = v;
         }
+        *reinterpret_cast<float4 *>(lds + i) = v;
         acc += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
     }
     __shared__ float red[16];
@@ -229,7 +231,7 @@ __global__ void k_add_rmsnorm(float *__restrict__ x,
 
     if (!QUANT) {
         for (int i = threadIdx.x * 4; i < n; i += blockDim.x * 4) {
-            const float4 v = *reinterpret_cast<const float4 *>(x + base + i);
+            const float4 v = *reinterpret_cast<const float4 *>(lds + i);
             const float4 wv = *reinterpret_cast<const float4 *>(w + i);
             float4 o;
             o.x = v.x * inv * wv.x; o.y = v.y * inv * wv.y;
@@ -241,7 +243,7 @@ __global__ void k_add_rmsnorm(float *__restrict__ x,
         const int lane32 = threadIdx.x & 31;
         for (int blk = threadIdx.x / 32; blk < nb; blk += blockDim.x / 32) {
             const int i = blk * QB + lane32;
-            const float v = x[base + i] * inv * w[i];
+            const float v = lds[i] * inv * w[i];
             const float amax = group32_reduce_max(fabsf(v));
             const float d = amax / 127.0f;
             const float qinv = d > 0.0f ? 1.0f / d : 0.0f;
@@ -254,6 +256,20 @@ __global__ void k_add_rmsnorm(float *__restrict__ x,
             }
         }
     }
+}
+
+// embedding row gather (replaces torch index_select inside the decode graph
+// — the ATen gather kernel costs ~40 us/step there; reference OP_EMBEDDING,
+// nn-cpu-ops.cpp:982-1008).
+__global__ void k_embed_gather(const float *__restrict__ table,
+                               const long *__restrict__ tokens,
+                               float *__restrict__ x, int dim) {
+    const int b = blockIdx.y;
+    const int64_t src = (int64_t)tokens[b] * dim;
+    for (int i = (blockIdx.x * blockDim.x + threadIdx.x) * 4; i < dim;
+         i += gridDim.x * blockDim.x * 4)
+        *reinterpret_cast<float4 *>(x + (int64_t)b * dim + i) =
+            *reinterpret_cast<const float4 *>(table + src + i);
 }
 
 // argmax pack: monotonic unsigned ordering of (float value, smallest index
@@ -485,14 +501,21 @@ __global__ void k_attn_split(const float *__restrict__ q, int q_ld,
     }
 }
 
-template <int VEC>
+// Stage 2: combine the S split partials. QUANT=true additionally emits the
+// Q80 triple of the attention output directly (each workgroup owns a whole
+// head = hd/32 quant blocks), eliminating the separate cast kernel the
+// reference runs before the wo matmul.
+template <int VEC, bool QUANT>
 __global__ void k_attn_combine(const float *__restrict__ ml_scratch,
                                const float *__restrict__ o_scratch,
                                float *__restrict__ y,
+                               int8_t *__restrict__ zq,
+                               float *__restrict__ zs,
+                               float *__restrict__ zbs,
                                int n_heads0, int S) {
     const int h0 = blockIdx.x;
     const int b = blockIdx.y;
-    const int hd = VEC * WAVE;
+    const int hd = VEC * WAVE;  // blockDim.x == hd
     const int64_t base = ((int64_t)b * n_heads0 + h0) * S;
     float M = -1e30f;
     for (int sp = 0; sp < S; sp++)
@@ -501,12 +524,26 @@ __global__ void k_attn_combine(const float *__restrict__ ml_scratch,
     for (int sp = 0; sp < S; sp++)
         L += ml_scratch[(base + sp) * 2 + 1] * __expf(ml_scratch[(base + sp) * 2] - M);
     const float invL = 1.0f / L;
-    for (int i = threadIdx.x; i < hd; i += blockDim.x) {
-        float acc = 0.0f;
-        for (int sp = 0; sp < S; sp++)
-            acc += o_scratch[(base + sp) * hd + i]
-                 * __expf(ml_scratch[(base + sp) * 2] - M);
-        y[((int64_t)b * n_heads0 + h0) * hd + i] = acc * invL;
+    const int i = threadIdx.x;
+    float acc = 0.0f;
+    for (int sp = 0; sp < S; sp++)
+        acc += o_scratch[(base + sp) * hd + i]
+             * __expf(ml_scratch[(base + sp) * 2] - M);
+    const float v = acc * invL;
+    if (!QUANT) {
+        y[((int64_t)b * n_heads0 + h0) * hd + i] = v;
+    } else {
+        const float amax = group32_reduce_max(fabsf(v));
+        const float d = amax / 127.0f;
+        const float qinv = d > 0.0f ? 1.0f / d : 0.0f;
+        const float qf = rintf(v * qinv);
+        zq[((int64_t)b * n_heads0 + h0) * hd + i] = (int8_t)qf;
+        const float bsum = group32_reduce_sum(qf);
+        if ((i & 31) == 0) {
+            const int blk = (h0 * hd + i) / QB;
+            zs[(int64_t)b * (n_heads0 * hd / QB) + blk] = d;
+            zbs[(int64_t)b * (n_heads0 * hd / QB) + blk] = bsum;
+        }
     }
 }
 
@@ -827,8 +864,9 @@ void q40_gemv(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
     };
     std::integral_constant<int, 1> r1;
     std::integral_constant<int, 2> r2;
+    std::integral_constant<int, 4> r4;
     switch (batch) {
-        case 1: launch(std::integral_constant<int, 1>{}, r2); break;
+        case 1: launch(std::integral_constant<int, 1>{}, r4); break;
         case 2: launch(std::integral_constant<int, 2>{}, r2); break;
         case 4: launch(std::integral_constant<int, 4>{}, r2); break;
         case 8: launch(std::integral_constant<int, 8>{}, r1); break;
@@ -840,7 +878,7 @@ void q40_gemv(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
 
 int64_t q40_gemv_argmax_blocks(int64_t d) {
     // number of stage-1 argmax scratch entries for a batch-1 GEMV over d rows
-    return ceil_div(d, 4 * 2);  // waves_per_block * RPW
+    return ceil_div(d, 4 * 4);  // waves_per_block * RPW(batch=1)
 }
 
 void q40_gemv_grouped(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
@@ -891,33 +929,48 @@ void kv_append(torch::Tensor k, torch::Tensor v, torch::Tensor kc,
 void attn(torch::Tensor q, int64_t q_ld, torch::Tensor kc, torch::Tensor vc,
           torch::Tensor y, torch::Tensor pos, int64_t batch, int64_t n_heads0,
           int64_t kv_mul, int64_t head_dim, int64_t splits,
-          torch::Tensor ml_scratch, torch::Tensor o_scratch) {
+          torch::Tensor ml_scratch, torch::Tensor o_scratch,
+          c10::optional<torch::Tensor> zq = c10::nullopt,
+          c10::optional<torch::Tensor> zs = c10::nullopt,
+          c10::optional<torch::Tensor> zbs = c10::nullopt) {
     CHECK_CUDA(q);
     const int kv_dim0 = kc.size(1);
     const float scale = 1.0f / sqrtf((float)head_dim);
     const dim3 grid(n_heads0, batch, splits);
     const dim3 cgrid(n_heads0, batch);
-    if (head_dim == 128) {
-        hipLaunchKernelGGL(k_attn_split<2>, grid, dim3(256), 0, cur_stream(),
+    const bool quant = zq.has_value();
+    auto run = [&](auto vec_const) {
+        constexpr int V = decltype(vec_const)::value;
+        hipLaunchKernelGGL(k_attn_split<V>, grid, dim3(256), 0, cur_stream(),
                            q.data_ptr<float>(), (int)q_ld, kc.data_ptr<float>(),
                            vc.data_ptr<float>(), pos.data_ptr<int>(),
                            (int)n_heads0, (int)kv_mul, kv_dim0, scale,
                            ml_scratch.data_ptr<float>(), o_scratch.data_ptr<float>());
-        hipLaunchKernelGGL(k_attn_combine<2>, cgrid, dim3(128), 0, cur_stream(),
-                           ml_scratch.data_ptr<float>(), o_scratch.data_ptr<float>(),
-                           y.data_ptr<float>(), (int)n_heads0, (int)splits);
-    } else if (head_dim == 64) {
-        hipLaunchKernelGGL(k_attn_split<1>, grid, dim3(256), 0, cur_stream(),
-                           q.data_ptr<float>(), (int)q_ld, kc.data_ptr<float>(),
-                           vc.data_ptr<float>(), pos.data_ptr<int>(),
-                           (int)n_heads0, (int)kv_mul, kv_dim0, scale,
-                           ml_scratch.data_ptr<float>(), o_scratch.data_ptr<float>());
-        hipLaunchKernelGGL(k_attn_combine<1>, cgrid, dim3(64), 0, cur_stream(),
-                           ml_scratch.data_ptr<float>(), o_scratch.data_ptr<float>(),
-                           y.data_ptr<float>(), (int)n_heads0, (int)splits);
-    } else {
-        TORCH_CHECK(false, "unsupported head_dim ", head_dim);
-    }
+        if (quant)
+            hipLaunchKernelGGL((k_attn_combine<V, true>), cgrid, dim3(V * WAVE), 0,
+                               cur_stream(), ml_scratch.data_ptr<float>(),
+                               o_scratch.data_ptr<float>(), nullptr,
+                               zq->data_ptr<int8_t>(), zs->data_ptr<float>(),
+                               zbs->data_ptr<float>(), (int)n_heads0, (int)splits);
+        else
+            hipLaunchKernelGGL((k_attn_combine<V, false>), cgrid, dim3(V * WAVE), 0,
+                               cur_stream(), ml_scratch.data_ptr<float>(),
+                               o_scratch.data_ptr<float>(), y.data_ptr<float>(),
+                               nullptr, nullptr, nullptr, (int)n_heads0, (int)splits);
+    };
+    if (head_dim == 128) run(std::integral_constant<int, 2>{});
+    else if (head_dim == 64) run(std::integral_constant<int, 1>{});
+    else TORCH_CHECK(false, "unsupported head_dim ", head_dim);
+}
+
+void embed_gather(torch::Tensor table, torch::Tensor tokens, torch::Tensor x,
+                  int64_t batch) {
+    CHECK_CUDA(table);
+    const int dim = table.size(1);
+    const dim3 grid(ceil_div(dim / 4, 256), batch);
+    hipLaunchKernelGGL(k_embed_gather, grid, dim3(256), 0, cur_stream(),
+                       table.data_ptr<float>(), (const long *)tokens.data_ptr<int64_t>(),
+                       x.data_ptr<float>(), dim);
 }
 
 void swiglu_q80(torch::Tensor a, torch::Tensor g, int64_t lda, int64_t n,
@@ -947,12 +1000,12 @@ void add_rmsnorm(torch::Tensor x, c10::optional<torch::Tensor> partial,
     const bool add = partial.has_value();
     const float *pp = add ? partial->data_ptr<float>() : nullptr;
     if (add)
-        hipLaunchKernelGGL((k_add_rmsnorm<true, false>), dim3(rows), dim3(1024), 0,
-                           cur_stream(), x.data_ptr<float>(), pp, w.data_ptr<float>(),
+        hipLaunchKernelGGL((k_add_rmsnorm<true, false>), dim3(rows), dim3(1024),
+                           n * 4, cur_stream(), x.data_ptr<float>(), pp, w.data_ptr<float>(),
                            y.data_ptr<float>(), nullptr, nullptr, nullptr, n, (float)eps);
     else
-        hipLaunchKernelGGL((k_add_rmsnorm<false, false>), dim3(rows), dim3(1024), 0,
-                           cur_stream(), x.data_ptr<float>(), pp, w.data_ptr<float>(),
+        hipLaunchKernelGGL((k_add_rmsnorm<false, false>), dim3(rows), dim3(1024),
+                           n * 4, cur_stream(), x.data_ptr<float>(), pp, w.data_ptr<float>(),
                            y.data_ptr<float>(), nullptr, nullptr, nullptr, n, (float)eps);
 }
 
@@ -965,13 +1018,13 @@ void add_rmsnorm_q80(torch::Tensor x, c10::optional<torch::Tensor> partial,
     const bool add = partial.has_value();
     const float *pp = add ? partial->data_ptr<float>() : nullptr;
     if (add)
-        hipLaunchKernelGGL((k_add_rmsnorm<true, true>), dim3(rows), dim3(1024), 0,
-                           cur_stream(), x.data_ptr<float>(), pp, w.data_ptr<float>(),
+        hipLaunchKernelGGL((k_add_rmsnorm<true, true>), dim3(rows), dim3(1024),
+                           n * 4, cur_stream(), x.data_ptr<float>(), pp, w.data_ptr<float>(),
                            nullptr, q.data_ptr<int8_t>(), s.data_ptr<float>(),
                            bs.data_ptr<float>(), n, (float)eps);
     else
-        hipLaunchKernelGGL((k_add_rmsnorm<false, true>), dim3(rows), dim3(1024), 0,
-                           cur_stream(), x.data_ptr<float>(), pp, w.data_ptr<float>(),
+        hipLaunchKernelGGL((k_add_rmsnorm<false, true>), dim3(rows), dim3(1024),
+                           n * 4, cur_stream(), x.data_ptr<float>(), pp, w.data_ptr<float>(),
                            nullptr, q.data_ptr<int8_t>(), s.data_ptr<float>(),
                            bs.data_ptr<float>(), n, (float)eps);
 }
@@ -1053,7 +1106,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rope", &rope);
     m.def("rope_kv", &rope_kv);
     m.def("kv_append", &kv_append);
-    m.def("attn", &attn);
+    m.def("attn", &attn, py::arg("q"), py::arg("q_ld"), py::arg("kc"), py::arg("vc"),
+          py::arg("y"), py::arg("pos"), py::arg("batch"), py::arg("n_heads0"),
+          py::arg("kv_mul"), py::arg("head_dim"), py::arg("splits"),
+          py::arg("ml_scratch"), py::arg("o_scratch"),
+          py::arg("zq") = py::none(), py::arg("zs") = py::none(),
+          py::arg("zbs") = py::none());
+    m.def("embed_gather", &embed_gather);
     m.def("swiglu_q80", &swiglu_q80);
     m.def("silu_mul", &silu_mul);
     m.def("sync_pack", &sync_pack);
