@@ -204,7 +204,8 @@ class HipTransformer:
         self.ff_out = torch.zeros(NB, 2 * c.ff_dim0, device=dev)
         self.dq = QuantBuf(NB, c.ff_dim0, dev)
         self.logits0 = torch.zeros(NB, c.vocab0, device=dev)
-        self.amax_blocks = (c.vocab0 + 15) // 16  # waves_per_block(4) * RPW(4)
+        rpw = 4 if c.vocab0 >= 16384 else (2 if c.vocab0 >= 2048 else 1)
+        self.amax_blocks = -(-c.vocab0 // (4 * rpw))  # mirrors gemv RPW choice
         self.amax_scratch = torch.zeros(self.amax_blocks, dtype=torch.int64, device=dev)
         self.attn_splits = 8
         self.attn_ml = torch.zeros(NB * c.n_heads0 * self.attn_splits * 2, device=dev)

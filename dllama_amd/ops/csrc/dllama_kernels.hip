@@ -866,7 +866,12 @@ void q40_gemv(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
     std::integral_constant<int, 2> r2;
     std::integral_constant<int, 4> r4;
     switch (batch) {
-        case 1: launch(std::integral_constant<int, 1>{}, r4); break;
+        case 1:
+            // keep >= ~2 workgroups/CU: more rows/wave only when d is huge
+            if (d >= 16384) launch(std::integral_constant<int, 1>{}, r4);
+            else if (d >= 2048) launch(std::integral_constant<int, 1>{}, r2);
+            else launch(std::integral_constant<int, 1>{}, r1);
+            break;
         case 2: launch(std::integral_constant<int, 2>{}, r2); break;
         case 4: launch(std::integral_constant<int, 4>{}, r2); break;
         case 8: launch(std::integral_constant<int, 8>{}, r1); break;
@@ -877,8 +882,10 @@ void q40_gemv(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
 }
 
 int64_t q40_gemv_argmax_blocks(int64_t d) {
-    // number of stage-1 argmax scratch entries for a batch-1 GEMV over d rows
-    return ceil_div(d, 4 * 4);  // waves_per_block * RPW(batch=1)
+    // stage-1 argmax scratch entries for a batch-1 GEMV over d rows;
+    // must mirror the RPW selection above
+    const int rpw = d >= 16384 ? 4 : (d >= 2048 ? 2 : 1);
+    return ceil_div(d, 4 * rpw);
 }
 
 void q40_gemv_grouped(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
