@@ -204,8 +204,9 @@ class HipTransformer:
         self.ff_out = torch.zeros(NB, 2 * c.ff_dim0, device=dev)
         self.dq = QuantBuf(NB, c.ff_dim0, dev)
         self.logits0 = torch.zeros(NB, c.vocab0, device=dev)
-        rpw = 4 if c.vocab0 >= 16384 else (2 if c.vocab0 >= 2048 else 1)
+        rpw = 2 if c.vocab0 >= 2048 else 1
         self.amax_blocks = -(-c.vocab0 // (4 * rpw))  # mirrors gemv RPW choice
+        self.ssq = torch.zeros(2 * c.n_layers + 1, NB, device=dev)
         self.amax_scratch = torch.zeros(self.amax_blocks, dtype=torch.int64, device=dev)
         self.attn_splits = 8
         self.attn_ml = torch.zeros(NB * c.n_heads0 * self.attn_splits * 2, device=dev)
@@ -237,8 +238,9 @@ class HipTransformer:
 
     # ------------------------------------------------------------ forward
 
-    def _sync_partial(self, B: int, NB: int):
-        """TP>1: all-reduce self.partial[:NB] into x (+= sum of partials)."""
+    def _sync_partial(self, NB: int, slot: int):
+        """TP>1: all-reduce self.partial[:NB] into x (+= sum of partials),
+        accumulating the residual row's sum-of-squares into ssq[slot]."""
         c = self.cfg
         if c.sync_type == Q80:
             nb_dim = c.dim // QB
@@ -249,24 +251,22 @@ class HipTransformer:
             self.k.sync_pack(q.q[:NB], q.s[:NB], out)
             inb = self.sync_in[:, : NB * row_bytes]
             self.comm.all_gather(inb, out)
-            self.k.merge_add(self.x[:NB], inb)
+            self.k.merge_add(self.x[:NB], inb, self.ssq[slot])
         else:
             self.comm.allreduce_(self.partial[:NB])
-            self.k.add_(self.x[:NB], self.partial[:NB])
+            self.k.add_ssq(self.x[:NB], self.partial[:NB], self.ssq[slot], NB)
 
-    def _merge_norm(self, pending, w, NB: int, quant: bool):
-        """Fold the pending partial into x and norm: TP=1 fuses
-        add+rmsnorm(+q80) into one kernel; TP>1 syncs first."""
+    def _proj_merge(self, lin: Linear, qb: QuantBuf, slot: int, NB: int):
+        """Down-projection + residual fold: TP=1 fuses the add and the ssq
+        accumulation into the GEMV epilogue; TP>1 produces the partial and
+        syncs it."""
         k = self.k
-        if pending is not None and self.cfg.world > 1:
-            self._sync_partial(NB, NB)
-            pending = None
-        if quant:
-            k.add_rmsnorm_q80(self.x[:NB], pending, w, self.xq.q[:NB],
-                              self.xq.s[:NB], self.xq.bs[:NB], self.cfg.norm_eps)
+        if self.cfg.world == 1:
+            k.q40_gemv_resid(lin.qs, lin.scales, qb.q, qb.s, qb.bs,
+                             self.x, self.ssq[slot], NB)
         else:
-            k.add_rmsnorm(self.x[:NB], pending, w, self.t_norm[:NB],
-                          self.cfg.norm_eps)
+            k.q40_gemv(lin.qs, lin.scales, qb.q, qb.s, qb.bs, self.partial, NB)
+            self._sync_partial(NB, slot)
 
     def forward_buffers(self, B: int):
         """Run one step over tokens[:B] at positions pos..pos+B-1, writing
@@ -275,50 +275,64 @@ class HipTransformer:
         c, k = self.cfg, self.k
         NB = _pow2_batch(B)
         x = self.x
-        k.embed_gather(self.embedding, self.tokens, x, NB)
-        pending = None  # partial output not yet folded into the residual
+        self.ssq.zero_()
+        k.embed_gather(self.embedding, self.tokens, x, NB, self.ssq[0])
 
         kv_mul = c.n_heads0 // max(1, c.kv_dim0 // c.head_dim)
+        fused_rope = self.rope_style == 0 and not c.is_qwen3
+        slot = 0
         for l, lw in enumerate(self.layers):
             # attention block
-            self._merge_norm(pending, lw["norm0"], NB, quant=True)
-            k.q40_gemv(lw["qkv"].qs, lw["qkv"].scales, self.xq.q, self.xq.s,
-                       self.xq.bs, self.qkv_out, NB)
-            if c.is_qwen3:
-                k.rmsnorm_rows_s(self.qkv_out, self.qkv_ld, 0,
-                                 c.q_dim0 // c.head_dim, B, lw["q_norm"],
-                                 c.head_dim, c.norm_eps)
-                k.rmsnorm_rows_s(self.qkv_out, self.qkv_ld, c.q_dim0,
-                                 c.kv_dim0 // c.head_dim, B, lw["k_norm"],
-                                 c.head_dim, c.norm_eps)
-            k.rope_kv(self.qkv_out, self.qkv_ld, c.q_dim0, c.kv_dim0,
-                      self.rope_cache, self.pos, self.k_cache[l], self.v_cache[l],
-                      c.head_dim, self.rope_style, B)
+            k.norm_quant(x[:NB], lw["norm0"], self.ssq[slot], self.xq.q[:NB],
+                         self.xq.s[:NB], self.xq.bs[:NB], NB, c.norm_eps)
+            if fused_rope:
+                # GEMV epilogue rotates q/k and scatters k,v into the cache
+                k.q40_gemv_rope(lw["qkv"].qs, lw["qkv"].scales, self.xq.q,
+                                self.xq.s, self.xq.bs, self.qkv_out, NB,
+                                self.rope_cache, self.pos, self.k_cache[l],
+                                self.v_cache[l], c.q_dim0, c.kv_dim0, c.head_dim)
+            else:
+                k.q40_gemv(lw["qkv"].qs, lw["qkv"].scales, self.xq.q, self.xq.s,
+                           self.xq.bs, self.qkv_out, NB)
+                if c.is_qwen3:
+                    k.rmsnorm_rows_s(self.qkv_out, self.qkv_ld, 0,
+                                     c.q_dim0 // c.head_dim, B, lw["q_norm"],
+                                     c.head_dim, c.norm_eps)
+                    k.rmsnorm_rows_s(self.qkv_out, self.qkv_ld, c.q_dim0,
+                                     c.kv_dim0 // c.head_dim, B, lw["k_norm"],
+                                     c.head_dim, c.norm_eps)
+                k.rope_kv(self.qkv_out, self.qkv_ld, c.q_dim0, c.kv_dim0,
+                          self.rope_cache, self.pos, self.k_cache[l],
+                          self.v_cache[l], c.head_dim, self.rope_style, B)
             k.attn(self.qkv_out, self.qkv_ld, self.k_cache[l], self.v_cache[l],
                    self.zbuf[:B], self.pos, B, c.n_heads0, kv_mul, c.head_dim,
                    self.attn_splits, self.attn_ml, self.attn_o,
                    self.zq.q, self.zq.s, self.zq.bs)
-            k.q40_gemv(lw["wo"].qs, lw["wo"].scales, self.zq.q, self.zq.s,
-                       self.zq.bs, self.partial, NB)
-            pending = self.partial
+            self._proj_merge(lw["wo"], self.zq, slot + 1, NB)
+            slot += 1
 
             # ffn block
             if c.is_moe:
-                self._merge_norm(pending, lw["norm1"], NB, quant=False)
-                pending = None
+                k.norm_f32(x[:NB], lw["norm1"], self.ssq[slot], self.t_norm[:NB],
+                           NB, c.norm_eps)
                 self._moe_ffn(B, NB, lw)
+                if c.world == 1:
+                    k.add_ssq(x[:NB], self.partial[:NB], self.ssq[slot + 1], NB)
+                else:
+                    self._sync_partial(NB, slot + 1)
             else:
-                self._merge_norm(pending, lw["norm1"], NB, quant=True)
+                k.norm_quant(x[:NB], lw["norm1"], self.ssq[slot], self.xq.q[:NB],
+                             self.xq.s[:NB], self.xq.bs[:NB], NB, c.norm_eps)
                 k.q40_gemv(lw["w13"].qs, lw["w13"].scales, self.xq.q, self.xq.s,
                            self.xq.bs, self.ff_out, NB)
                 k.swiglu_q80(self.ff_out, self.ff_out[:, c.ff_dim0:],
                              2 * c.ff_dim0, c.ff_dim0, NB, self.dq.q[:NB],
                              self.dq.s[:NB], self.dq.bs[:NB])
-                k.q40_gemv(lw["w2"].qs, lw["w2"].scales, self.dq.q, self.dq.s,
-                           self.dq.bs, self.partial, NB)
-            pending = self.partial
+                self._proj_merge(lw["w2"], self.dq, slot + 1, NB)
+            slot += 1
 
-        self._merge_norm(pending, self.final_norm, NB, quant=True)
+        k.norm_quant(x[:NB], self.final_norm, self.ssq[slot], self.xq.q[:NB],
+                     self.xq.s[:NB], self.xq.bs[:NB], NB, c.norm_eps)
         use_amax = (self.greedy_feedback and B == 1 and c.world == 1)
         k.q40_gemv(self.wcls.qs, self.wcls.scales, self.xq.q, self.xq.s,
                    self.xq.bs, self.logits0, NB,

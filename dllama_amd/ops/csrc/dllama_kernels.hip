@@ -263,13 +263,104 @@ __global__ void k_add_rmsnorm(float *__restrict__ x,
 // nn-cpu-ops.cpp:982-1008).
 __global__ void k_embed_gather(const float *__restrict__ table,
                                const long *__restrict__ tokens,
-                               float *__restrict__ x, int dim) {
+                               float *__restrict__ x, int dim,
+                               float *__restrict__ ssq) {
     const int b = blockIdx.y;
     const int64_t src = (int64_t)tokens[b] * dim;
+    float local = 0.0f;
     for (int i = (blockIdx.x * blockDim.x + threadIdx.x) * 4; i < dim;
-         i += gridDim.x * blockDim.x * 4)
-        *reinterpret_cast<float4 *>(x + (int64_t)b * dim + i) =
-            *reinterpret_cast<const float4 *>(table + src + i);
+         i += gridDim.x * blockDim.x * 4) {
+        const float4 v = *reinterpret_cast<const float4 *>(table + src + i);
+        *reinterpret_cast<float4 *>(x + (int64_t)b * dim + i) = v;
+        local += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+    }
+    if (ssq != nullptr) {
+        local = wave_reduce_sum(local);
+        __shared__ float red[16];
+        const int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+        if (lane == 0) red[wid] = local;
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            float t = 0.0f;
+            for (int i = 0; i < blockDim.x / WAVE; i++) t += red[i];
+            atomicAdd(ssq + b, t);
+        }
+    }
+}
+
+// ------------------------------------------- wide norm from precomputed ssq
+// The residual row's sum-of-squares arrives precomputed (GEMV RESID epilogue
+// / embed / merge-add), so the rmsnorm becomes one wide pass with no
+// in-kernel reduction: quantized (Q80 triple) or f32 output.
+__global__ void k_norm_quant(const float *__restrict__ x,
+                             const float *__restrict__ w,
+                             const float *__restrict__ ssq,
+                             int8_t *__restrict__ q,
+                             float *__restrict__ s,
+                             float *__restrict__ bs,
+                             int n, float eps) {
+    const int b = blockIdx.y;
+    const float inv = rsqrtf(ssq[b] / n + eps);
+    const int nb = n / QB;
+    const int gid = blockIdx.x * blockDim.x + threadIdx.x;
+    const int blk = gid / 32;
+    const int lane = threadIdx.x & 31;
+    if (blk >= nb) return;
+    const int i = blk * QB + lane;
+    const float v = x[(int64_t)b * n + i] * inv * w[i];
+    const float amax = group32_reduce_max(fabsf(v));
+    const float dd = amax / 127.0f;
+    const float qinv = dd > 0.0f ? 1.0f / dd : 0.0f;
+    const float qf = rintf(v * qinv);
+    q[(int64_t)b * n + i] = (int8_t)qf;
+    const float bsum = group32_reduce_sum(qf);
+    if (lane == 0) {
+        s[(int64_t)b * nb + blk] = dd;
+        bs[(int64_t)b * nb + blk] = bsum;
+    }
+}
+
+__global__ void k_norm_f32(const float *__restrict__ x,
+                           const float *__restrict__ w,
+                           const float *__restrict__ ssq,
+                           float *__restrict__ y,
+                           int n, float eps) {
+    const int b = blockIdx.y;
+    const float inv = rsqrtf(ssq[b] / n + eps);
+    for (int i = (blockIdx.x * blockDim.x + threadIdx.x) * 4; i < n;
+         i += gridDim.x * blockDim.x * 4) {
+        const float4 v = *reinterpret_cast<const float4 *>(x + (int64_t)b * n + i);
+        const float4 wv = *reinterpret_cast<const float4 *>(w + i);
+        float4 o;
+        o.x = v.x * inv * wv.x; o.y = v.y * inv * wv.y;
+        o.z = v.z * inv * wv.z; o.w = v.w * inv * wv.w;
+        *reinterpret_cast<float4 *>(y + (int64_t)b * n + i) = o;
+    }
+}
+
+// x[b] += p[b] with per-row sum-of-squares accumulation (residual fold for
+// paths that produce a separate partial: MoE weighted sum, f32 TP sync).
+__global__ void k_add_ssq(float *__restrict__ x,
+                          const float *__restrict__ p,
+                          float *__restrict__ ssq, int n) {
+    const int b = blockIdx.y;
+    float local = 0.0f;
+    for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * blockDim.x) {
+        const float v = x[(int64_t)b * n + i] + p[(int64_t)b * n + i];
+        x[(int64_t)b * n + i] = v;
+        local += v * v;
+    }
+    local = wave_reduce_sum(local);
+    __shared__ float red[16];
+    const int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    if (lane == 0) red[wid] = local;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float t = 0.0f;
+        for (int i = 0; i < blockDim.x / WAVE; i++) t += red[i];
+        atomicAdd(ssq + b, t);
+    }
 }
 
 // argmax pack: monotonic unsigned ordering of (float value, smallest index
@@ -325,10 +416,20 @@ __device__ __forceinline__ int q40_block_dot(const uint4 wq, const int4 x0,
     return idot;
 }
 
+// GEMV epilogue modes: fusing the ops that FOLLOW a matmul into its tail
+// removes whole kernels from the per-layer chain (the reference runs each
+// as its own op, llm.cpp:263-557).
+#define EPI_NONE 0
+#define EPI_RESID 1  // x[b,row] += v; atomically accumulate sum(x'^2) per b
+                     //  -> the next norm becomes a single wide pass
+#define EPI_ROPE 2   // llama rope: rows (2j,2j+1) are a rotation pair held by
+                     //  one RPW=2 wave; q rotated into y, k rotated into the
+                     //  KV cache, v copied into the cache
+
 // RPW = rows per wave: processing 2 rows per wave doubles the independent
 // 16B weight loads in flight per lane (the decode GEMV is HBM-latency
 // limited at 1 row/wave). NB = batch columns.
-template <int NB, int RPW>
+template <int NB, int RPW, int EPI>
 __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
                            const __half *__restrict__ scales,
                            const int8_t *__restrict__ xq,
@@ -336,7 +437,14 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
                            const float *__restrict__ xbs,
                            float *__restrict__ y,
                            int d, int n,
-                           unsigned long long *__restrict__ amax_scratch) {
+                           unsigned long long *__restrict__ amax_scratch,
+                           float *__restrict__ x_resid,
+                           float *__restrict__ ssq,
+                           const float *__restrict__ rope_cache,
+                           const int *__restrict__ pos,
+                           float *__restrict__ kc,
+                           float *__restrict__ vc,
+                           int q_dim0, int kv_dim0, int hd) {
     const int wpb = blockDim.x / WAVE;
     const int wid = threadIdx.x / WAVE;
     const int row0 = (blockIdx.x * wpb + wid) * RPW;
@@ -403,20 +511,55 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
     }
 
     unsigned long long wave_best = 0ull;
+    float ssq_local[NB];
     #pragma unroll
-    for (int r = 0; r < RPW; r++) {
-        const int row = rbase + r;
-        const bool ok = wave_valid && row < d;
+    for (int b = 0; b < NB; b++) ssq_local[b] = 0.0f;
+    const int pos0 = (EPI == EPI_ROPE) ? pos[0] : 0;
+
+    #pragma unroll
+    for (int b = 0; b < NB; b++) {
+        float v[RPW];
         #pragma unroll
-        for (int b = 0; b < NB; b++) {
-            const float v = wave_reduce_sum(acc[r][b]);
-            if (lane == 0 && ok) {
-                y[(int64_t)b * d + row] = v;
-                if (NB == 1) wave_best = max(wave_best, argmax_pack(v, row));
+        for (int r = 0; r < RPW; r++) v[r] = wave_reduce_sum(acc[r][b]);
+        if (lane != 0 || !wave_valid) continue;
+        if (EPI == EPI_ROPE) {
+            // RPW==2: rows (rbase, rbase+1) form one llama rotation pair
+            const int pb = pos0 + b;
+            if (rbase < q_dim0) {
+                const int j = (rbase % hd) >> 1;
+                const float cr = rope_cache[(int64_t)pb * hd + 2 * j];
+                const float ci = rope_cache[(int64_t)pb * hd + 2 * j + 1];
+                y[(int64_t)b * d + rbase] = v[0] * cr - v[1 % RPW] * ci;
+                y[(int64_t)b * d + rbase + 1] = v[0] * ci + v[1 % RPW] * cr;
+            } else if (rbase < q_dim0 + kv_dim0) {
+                const int rk = rbase - q_dim0;
+                const int j = (rk % hd) >> 1;
+                const float cr = rope_cache[(int64_t)pb * hd + 2 * j];
+                const float ci = rope_cache[(int64_t)pb * hd + 2 * j + 1];
+                kc[(int64_t)pb * kv_dim0 + rk] = v[0] * cr - v[1 % RPW] * ci;
+                kc[(int64_t)pb * kv_dim0 + rk + 1] = v[0] * ci + v[1 % RPW] * cr;
+            } else {
+                const int rv = rbase - q_dim0 - kv_dim0;
+                vc[(int64_t)pb * kv_dim0 + rv] = v[0];
+                if (RPW == 2) vc[(int64_t)pb * kv_dim0 + rv + 1] = v[1 % RPW];
+            }
+        } else {
+            #pragma unroll
+            for (int r = 0; r < RPW; r++) {
+                const int row = rbase + r;
+                if (row >= d) continue;
+                if (EPI == EPI_RESID) {
+                    const float xv = x_resid[(int64_t)b * d + row] + v[r];
+                    x_resid[(int64_t)b * d + row] = xv;
+                    ssq_local[b] += xv * xv;
+                } else {
+                    y[(int64_t)b * d + row] = v[r];
+                    if (NB == 1) wave_best = max(wave_best, argmax_pack(v[r], row));
+                }
             }
         }
     }
-    if (NB == 1 && amax_scratch != nullptr) {
+    if (EPI == EPI_NONE && NB == 1 && amax_scratch != nullptr) {
         // stage 1 of the greedy argmax: one packed best per workgroup
         __shared__ unsigned long long wb[16];
         if (lane == 0) wb[wid] = wave_best;
@@ -425,6 +568,18 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
             unsigned long long best = wb[0];
             for (int i = 1; i < wpb; i++) best = max(best, wb[i]);
             amax_scratch[blockIdx.x] = best;
+        }
+    }
+    if (EPI == EPI_RESID) {
+        __shared__ float sred[4][NB];
+        if (lane == 0)
+            #pragma unroll
+            for (int b = 0; b < NB; b++) sred[wid][b] = wave_valid ? ssq_local[b] : 0.0f;
+        __syncthreads();
+        if (threadIdx.x < NB) {
+            float t = 0.0f;
+            for (int wv = 0; wv < wpb; wv++) t += sred[wv][threadIdx.x];
+            atomicAdd(ssq + threadIdx.x, t);
         }
     }
 }
@@ -761,13 +916,15 @@ __global__ void k_sync_pack(const int8_t *__restrict__ q,
 // (reference OP_MERGE_ADD, nn-cpu-ops.cpp:920-957).
 __global__ void k_merge_add(float *__restrict__ x,
                             const uint8_t *__restrict__ bufs,
+                            float *__restrict__ ssq,
                             int world, int n, int rows) {
     const int nb = n / QB;
     const int row_bytes = n + 2 * nb;
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
-         i < (int64_t)rows * n; i += (int64_t)gridDim.x * blockDim.x) {
-        const int r = i / n, c = i % n;
-        float acc = x[i];
+    const int r = blockIdx.y;
+    float local = 0.0f;
+    for (int c = blockIdx.x * blockDim.x + threadIdx.x; c < n;
+         c += gridDim.x * blockDim.x) {
+        float acc = x[(int64_t)r * n + c];
         for (int w = 0; w < world; w++) {
             const uint8_t *row = bufs + ((int64_t)w * rows + r) * row_bytes;
             const int8_t qv = (int8_t)row[c];
@@ -776,7 +933,20 @@ __global__ void k_merge_add(float *__restrict__ x,
             const __half h = *reinterpret_cast<const __half *>(&u);
             acc = fmaf((float)qv, __half2float(h), acc);
         }
-        x[i] = acc;
+        x[(int64_t)r * n + c] = acc;
+        local += acc * acc;
+    }
+    if (ssq != nullptr) {
+        local = wave_reduce_sum(local);
+        __shared__ float red[16];
+        const int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+        if (lane == 0) red[wid] = local;
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            float t = 0.0f;
+            for (int i = 0; i < blockDim.x / WAVE; i++) t += red[i];
+            atomicAdd(ssq + r, t);
+        }
     }
 }
 
@@ -839,38 +1009,61 @@ void rmsnorm_rows(torch::Tensor x, torch::Tensor w, torch::Tensor y, double eps)
                        hd, (float)eps);
 }
 
-void q40_gemv(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
-              torch::Tensor xs, torch::Tensor xbs, torch::Tensor y, int64_t batch,
-              c10::optional<torch::Tensor> amax_slot = c10::nullopt) {
-    CHECK_CUDA(qs); CHECK_CONT(qs); CHECK_CONT(xq);
+struct GemvEpi {
+    unsigned long long *slot = nullptr;
+    float *x_resid = nullptr;
+    float *ssq = nullptr;
+    const float *cache = nullptr;
+    const int *pos = nullptr;
+    float *kc = nullptr;
+    float *vc = nullptr;
+    int q_dim0 = 0, kv_dim0 = 0, hd = 0;
+    bool force_rpw2 = false;
+};
+
+template <int EPI>
+static void gemv_launch(torch::Tensor &qs, torch::Tensor &scales, torch::Tensor &xq,
+                        torch::Tensor &xs, torch::Tensor &xbs, float *y,
+                        int64_t batch, const GemvEpi &e) {
     const int d = qs.size(0);
     const int n = qs.size(1) * 2;
     TORCH_CHECK(xq.size(-1) == n, "x width mismatch");
     TORCH_CHECK(n % QB == 0, "n must be a multiple of 32");
-    unsigned long long *slot = nullptr;
-    if (amax_slot.has_value())
-        slot = reinterpret_cast<unsigned long long *>(amax_slot->data_ptr<int64_t>());
     const int waves_per_block = 4;
     const dim3 block(waves_per_block * WAVE);
     auto launch = [&](auto nb_const, auto rpw_const) {
         constexpr int RPW = decltype(rpw_const)::value;
         const dim3 grid(ceil_div(d, waves_per_block * RPW));
-        hipLaunchKernelGGL((k_q40_gemv<decltype(nb_const)::value, RPW>), grid, block,
-                           0, cur_stream(),
+        hipLaunchKernelGGL((k_q40_gemv<decltype(nb_const)::value, RPW, EPI>), grid,
+                           block, 0, cur_stream(),
                            qs.data_ptr<uint8_t>(),
                            reinterpret_cast<const __half *>(scales.data_ptr<at::Half>()),
                            xq.data_ptr<int8_t>(), xs.data_ptr<float>(),
-                           xbs.data_ptr<float>(), y.data_ptr<float>(), d, n, slot);
+                           xbs.data_ptr<float>(), y, d, n, e.slot, e.x_resid, e.ssq,
+                           e.cache, e.pos, e.kc, e.vc, e.q_dim0, e.kv_dim0, e.hd);
     };
     std::integral_constant<int, 1> r1;
     std::integral_constant<int, 2> r2;
-    std::integral_constant<int, 4> r4;
+    if (EPI == EPI_ROPE) {
+        // rotation pairs require RPW=2 at every batch size
+        switch (batch) {
+            case 1: launch(std::integral_constant<int, 1>{}, r2); break;
+            case 2: launch(std::integral_constant<int, 2>{}, r2); break;
+            case 4: launch(std::integral_constant<int, 4>{}, r2); break;
+            case 8: launch(std::integral_constant<int, 8>{}, r2); break;
+            case 16: launch(std::integral_constant<int, 16>{}, r2); break;
+            case 32: launch(std::integral_constant<int, 32>{}, r2); break;
+            default: TORCH_CHECK(false, "unsupported batch ", batch);
+        }
+        return;
+    }
     switch (batch) {
         case 1:
-            // keep >= ~2 workgroups/CU: more rows/wave only when d is huge
-            if (d >= 16384) launch(std::integral_constant<int, 1>{}, r4);
-            else if (d >= 2048) launch(std::integral_constant<int, 1>{}, r2);
-            else launch(std::integral_constant<int, 1>{}, r1);
+            // RPW=2 is the sweet spot (RPW=4 halves wave count and loses)
+            if (d >= 2048 || e.force_rpw2)
+                launch(std::integral_constant<int, 1>{}, r2);
+            else
+                launch(std::integral_constant<int, 1>{}, r1);
             break;
         case 2: launch(std::integral_constant<int, 2>{}, r2); break;
         case 4: launch(std::integral_constant<int, 4>{}, r2); break;
@@ -881,10 +1074,49 @@ void q40_gemv(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
     }
 }
 
+void q40_gemv(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
+              torch::Tensor xs, torch::Tensor xbs, torch::Tensor y, int64_t batch,
+              c10::optional<torch::Tensor> amax_slot = c10::nullopt) {
+    CHECK_CUDA(qs); CHECK_CONT(qs); CHECK_CONT(xq);
+    GemvEpi e;
+    if (amax_slot.has_value())
+        e.slot = reinterpret_cast<unsigned long long *>(amax_slot->data_ptr<int64_t>());
+    gemv_launch<EPI_NONE>(qs, scales, xq, xs, xbs, y.data_ptr<float>(), batch, e);
+}
+
+void q40_gemv_resid(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
+                    torch::Tensor xs, torch::Tensor xbs, torch::Tensor x,
+                    torch::Tensor ssq, int64_t batch) {
+    CHECK_CUDA(qs); CHECK_CONT(qs); CHECK_CONT(xq);
+    TORCH_CHECK(batch <= 32, "RESID epilogue LDS sized for batch<=32");
+    GemvEpi e;
+    e.x_resid = x.data_ptr<float>();
+    e.ssq = ssq.data_ptr<float>();
+    gemv_launch<EPI_RESID>(qs, scales, xq, xs, xbs, x.data_ptr<float>(), batch, e);
+}
+
+void q40_gemv_rope(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
+                   torch::Tensor xs, torch::Tensor xbs, torch::Tensor y,
+                   int64_t batch, torch::Tensor cache, torch::Tensor pos,
+                   torch::Tensor kc, torch::Tensor vc, int64_t q_dim0,
+                   int64_t kv_dim0, int64_t head_dim) {
+    CHECK_CUDA(qs); CHECK_CONT(qs); CHECK_CONT(xq);
+    GemvEpi e;
+    e.cache = cache.data_ptr<float>();
+    e.pos = pos.data_ptr<int>();
+    e.kc = kc.data_ptr<float>();
+    e.vc = vc.data_ptr<float>();
+    e.q_dim0 = (int)q_dim0;
+    e.kv_dim0 = (int)kv_dim0;
+    e.hd = (int)head_dim;
+    e.force_rpw2 = true;  // rotation pairs live in one RPW=2 wave
+    gemv_launch<EPI_ROPE>(qs, scales, xq, xs, xbs, y.data_ptr<float>(), batch, e);
+}
+
 int64_t q40_gemv_argmax_blocks(int64_t d) {
     // stage-1 argmax scratch entries for a batch-1 GEMV over d rows;
     // must mirror the RPW selection above
-    const int rpw = d >= 16384 ? 4 : (d >= 2048 ? 2 : 1);
+    const int rpw = d >= 2048 ? 2 : 1;
     return ceil_div(d, 4 * rpw);
 }
 
@@ -971,13 +1203,45 @@ void attn(torch::Tensor q, int64_t q_ld, torch::Tensor kc, torch::Tensor vc,
 }
 
 void embed_gather(torch::Tensor table, torch::Tensor tokens, torch::Tensor x,
-                  int64_t batch) {
+                  int64_t batch, c10::optional<torch::Tensor> ssq = c10::nullopt) {
     CHECK_CUDA(table);
     const int dim = table.size(1);
     const dim3 grid(ceil_div(dim / 4, 256), batch);
     hipLaunchKernelGGL(k_embed_gather, grid, dim3(256), 0, cur_stream(),
                        table.data_ptr<float>(), (const long *)tokens.data_ptr<int64_t>(),
-                       x.data_ptr<float>(), dim);
+                       x.data_ptr<float>(), dim,
+                       ssq.has_value() ? ssq->data_ptr<float>() : nullptr);
+}
+
+void norm_quant(torch::Tensor x, torch::Tensor w, torch::Tensor ssq,
+                torch::Tensor q, torch::Tensor s, torch::Tensor bs,
+                int64_t batch, double eps) {
+    CHECK_CUDA(x);
+    const int n = x.size(-1);
+    const dim3 grid(ceil_div(n, 256), batch);
+    hipLaunchKernelGGL(k_norm_quant, grid, dim3(256), 0, cur_stream(),
+                       x.data_ptr<float>(), w.data_ptr<float>(), ssq.data_ptr<float>(),
+                       q.data_ptr<int8_t>(), s.data_ptr<float>(), bs.data_ptr<float>(),
+                       n, (float)eps);
+}
+
+void norm_f32(torch::Tensor x, torch::Tensor w, torch::Tensor ssq,
+              torch::Tensor y, int64_t batch, double eps) {
+    CHECK_CUDA(x);
+    const int n = x.size(-1);
+    const dim3 grid(ceil_div(n / 4, 256), batch);
+    hipLaunchKernelGGL(k_norm_f32, grid, dim3(256), 0, cur_stream(),
+                       x.data_ptr<float>(), w.data_ptr<float>(), ssq.data_ptr<float>(),
+                       y.data_ptr<float>(), n, (float)eps);
+}
+
+void add_ssq(torch::Tensor x, torch::Tensor p, torch::Tensor ssq, int64_t batch) {
+    CHECK_CUDA(x);
+    const int n = x.size(-1);
+    const dim3 grid(ceil_div(n, 1024), batch);
+    hipLaunchKernelGGL(k_add_ssq, grid, dim3(256), 0, cur_stream(),
+                       x.data_ptr<float>(), p.data_ptr<float>(),
+                       ssq.data_ptr<float>(), n);
 }
 
 void swiglu_q80(torch::Tensor a, torch::Tensor g, int64_t lda, int64_t n,
@@ -1077,13 +1341,15 @@ void sync_pack(torch::Tensor q, torch::Tensor s, torch::Tensor buf) {
                        buf.data_ptr<uint8_t>(), n, rows);
 }
 
-void merge_add(torch::Tensor x, torch::Tensor bufs) {
+void merge_add(torch::Tensor x, torch::Tensor bufs,
+               c10::optional<torch::Tensor> ssq = c10::nullopt) {
     CHECK_CUDA(x);
     const int n = x.size(-1);
     const int rows = x.numel() / n;
     const int world = bufs.size(0);
-    hipLaunchKernelGGL(k_merge_add, dim3(ceil_div((int64_t)rows * n, 256)), dim3(256),
+    hipLaunchKernelGGL(k_merge_add, dim3(ceil_div(n, 1024), rows), dim3(256),
                        0, cur_stream(), x.data_ptr<float>(), bufs.data_ptr<uint8_t>(),
+                       ssq.has_value() ? ssq->data_ptr<float>() : nullptr,
                        world, n, rows);
 }
 
@@ -1109,6 +1375,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("q40_gemv", &q40_gemv, py::arg("qs"), py::arg("scales"), py::arg("xq"),
           py::arg("xs"), py::arg("xbs"), py::arg("y"), py::arg("batch"),
           py::arg("amax_slot") = py::none());
+    m.def("q40_gemv_resid", &q40_gemv_resid);
+    m.def("q40_gemv_rope", &q40_gemv_rope);
+    m.def("norm_quant", &norm_quant);
+    m.def("norm_f32", &norm_f32);
+    m.def("add_ssq", &add_ssq);
     m.def("q40_gemv_grouped", &q40_gemv_grouped);
     m.def("rope", &rope);
     m.def("rope_kv", &rope_kv);
@@ -1119,11 +1390,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("ml_scratch"), py::arg("o_scratch"),
           py::arg("zq") = py::none(), py::arg("zs") = py::none(),
           py::arg("zbs") = py::none());
-    m.def("embed_gather", &embed_gather);
+    m.def("embed_gather", &embed_gather, py::arg("table"), py::arg("tokens"), py::arg("x"), py::arg("batch"), py::arg("ssq") = py::none());
     m.def("swiglu_q80", &swiglu_q80);
     m.def("silu_mul", &silu_mul);
     m.def("sync_pack", &sync_pack);
-    m.def("merge_add", &merge_add);
+    m.def("merge_add", &merge_add, py::arg("x"), py::arg("bufs"), py::arg("ssq") = py::none());
     m.def("add_", &add_);
     m.def("pos_inc", &pos_inc);
     m.def("token_from_argmax", &token_from_argmax);
