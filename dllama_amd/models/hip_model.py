@@ -206,7 +206,8 @@ class HipTransformer:
         self.logits0 = torch.zeros(NB, c.vocab0, device=dev)
         rpw = 2 if c.vocab0 >= 2048 else 1
         self.amax_blocks = -(-c.vocab0 // (4 * rpw))  # mirrors gemv RPW choice
-        self.ssq = torch.zeros(2 * c.n_layers + 1, NB, device=dev)
+        # sum-of-squares accumulators: [slot, batch, 16-way spread]
+        self.ssq = torch.zeros(2 * c.n_layers + 1, NB, 16, device=dev)
         self.amax_scratch = torch.zeros(self.amax_blocks, dtype=torch.int64, device=dev)
         self.attn_splits = 8
         self.attn_ml = torch.zeros(NB * c.n_heads0 * self.attn_splits * 2, device=dev)
@@ -280,20 +281,37 @@ class HipTransformer:
 
         kv_mul = c.n_heads0 // max(1, c.kv_dim0 // c.head_dim)
         fused_rope = self.rope_style == 0 and not c.is_qwen3
+        fused_norm = NB <= 4  # decode: norm+quant lives in the GEMV prologue
         slot = 0
+
+        def norm_gemv(lin, wn, slot, out, amax=None):
+            """normed+quantized x -> GEMV (fused prologue for decode)."""
+            if fused_norm:
+                k.q40_gemv_nq(lin.qs, lin.scales, x, wn, self.ssq[slot],
+                              c.norm_eps, out, NB, amax)
+            else:
+                k.norm_quant(x[:NB], wn, self.ssq[slot], self.xq.q[:NB],
+                             self.xq.s[:NB], self.xq.bs[:NB], NB, c.norm_eps)
+                k.q40_gemv(lin.qs, lin.scales, self.xq.q, self.xq.s,
+                           self.xq.bs, out, NB, amax)
+
         for l, lw in enumerate(self.layers):
             # attention block
-            k.norm_quant(x[:NB], lw["norm0"], self.ssq[slot], self.xq.q[:NB],
-                         self.xq.s[:NB], self.xq.bs[:NB], NB, c.norm_eps)
-            if fused_rope:
-                # GEMV epilogue rotates q/k and scatters k,v into the cache
+            if fused_rope and fused_norm:
+                k.q40_gemv_nq_rope(lw["qkv"].qs, lw["qkv"].scales, x,
+                                   lw["norm0"], self.ssq[slot], c.norm_eps,
+                                   self.qkv_out, NB, self.rope_cache, self.pos,
+                                   self.k_cache[l], self.v_cache[l],
+                                   c.q_dim0, c.kv_dim0, c.head_dim)
+            elif fused_rope:
+                k.norm_quant(x[:NB], lw["norm0"], self.ssq[slot], self.xq.q[:NB],
+                             self.xq.s[:NB], self.xq.bs[:NB], NB, c.norm_eps)
                 k.q40_gemv_rope(lw["qkv"].qs, lw["qkv"].scales, self.xq.q,
                                 self.xq.s, self.xq.bs, self.qkv_out, NB,
                                 self.rope_cache, self.pos, self.k_cache[l],
                                 self.v_cache[l], c.q_dim0, c.kv_dim0, c.head_dim)
             else:
-                k.q40_gemv(lw["qkv"].qs, lw["qkv"].scales, self.xq.q, self.xq.s,
-                           self.xq.bs, self.qkv_out, NB)
+                norm_gemv(lw["qkv"], lw["norm0"], slot, self.qkv_out)
                 if c.is_qwen3:
                     k.rmsnorm_rows_s(self.qkv_out, self.qkv_ld, 0,
                                      c.q_dim0 // c.head_dim, B, lw["q_norm"],
@@ -321,22 +339,16 @@ class HipTransformer:
                 else:
                     self._sync_partial(NB, slot + 1)
             else:
-                k.norm_quant(x[:NB], lw["norm1"], self.ssq[slot], self.xq.q[:NB],
-                             self.xq.s[:NB], self.xq.bs[:NB], NB, c.norm_eps)
-                k.q40_gemv(lw["w13"].qs, lw["w13"].scales, self.xq.q, self.xq.s,
-                           self.xq.bs, self.ff_out, NB)
+                norm_gemv(lw["w13"], lw["norm1"], slot, self.ff_out)
                 k.swiglu_q80(self.ff_out, self.ff_out[:, c.ff_dim0:],
                              2 * c.ff_dim0, c.ff_dim0, NB, self.dq.q[:NB],
                              self.dq.s[:NB], self.dq.bs[:NB])
                 self._proj_merge(lw["w2"], self.dq, slot + 1, NB)
             slot += 1
 
-        k.norm_quant(x[:NB], self.final_norm, self.ssq[slot], self.xq.q[:NB],
-                     self.xq.s[:NB], self.xq.bs[:NB], NB, c.norm_eps)
         use_amax = (self.greedy_feedback and B == 1 and c.world == 1)
-        k.q40_gemv(self.wcls.qs, self.wcls.scales, self.xq.q, self.xq.s,
-                   self.xq.bs, self.logits0, NB,
-                   self.amax_scratch if use_amax else None)
+        norm_gemv(self.wcls, self.final_norm, slot, self.logits0,
+                  self.amax_scratch if use_amax else None)
         if c.world > 1:
             self.comm.all_gather(self.logits_gather[:, :NB], self.logits0[:NB])
         if self.greedy_feedback and B == 1:

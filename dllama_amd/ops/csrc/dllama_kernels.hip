@@ -20,6 +20,7 @@
 
 #define WAVE 64
 #define QB 32  // quant block size
+#define SSQ_SPREAD 16
 
 static inline int ceil_div(int64_t a, int64_t b) { return (int)((a + b - 1) / b); }
 
@@ -55,6 +56,13 @@ __device__ __forceinline__ float group32_reduce_sum(float v) {
     for (int off = 16; off > 0; off >>= 1)
         v += __shfl_xor(v, off, 32);
     return v;
+}
+
+__device__ __forceinline__ float ssq_total(const float *ssq, int b) {
+    float t = 0.0f;
+    #pragma unroll
+    for (int k = 0; k < SSQ_SPREAD; k++) t += ssq[b * SSQ_SPREAD + k];
+    return t;
 }
 
 // ------------------------------------------------------------------ q80 quantize
@@ -283,7 +291,7 @@ __global__ void k_embed_gather(const float *__restrict__ table,
         if (threadIdx.x == 0) {
             float t = 0.0f;
             for (int i = 0; i < blockDim.x / WAVE; i++) t += red[i];
-            atomicAdd(ssq + b, t);
+            atomicAdd(ssq + b * SSQ_SPREAD + (blockIdx.x & (SSQ_SPREAD - 1)), t);
         }
     }
 }
@@ -300,7 +308,7 @@ __global__ void k_norm_quant(const float *__restrict__ x,
                              float *__restrict__ bs,
                              int n, float eps) {
     const int b = blockIdx.y;
-    const float inv = rsqrtf(ssq[b] / n + eps);
+    const float inv = rsqrtf(ssq_total(ssq, b) / n + eps);
     const int nb = n / QB;
     const int gid = blockIdx.x * blockDim.x + threadIdx.x;
     const int blk = gid / 32;
@@ -326,7 +334,7 @@ __global__ void k_norm_f32(const float *__restrict__ x,
                            float *__restrict__ y,
                            int n, float eps) {
     const int b = blockIdx.y;
-    const float inv = rsqrtf(ssq[b] / n + eps);
+    const float inv = rsqrtf(ssq_total(ssq, b) / n + eps);
     for (int i = (blockIdx.x * blockDim.x + threadIdx.x) * 4; i < n;
          i += gridDim.x * blockDim.x * 4) {
         const float4 v = *reinterpret_cast<const float4 *>(x + (int64_t)b * n + i);
@@ -359,7 +367,7 @@ __global__ void k_add_ssq(float *__restrict__ x,
     if (threadIdx.x == 0) {
         float t = 0.0f;
         for (int i = 0; i < blockDim.x / WAVE; i++) t += red[i];
-        atomicAdd(ssq + b, t);
+        atomicAdd(ssq + b * SSQ_SPREAD + (blockIdx.x & (SSQ_SPREAD - 1)), t);
     }
 }
 
@@ -420,16 +428,63 @@ __device__ __forceinline__ int q40_block_dot(const uint4 wq, const int4 x0,
 // removes whole kernels from the per-layer chain (the reference runs each
 // as its own op, llm.cpp:263-557).
 #define EPI_NONE 0
-#define EPI_RESID 1  // x[b,row] += v; atomically accumulate sum(x'^2) per b
-                     //  -> the next norm becomes a single wide pass
+#define EPI_RESID 1  // x[b,row] += v; accumulate sum(x'^2) into 16-way-spread
+                     //  slots -> the next norm becomes a single wide pass
 #define EPI_ROPE 2   // llama rope: rows (2j,2j+1) are a rotation pair held by
                      //  one RPW=2 wave; q rotated into y, k rotated into the
                      //  KV cache, v copied into the cache
 
+// ssq accumulators are spread over SSQ_SPREAD slots per batch row: a
+// single-address atomicAdd from every workgroup serializes on one cacheline
+// (measured: doubled the 4096-row GEMV).
+
+// PRO=1 prologue: the GEMV consumes the f32 residual row directly, applying
+// rmsnorm (from the precomputed ssq) and Q80-quantizing lane-locally —
+// numerically identical to the separate norm+cast kernels, but zero extra
+// launches. Each lane quantizes only the blocks it dots.
+__device__ __forceinline__ void q80_quant_block(const float *__restrict__ xsrc,
+                                                const float *__restrict__ wsrc,
+                                                const float inv,
+                                                int4 *lo, int4 *hi,
+                                                float *scale, float *bsum) {
+    float v[QB];
+    #pragma unroll
+    for (int t = 0; t < 8; t++) {
+        const float4 xv = reinterpret_cast<const float4 *>(xsrc)[t];
+        const float4 wv = reinterpret_cast<const float4 *>(wsrc)[t];
+        v[4 * t] = xv.x * inv * wv.x;
+        v[4 * t + 1] = xv.y * inv * wv.y;
+        v[4 * t + 2] = xv.z * inv * wv.z;
+        v[4 * t + 3] = xv.w * inv * wv.w;
+    }
+    float amax = 0.0f;
+    #pragma unroll
+    for (int i = 0; i < QB; i++) amax = fmaxf(amax, fabsf(v[i]));
+    const float dd = amax / 127.0f;
+    const float qinv = dd > 0.0f ? 1.0f / dd : 0.0f;
+    int words[8];
+    float bs = 0.0f;
+    #pragma unroll
+    for (int t = 0; t < 8; t++) {
+        int w = 0;
+        #pragma unroll
+        for (int e = 0; e < 4; e++) {
+            const float qf = rintf(v[4 * t + e] * qinv);
+            bs += qf;
+            w |= ((int)qf & 0xFF) << (8 * e);
+        }
+        words[t] = w;
+    }
+    *lo = make_int4(words[0], words[1], words[2], words[3]);
+    *hi = make_int4(words[4], words[5], words[6], words[7]);
+    *scale = dd;
+    *bsum = bs;
+}
+
 // RPW = rows per wave: processing 2 rows per wave doubles the independent
 // 16B weight loads in flight per lane (the decode GEMV is HBM-latency
 // limited at 1 row/wave). NB = batch columns.
-template <int NB, int RPW, int EPI>
+template <int NB, int RPW, int EPI, int PRO>
 __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
                            const __half *__restrict__ scales,
                            const int8_t *__restrict__ xq,
@@ -444,7 +499,11 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
                            const int *__restrict__ pos,
                            float *__restrict__ kc,
                            float *__restrict__ vc,
-                           int q_dim0, int kv_dim0, int hd) {
+                           int q_dim0, int kv_dim0, int hd,
+                           const float *__restrict__ xf,
+                           const float *__restrict__ wnorm,
+                           const float *__restrict__ ssq_in,
+                           float eps) {
     const int wpb = blockDim.x / WAVE;
     const int wid = threadIdx.x / WAVE;
     const int row0 = (blockIdx.x * wpb + wid) * RPW;
@@ -469,6 +528,13 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
         #pragma unroll
         for (int b = 0; b < NB; b++) acc[r][b] = 0.0f;
 
+    float invb[NB];
+    if (PRO) {
+        #pragma unroll
+        for (int b = 0; b < NB; b++)
+            invb[b] = rsqrtf(ssq_total(ssq_in, b) / n + eps);
+    }
+
     for (int jp = lane; jp < nbp; jp += WAVE) {
         const int j = jp << 1;
         uint4 wq0[RPW], wq1[RPW];
@@ -481,10 +547,20 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
         }
         #pragma unroll
         for (int b = 0; b < NB; b++) {
-            const int4 *xrow = reinterpret_cast<const int4 *>(xq + (int64_t)b * n) + j * 2;
-            const int4 x0 = xrow[0], x1 = xrow[1], x2 = xrow[2], x3 = xrow[3];
-            const float2 sx = *reinterpret_cast<const float2 *>(xs + (int64_t)b * nb + j);
-            const float2 bsum = *reinterpret_cast<const float2 *>(xbs + (int64_t)b * nb + j);
+            int4 x0, x1, x2, x3;
+            float2 sx, bsum;
+            if (PRO) {
+                q80_quant_block(xf + (int64_t)b * n + j * QB, wnorm + j * QB,
+                                invb[b], &x0, &x1, &sx.x, &bsum.x);
+                q80_quant_block(xf + (int64_t)b * n + (j + 1) * QB,
+                                wnorm + (j + 1) * QB, invb[b], &x2, &x3,
+                                &sx.y, &bsum.y);
+            } else {
+                const int4 *xrow = reinterpret_cast<const int4 *>(xq + (int64_t)b * n) + j * 2;
+                x0 = xrow[0]; x1 = xrow[1]; x2 = xrow[2]; x3 = xrow[3];
+                sx = *reinterpret_cast<const float2 *>(xs + (int64_t)b * nb + j);
+                bsum = *reinterpret_cast<const float2 *>(xbs + (int64_t)b * nb + j);
+            }
             #pragma unroll
             for (int r = 0; r < RPW; r++) {
                 const int idot0 = q40_block_dot(wq0[r], x0, x1);
@@ -502,10 +578,19 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
             const float sw1 = __half2float(srow[r][j]);
             #pragma unroll
             for (int b = 0; b < NB; b++) {
-                const int4 *xb = reinterpret_cast<const int4 *>(xq + (int64_t)b * n) + j * 2;
-                const int idot = q40_block_dot(wq, xb[0], xb[1]);
-                acc[r][b] = fmaf(sw1 * xs[(int64_t)b * nb + j],
-                                 (float)idot - 8.0f * xbs[(int64_t)b * nb + j], acc[r][b]);
+                int4 xb0, xb1;
+                float sx1, bs1;
+                if (PRO) {
+                    q80_quant_block(xf + (int64_t)b * n + j * QB, wnorm + j * QB,
+                                    invb[b], &xb0, &xb1, &sx1, &bs1);
+                } else {
+                    const int4 *xb = reinterpret_cast<const int4 *>(xq + (int64_t)b * n) + j * 2;
+                    xb0 = xb[0]; xb1 = xb[1];
+                    sx1 = xs[(int64_t)b * nb + j];
+                    bs1 = xbs[(int64_t)b * nb + j];
+                }
+                const int idot = q40_block_dot(wq, xb0, xb1);
+                acc[r][b] = fmaf(sw1 * sx1, (float)idot - 8.0f * bs1, acc[r][b]);
             }
         }
     }
@@ -579,7 +664,7 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
         if (threadIdx.x < NB) {
             float t = 0.0f;
             for (int wv = 0; wv < wpb; wv++) t += sred[wv][threadIdx.x];
-            atomicAdd(ssq + threadIdx.x, t);
+            atomicAdd(ssq + threadIdx.x * SSQ_SPREAD + (blockIdx.x & (SSQ_SPREAD - 1)), t);
         }
     }
 }
@@ -945,7 +1030,7 @@ __global__ void k_merge_add(float *__restrict__ x,
         if (threadIdx.x == 0) {
             float t = 0.0f;
             for (int i = 0; i < blockDim.x / WAVE; i++) t += red[i];
-            atomicAdd(ssq + r, t);
+            atomicAdd(ssq + r * SSQ_SPREAD + (blockIdx.x & (SSQ_SPREAD - 1)), t);
         }
     }
 }
@@ -1019,28 +1104,36 @@ struct GemvEpi {
     float *vc = nullptr;
     int q_dim0 = 0, kv_dim0 = 0, hd = 0;
     bool force_rpw2 = false;
+    // PRO=1 (fused norm+quant prologue) inputs
+    const float *xf = nullptr;
+    const float *wnorm = nullptr;
+    const float *ssq_in = nullptr;
+    float eps = 0.0f;
 };
 
-template <int EPI>
+template <int EPI, int PRO>
 static void gemv_launch(torch::Tensor &qs, torch::Tensor &scales, torch::Tensor &xq,
                         torch::Tensor &xs, torch::Tensor &xbs, float *y,
                         int64_t batch, const GemvEpi &e) {
     const int d = qs.size(0);
     const int n = qs.size(1) * 2;
-    TORCH_CHECK(xq.size(-1) == n, "x width mismatch");
+    TORCH_CHECK(PRO == 1 || xq.size(-1) == n, "x width mismatch");
     TORCH_CHECK(n % QB == 0, "n must be a multiple of 32");
     const int waves_per_block = 4;
     const dim3 block(waves_per_block * WAVE);
     auto launch = [&](auto nb_const, auto rpw_const) {
         constexpr int RPW = decltype(rpw_const)::value;
         const dim3 grid(ceil_div(d, waves_per_block * RPW));
-        hipLaunchKernelGGL((k_q40_gemv<decltype(nb_const)::value, RPW, EPI>), grid,
+        hipLaunchKernelGGL((k_q40_gemv<decltype(nb_const)::value, RPW, EPI, PRO>), grid,
                            block, 0, cur_stream(),
                            qs.data_ptr<uint8_t>(),
                            reinterpret_cast<const __half *>(scales.data_ptr<at::Half>()),
-                           xq.data_ptr<int8_t>(), xs.data_ptr<float>(),
-                           xbs.data_ptr<float>(), y, d, n, e.slot, e.x_resid, e.ssq,
-                           e.cache, e.pos, e.kc, e.vc, e.q_dim0, e.kv_dim0, e.hd);
+                           PRO ? nullptr : xq.data_ptr<int8_t>(),
+                           PRO ? nullptr : xs.data_ptr<float>(),
+                           PRO ? nullptr : xbs.data_ptr<float>(), y, d, n,
+                           e.slot, e.x_resid, e.ssq,
+                           e.cache, e.pos, e.kc, e.vc, e.q_dim0, e.kv_dim0, e.hd,
+                           e.xf, e.wnorm, e.ssq_in, e.eps);
     };
     std::integral_constant<int, 1> r1;
     std::integral_constant<int, 2> r2;
@@ -1081,7 +1174,7 @@ void q40_gemv(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
     GemvEpi e;
     if (amax_slot.has_value())
         e.slot = reinterpret_cast<unsigned long long *>(amax_slot->data_ptr<int64_t>());
-    gemv_launch<EPI_NONE>(qs, scales, xq, xs, xbs, y.data_ptr<float>(), batch, e);
+    gemv_launch<EPI_NONE, 0>(qs, scales, xq, xs, xbs, y.data_ptr<float>(), batch, e);
 }
 
 void q40_gemv_resid(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
@@ -1092,7 +1185,7 @@ void q40_gemv_resid(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
     GemvEpi e;
     e.x_resid = x.data_ptr<float>();
     e.ssq = ssq.data_ptr<float>();
-    gemv_launch<EPI_RESID>(qs, scales, xq, xs, xbs, x.data_ptr<float>(), batch, e);
+    gemv_launch<EPI_RESID, 0>(qs, scales, xq, xs, xbs, x.data_ptr<float>(), batch, e);
 }
 
 void q40_gemv_rope(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
@@ -1110,7 +1203,47 @@ void q40_gemv_rope(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
     e.kv_dim0 = (int)kv_dim0;
     e.hd = (int)head_dim;
     e.force_rpw2 = true;  // rotation pairs live in one RPW=2 wave
-    gemv_launch<EPI_ROPE>(qs, scales, xq, xs, xbs, y.data_ptr<float>(), batch, e);
+    gemv_launch<EPI_ROPE, 0>(qs, scales, xq, xs, xbs, y.data_ptr<float>(), batch, e);
+}
+
+void q40_gemv_nq(torch::Tensor qs, torch::Tensor scales, torch::Tensor x,
+                 torch::Tensor wnorm, torch::Tensor ssq, double eps,
+                 torch::Tensor y, int64_t batch,
+                 c10::optional<torch::Tensor> amax_slot = c10::nullopt) {
+    // GEMV with fused rmsnorm+Q80-quantize prologue (decode batches)
+    CHECK_CUDA(qs); CHECK_CONT(qs);
+    TORCH_CHECK(batch <= 4, "PRO prologue is for small (decode) batches");
+    GemvEpi e;
+    e.xf = x.data_ptr<float>();
+    e.wnorm = wnorm.data_ptr<float>();
+    e.ssq_in = ssq.data_ptr<float>();
+    e.eps = (float)eps;
+    if (amax_slot.has_value())
+        e.slot = reinterpret_cast<unsigned long long *>(amax_slot->data_ptr<int64_t>());
+    gemv_launch<EPI_NONE, 1>(qs, scales, x, x, x, y.data_ptr<float>(), batch, e);
+}
+
+void q40_gemv_nq_rope(torch::Tensor qs, torch::Tensor scales, torch::Tensor x,
+                      torch::Tensor wnorm, torch::Tensor ssq, double eps,
+                      torch::Tensor y, int64_t batch, torch::Tensor cache,
+                      torch::Tensor pos, torch::Tensor kc, torch::Tensor vc,
+                      int64_t q_dim0, int64_t kv_dim0, int64_t head_dim) {
+    CHECK_CUDA(qs); CHECK_CONT(qs);
+    TORCH_CHECK(batch <= 4, "PRO prologue is for small (decode) batches");
+    GemvEpi e;
+    e.xf = x.data_ptr<float>();
+    e.wnorm = wnorm.data_ptr<float>();
+    e.ssq_in = ssq.data_ptr<float>();
+    e.eps = (float)eps;
+    e.cache = cache.data_ptr<float>();
+    e.pos = pos.data_ptr<int>();
+    e.kc = kc.data_ptr<float>();
+    e.vc = vc.data_ptr<float>();
+    e.q_dim0 = (int)q_dim0;
+    e.kv_dim0 = (int)kv_dim0;
+    e.hd = (int)head_dim;
+    e.force_rpw2 = true;
+    gemv_launch<EPI_ROPE, 1>(qs, scales, x, x, x, y.data_ptr<float>(), batch, e);
 }
 
 int64_t q40_gemv_argmax_blocks(int64_t d) {
@@ -1377,6 +1510,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("amax_slot") = py::none());
     m.def("q40_gemv_resid", &q40_gemv_resid);
     m.def("q40_gemv_rope", &q40_gemv_rope);
+    m.def("q40_gemv_nq", &q40_gemv_nq, py::arg("qs"), py::arg("scales"),
+          py::arg("x"), py::arg("wnorm"), py::arg("ssq"), py::arg("eps"),
+          py::arg("y"), py::arg("batch"), py::arg("amax_slot") = py::none());
+    m.def("q40_gemv_nq_rope", &q40_gemv_nq_rope);
     m.def("norm_quant", &norm_quant);
     m.def("norm_f32", &norm_f32);
     m.def("add_ssq", &add_ssq);
