@@ -281,7 +281,10 @@ class HipTransformer:
 
         kv_mul = c.n_heads0 // max(1, c.kv_dim0 // c.head_dim)
         fused_rope = self.rope_style == 0 and not c.is_qwen3
-        fused_norm = NB <= 4  # decode: norm+quant lives in the GEMV prologue
+        # Fusing norm+quant into the GEMV prologue re-quantizes x per
+        # workgroup: measured VALU-bound (2x slower on the big GEMVs).
+        # Kept behind a flag for shapes where it might win; off by default.
+        fused_norm = getattr(self, "use_fused_norm", False) and NB <= 4
         slot = 0
 
         def norm_gemv(lin, wn, slot, out, amax=None):
