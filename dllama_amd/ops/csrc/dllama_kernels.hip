@@ -58,6 +58,13 @@ __device__ __forceinline__ float group32_reduce_sum(float v) {
     return v;
 }
 
+__device__ __forceinline__ float group16_reduce_sum(float v) {
+    #pragma unroll
+    for (int off = 8; off > 0; off >>= 1)
+        v += __shfl_xor(v, off, 16);
+    return v;
+}
+
 __device__ __forceinline__ float ssq_total(const float *ssq, int b) {
     float t = 0.0f;
     #pragma unroll
@@ -691,28 +698,52 @@ __global__ void k_attn_split(const float *__restrict__ q, int q_ld,
     const int lane = threadIdx.x % WAVE;
     const int kv_off = (h0 / kv_mul) * hd;
 
-    float qreg[VEC];
+    // Each wave's 4 groups of 16 lanes score 4 timesteps per iteration:
+    // the serial online-softmax chain advances once per 4 t (a 1-t-per-wave
+    // loop measured 22 us at pos=1024 — latency-chained, not memory-bound).
+    const int VEC16 = VEC * 4;  // q/k elems per lane within a 16-lane group
+    const int lane16 = lane & 15;
+    const int group = lane >> 4;
+
+    float qreg[VEC16];
     #pragma unroll
-    for (int v = 0; v < VEC; v++)
-        qreg[v] = q[(int64_t)b * q_ld + h0 * hd + lane * VEC + v] * scale;
+    for (int v = 0; v < VEC16; v++)
+        qreg[v] = q[(int64_t)b * q_ld + h0 * hd + lane16 * VEC16 + v] * scale;
 
     float m = -1e30f, l = 0.0f, o[VEC];
     #pragma unroll
     for (int v = 0; v < VEC; v++) o[v] = 0.0f;
 
-    for (int t = sp * 4 + wave; t < plen; t += 4 * S) {
-        const float *krow = kc + (int64_t)t * kv_dim0 + kv_off + lane * VEC;
+    for (int tb = (sp * 4 + wave) * 4; tb < plen; tb += 16 * S) {
+        const int tg = tb + group;  // this group's timestep
         float partial = 0.0f;
+        if (tg < plen) {
+            const float *krow = kc + (int64_t)tg * kv_dim0 + kv_off + lane16 * VEC16;
+            #pragma unroll
+            for (int v = 0; v < VEC16; v++) partial = fmaf(qreg[v], krow[v], partial);
+        }
+        const float sg = group16_reduce_sum(partial);
+        float s4[4];
         #pragma unroll
-        for (int v = 0; v < VEC; v++) partial = fmaf(qreg[v], krow[v], partial);
-        const float s = wave_reduce_sum(partial);
-        const float mn = fmaxf(m, s);
+        for (int gg = 0; gg < 4; gg++) {
+            s4[gg] = __shfl(sg, gg * 16, WAVE);
+            if (tb + gg >= plen) s4[gg] = -1e30f;
+        }
+        const float mn = fmaxf(fmaxf(fmaxf(m, s4[0]), fmaxf(s4[1], s4[2])), s4[3]);
         const float f = __expf(m - mn);
-        const float w = __expf(s - mn);
-        const float *vrow = vc + (int64_t)t * kv_dim0 + kv_off + lane * VEC;
-        l = l * f + w;
+        float w4[4];
         #pragma unroll
-        for (int v = 0; v < VEC; v++) o[v] = fmaf(o[v], f, w * vrow[v]);
+        for (int gg = 0; gg < 4; gg++) w4[gg] = __expf(s4[gg] - mn);
+        l = l * f + w4[0] + w4[1] + w4[2] + w4[3];
+        #pragma unroll
+        for (int v = 0; v < VEC; v++) o[v] *= f;
+        #pragma unroll
+        for (int gg = 0; gg < 4; gg++) {
+            if (tb + gg >= plen) break;
+            const float *vrow = vc + (int64_t)(tb + gg) * kv_dim0 + kv_off + lane * VEC;
+            #pragma unroll
+            for (int v = 0; v < VEC; v++) o[v] = fmaf(w4[gg], vrow[v], o[v]);
+        }
         m = mn;
     }
 
