@@ -209,10 +209,11 @@ class HipTransformer:
         # sum-of-squares accumulators: [slot, batch, 16-way spread]
         self.ssq = torch.zeros(2 * c.n_layers + 1, NB, 16, device=dev)
         self.amax_scratch = torch.zeros(self.amax_blocks, dtype=torch.int64, device=dev)
-        self.attn_splits = 8
+        self.attn_splits = 32  # fills the chip (H0*S workgroups) at long pos
         self.attn_ml = torch.zeros(NB * c.n_heads0 * self.attn_splits * 2, device=dev)
         self.attn_o = torch.zeros(NB * c.n_heads0 * self.attn_splits * c.head_dim,
                                   device=dev)
+        self.attn_counter = torch.zeros(NB * c.n_heads0, dtype=torch.int32, device=dev)
         if c.world > 1:
             self.logits_gather = torch.zeros(c.world, NB, c.vocab0, device=dev)
             if c.sync_type == Q80:
@@ -327,7 +328,7 @@ class HipTransformer:
             k.attn(self.qkv_out, self.qkv_ld, self.k_cache[l], self.v_cache[l],
                    self.zbuf[:B], self.pos, B, c.n_heads0, kv_mul, c.head_dim,
                    self.attn_splits, self.attn_ml, self.attn_o,
-                   self.zq.q, self.zq.s, self.zq.bs)
+                   self.attn_counter, self.zq.q, self.zq.s, self.zq.bs)
             self._proj_merge(lw["wo"], self.zq, slot + 1, NB)
             slot += 1
 

@@ -154,7 +154,8 @@ def test_kv_append_and_attn(k):
     S = 8
     ml = torch.zeros(B * H0 * S * 2, device=DEV)
     osc = torch.zeros(B * H0 * S * hd, device=DEV)
-    k.attn(q, H0 * hd, kc, vc, y, pos, B, H0, H0 // n_kv0, hd, S, ml, osc)
+    cnt = torch.zeros(B * H0, dtype=torch.int32, device=DEV)
+    k.attn(q, H0 * hd, kc, vc, y, pos, B, H0, H0 // n_kv0, hd, S, ml, osc, cnt)
     want = R.attention(q.cpu(), kc.cpu(), vc.cpu(), torch.tensor([4, 5]), H0, hd)
     assert torch.allclose(y.cpu(), want, atol=1e-4, rtol=1e-3), \
         (y.cpu() - want).abs().max().item()
@@ -172,7 +173,8 @@ def test_attn_long_context(k):
     S = 8
     ml = torch.zeros(B * H0 * S * 2, device=DEV)
     osc = torch.zeros(B * H0 * S * hd, device=DEV)
-    k.attn(q, H0 * hd, kc, vc, y, pos, B, H0, H0 // n_kv0, hd, S, ml, osc)
+    cnt = torch.zeros(B * H0, dtype=torch.int32, device=DEV)
+    k.attn(q, H0 * hd, kc, vc, y, pos, B, H0, H0 // n_kv0, hd, S, ml, osc, cnt)
     want = R.attention(q.cpu(), kc.cpu(), vc.cpu(), torch.tensor([298]), H0, hd)
     assert torch.allclose(y.cpu(), want, atol=1e-4, rtol=1e-3)
 

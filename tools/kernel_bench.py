@@ -73,14 +73,18 @@ vc = torch.randn(seq, kvd, device=dev, generator=g)
 qv = torch.randn(1, H0 * hd, device=dev, generator=g)
 z = torch.zeros(1, H0 * hd, device=dev)
 pos = torch.tensor([1024], dtype=torch.int32, device=dev)
-S = 8
+S = 32
 ml = torch.zeros(1 * H0 * S * 2, device=dev)
 osc = torch.zeros(1 * H0 * S * hd, device=dev)
+cnt = torch.zeros(1 * H0, dtype=torch.int32, device=dev)
 zq3 = mkx(1, H0 * hd)
-print("attn(pos=1024) split+combine+quant: %6.2f us" %
+print("attn(pos=1024) fused+quant: %6.2f us" %
       bench(lambda: k.attn(qv, H0 * hd, kc, vc, z, pos, 1, H0, 4, hd, S, ml, osc,
-                           zq3[0], zq3[1], zq3[2])))
+                           cnt, zq3[0], zq3[1], zq3[2])))
 pos2 = torch.tensor([64], dtype=torch.int32, device=dev)
-print("attn(pos=64)  split+combine+quant: %6.2f us" %
+print("attn(pos=64)  fused+quant: %6.2f us" %
       bench(lambda: k.attn(qv, H0 * hd, kc, vc, z, pos2, 1, H0, 4, hd, S, ml, osc,
-                           zq3[0], zq3[1], zq3[2])))
+                           cnt, zq3[0], zq3[1], zq3[2])))
+xr = torch.randn(1, dim, device=dev, generator=g)
+pr = torch.randn(1, dim, device=dev, generator=g)
+print("add_ssq [1,4096]: %6.2f us" % bench(lambda: k.add_ssq(xr, pr, ssq[0], 1)))
