@@ -693,6 +693,7 @@ __global__ void k_attn_split(const float *__restrict__ q, int q_ld,
                              int8_t *__restrict__ zq,
                              float *__restrict__ zs,
                              float *__restrict__ zbs) {
+    (void)counter; (void)y; (void)zq; (void)zs; (void)zbs;
     const int h0 = blockIdx.x;
     const int b = blockIdx.y;
     const int sp = blockIdx.z;
@@ -776,43 +777,51 @@ __global__ void k_attn_split(const float *__restrict__ q, int q_ld,
         }
     }
 
-    // Last workgroup per (b, head) combines the S split partials in place —
-    // no separate combine kernel (one fewer launch + no global barrier).
-    __shared__ int is_last;
-    if (threadIdx.x == 0) {
-        __threadfence();
-        is_last = (atomicAdd(&counter[b * n_heads0 + h0], 1) == S - 1) ? 1 : 0;
-    }
-    __syncthreads();
-    if (!is_last) return;
-    if (threadIdx.x == 0) counter[b * n_heads0 + h0] = 0;  // reset for next step
-    const int i = threadIdx.x;
-    if (i >= hd) return;
+}
+
+// Stage 2: combine the S split partials. QUANT=true additionally emits the
+// Q80 triple of the attention output directly (each workgroup owns a whole
+// head = hd/32 quant blocks), eliminating the separate cast kernel the
+// reference runs before the wo matmul. (A fused last-block-combine variant
+// measured 2x SLOWER: per-workgroup threadfence + tail serialization.)
+template <int VEC, bool QUANT>
+__global__ void k_attn_combine(const float *__restrict__ ml_scratch,
+                               const float *__restrict__ o_scratch,
+                               float *__restrict__ y,
+                               int8_t *__restrict__ zq,
+                               float *__restrict__ zs,
+                               float *__restrict__ zbs,
+                               int n_heads0, int S) {
+    const int h0 = blockIdx.x;
+    const int b = blockIdx.y;
+    const int hd = VEC * WAVE;  // blockDim.x == hd
     const int64_t base = ((int64_t)b * n_heads0 + h0) * S;
-    float Mg = -1e30f;
-    for (int s2 = 0; s2 < S; s2++)
-        Mg = fmaxf(Mg, ml_scratch[(base + s2) * 2]);
-    float Lg = 0.0f;
-    for (int s2 = 0; s2 < S; s2++)
-        Lg += ml_scratch[(base + s2) * 2 + 1] * __expf(ml_scratch[(base + s2) * 2] - Mg);
-    const float invL = 1.0f / Lg;
-    float acc2 = 0.0f;
-    for (int s2 = 0; s2 < S; s2++)
-        acc2 += o_scratch[(base + s2) * hd + i] * __expf(ml_scratch[(base + s2) * 2] - Mg);
-    const float val = acc2 * invL;
+    float M = -1e30f;
+    for (int sp = 0; sp < S; sp++)
+        M = fmaxf(M, ml_scratch[(base + sp) * 2]);
+    float L = 0.0f;
+    for (int sp = 0; sp < S; sp++)
+        L += ml_scratch[(base + sp) * 2 + 1] * __expf(ml_scratch[(base + sp) * 2] - M);
+    const float invL = 1.0f / L;
+    const int i = threadIdx.x;
+    float acc = 0.0f;
+    for (int sp = 0; sp < S; sp++)
+        acc += o_scratch[(base + sp) * hd + i]
+             * __expf(ml_scratch[(base + sp) * 2] - M);
+    const float v = acc * invL;
     if (!QUANT) {
-        y[((int64_t)b * n_heads0 + h0) * hd + i] = val;
+        y[((int64_t)b * n_heads0 + h0) * hd + i] = v;
     } else {
-        const float amax2 = group32_reduce_max(fabsf(val));
-        const float dd = amax2 / 127.0f;
-        const float qinv = dd > 0.0f ? 1.0f / dd : 0.0f;
-        const float qf = rintf(val * qinv);
+        const float amax = group32_reduce_max(fabsf(v));
+        const float d = amax / 127.0f;
+        const float qinv = d > 0.0f ? 1.0f / d : 0.0f;
+        const float qf = rintf(v * qinv);
         zq[((int64_t)b * n_heads0 + h0) * hd + i] = (int8_t)qf;
-        const float bsum2 = group32_reduce_sum(qf);
+        const float bsum = group32_reduce_sum(qf);
         if ((i & 31) == 0) {
             const int blk = (h0 * hd + i) / QB;
-            zs[(int64_t)b * (n_heads0 * hd / QB) + blk] = dd;
-            zbs[(int64_t)b * (n_heads0 * hd / QB) + blk] = bsum2;
+            zs[(int64_t)b * (n_heads0 * hd / QB) + blk] = d;
+            zbs[(int64_t)b * (n_heads0 * hd / QB) + blk] = bsum;
         }
     }
 }
@@ -1339,32 +1348,31 @@ void attn(torch::Tensor q, int64_t q_ld, torch::Tensor kc, torch::Tensor vc,
     const int kv_dim0 = kc.size(1);
     const float scale = 1.0f / sqrtf((float)head_dim);
     const dim3 grid(n_heads0, batch, splits);
+    const dim3 cgrid(n_heads0, batch);
     const bool quant = zq.has_value();
-    auto run = [&](auto vec_const, auto quant_const) {
+    auto run = [&](auto vec_const) {
         constexpr int V = decltype(vec_const)::value;
-        constexpr bool Q = decltype(quant_const)::value;
-        hipLaunchKernelGGL((k_attn_split<V, Q>), grid, dim3(256), 0, cur_stream(),
+        hipLaunchKernelGGL((k_attn_split<V, false>), grid, dim3(256), 0, cur_stream(),
                            q.data_ptr<float>(), (int)q_ld, kc.data_ptr<float>(),
                            vc.data_ptr<float>(), pos.data_ptr<int>(),
                            (int)n_heads0, (int)kv_mul, kv_dim0, scale,
                            ml_scratch.data_ptr<float>(), o_scratch.data_ptr<float>(),
-                           counter.data_ptr<int>(),
-                           Q ? nullptr : y.data_ptr<float>(),
-                           Q ? zq->data_ptr<int8_t>() : nullptr,
-                           Q ? zs->data_ptr<float>() : nullptr,
-                           Q ? zbs->data_ptr<float>() : nullptr);
+                           counter.data_ptr<int>(), nullptr, nullptr, nullptr, nullptr);
+        if (quant)
+            hipLaunchKernelGGL((k_attn_combine<V, true>), cgrid, dim3(V * WAVE), 0,
+                               cur_stream(), ml_scratch.data_ptr<float>(),
+                               o_scratch.data_ptr<float>(), nullptr,
+                               zq->data_ptr<int8_t>(), zs->data_ptr<float>(),
+                               zbs->data_ptr<float>(), (int)n_heads0, (int)splits);
+        else
+            hipLaunchKernelGGL((k_attn_combine<V, false>), cgrid, dim3(V * WAVE), 0,
+                               cur_stream(), ml_scratch.data_ptr<float>(),
+                               o_scratch.data_ptr<float>(), y.data_ptr<float>(),
+                               nullptr, nullptr, nullptr, (int)n_heads0, (int)splits);
     };
-    std::integral_constant<bool, true> qt;
-    std::integral_constant<bool, false> qf;
-    if (head_dim == 128) {
-        if (quant) run(std::integral_constant<int, 2>{}, qt);
-        else run(std::integral_constant<int, 2>{}, qf);
-    } else if (head_dim == 64) {
-        if (quant) run(std::integral_constant<int, 1>{}, qt);
-        else run(std::integral_constant<int, 1>{}, qf);
-    } else {
-        TORCH_CHECK(false, "unsupported head_dim ", head_dim);
-    }
+    if (head_dim == 128) run(std::integral_constant<int, 2>{});
+    else if (head_dim == 64) run(std::integral_constant<int, 1>{});
+    else TORCH_CHECK(false, "unsupported head_dim ", head_dim);
 }
 
 void embed_gather(torch::Tensor table, torch::Tensor tokens, torch::Tensor x,
