@@ -209,7 +209,7 @@ class HipTransformer:
         # sum-of-squares accumulators: [slot, batch, 16-way spread]
         self.ssq = torch.zeros(2 * c.n_layers + 1, NB, 16, device=dev)
         self.amax_scratch = torch.zeros(self.amax_blocks, dtype=torch.int64, device=dev)
-        self.attn_splits = 16  # H0*S workgroups; balances chip fill vs combine cost
+        self.attn_splits = 8  # S=16,32 measured slower (combine reads S partials)
         self.attn_ml = torch.zeros(NB * c.n_heads0 * self.attn_splits * 2, device=dev)
         self.attn_o = torch.zeros(NB * c.n_heads0 * self.attn_splits * c.head_dim,
                                   device=dev)
