@@ -63,7 +63,7 @@ class DistComm(Comm):
         return x
 
     def all_gather(self, out, x):
-        dist.all_gather_into_tensor(out, x, group=self.group)
+        dist.all_gather_into_tensor(out.view(-1), x.reshape(-1), group=self.group)
         return out
 
     def broadcast_(self, x, src=0):
