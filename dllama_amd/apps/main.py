@@ -1,0 +1,229 @@
+"""`dllama` CLI — inference / chat / perplexity / worker modes.
+
+CLI parity with the reference (src/app.cpp:24-135, src/dllama.cpp): same
+flags, same 🔶/🔷 per-run stats output (dllama.cpp:104-115). On MI355X the
+reference's root+workers become torchrun ranks on one node:
+
+  # 1 GPU
+  dllama inference --model m.m --tokenizer t.t --prompt "Hello" --steps 64
+  # 8 GPUs (replaces root + 7 TCP workers; reference README "How to run")
+  torchrun --nproc-per-node 8 --master-addr 127.0.0.1 -m dllama_amd.apps.main \
+      inference --model m.m --tokenizer t.t --prompt "Hello" --steps 64
+"""
+
+from __future__ import annotations
+
+import argparse
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+from .. import model_file as mf
+from ..engine import InferenceEngine
+from ..models.config import ModelConfig
+from ..parallel.comm import SingleComm, init_dist_comm
+from ..quants import F32, Q80
+from ..tokenizer import (ChatItem, ChatTemplateGenerator, EosDetector, Sampler,
+                         TEMPLATE_UNKNOWN, Tokenizer, _TEMPLATE_NAMES, chat_stops)
+
+
+def build_parser() -> argparse.ArgumentParser:
+    p = argparse.ArgumentParser(prog="dllama")
+    p.add_argument("mode", choices=["inference", "chat", "perplexity", "worker"])
+    p.add_argument("--model", required=False)
+    p.add_argument("--tokenizer", required=False)
+    p.add_argument("--prompt", default=None)
+    p.add_argument("--steps", type=int, default=64)
+    p.add_argument("--buffer-float-type", default="q80", choices=["q80", "f32"],
+                   help="TP sync buffer quantization (reference --buffer-float-type)")
+    p.add_argument("--nthreads", type=int, default=os.cpu_count(),
+                   help="CPU backend threads (reference --nthreads)")
+    p.add_argument("--temperature", type=float, default=0.8)
+    p.add_argument("--topp", type=float, default=0.9)
+    p.add_argument("--seed", type=int, default=None)
+    p.add_argument("--chat-template", default=None,
+                   choices=[None, *_TEMPLATE_NAMES.keys()])
+    p.add_argument("--max-seq-len", type=int, default=0)
+    p.add_argument("--gpu-index", type=int, default=None,
+                   help="GPU to use (default LOCAL_RANK); -1 forces CPU")
+    p.add_argument("--n-batches", type=int, default=32,
+                   help="max prompt tokens per prefill step (reference nBatches)")
+    p.add_argument("--no-graph", action="store_true",
+                   help="disable hipGraph decode capture")
+    # accepted for reference CLI parity; meaningless on one xGMI node
+    p.add_argument("--workers", nargs="*", default=None,
+                   help="ignored: TP ranks come from torchrun (xGMI, not TCP)")
+    p.add_argument("--host", default="0.0.0.0")
+    p.add_argument("--port", type=int, default=9990)
+    p.add_argument("--net-turbo", type=int, default=1, help="ignored (no TCP mesh)")
+    return p
+
+
+def load_engine(args):
+    """Model + tokenizer + engine with the right backend and TP setup."""
+    comm = init_dist_comm()
+    sync = Q80 if args.buffer_float_type == "q80" else F32
+    m = mf.ModelFile(args.model, max_seq_len=args.max_seq_len, sync_type=sync)
+    use_gpu = torch.cuda.is_available() and args.gpu_index != -1
+    cfg = ModelConfig.from_header(m.header, world=comm.world, rank=comm.rank)
+    if comm.rank == 0:
+        _print_header(m.header, cfg, use_gpu)
+    if use_gpu:
+        from ..models.hip_model import HipTransformer
+        dev = args.gpu_index if args.gpu_index is not None \
+            else int(os.environ.get("LOCAL_RANK", "0"))
+        model = HipTransformer.from_file(m, cfg, device=f"cuda:{dev}", comm=comm,
+                                         n_batches=args.n_batches)
+        if not args.no_graph and comm.world == 1:
+            model.capture_decode_graph()
+    else:
+        from ..models.cpu_model import CpuTransformer
+        torch.set_num_threads(max(1, args.nthreads))
+        model = CpuTransformer(m, cfg, comm)
+    tok = Tokenizer(args.tokenizer) if args.tokenizer else None
+    seed = args.seed if args.seed is not None else int(time.time())
+    sampler = Sampler(m.header.vocab_size, args.temperature, args.topp, seed)
+    return InferenceEngine(model, tok, sampler, n_batches=args.n_batches), m, comm
+
+
+def _print_header(h, cfg, use_gpu):
+    print(f"💡 Arch: {'Llama' if h.arch_type == mf.ARCH_LLAMA else 'Qwen3 MoE' if h.arch_type == mf.ARCH_QWEN3_MOE else 'Qwen3'}")
+    print(f"💡 Dim: {h.dim}\n💡 HiddenDim: {h.hidden_dim}\n💡 nLayers: {h.n_layers}")
+    print(f"💡 nHeads: {h.n_heads}\n💡 nKvHeads: {h.n_kv_heads}\n💡 HeadDim: {h.head_dim}")
+    print(f"💡 VocabSize: {h.vocab_size}\n💡 SeqLen: {h.seq_len}")
+    if h.n_experts:
+        print(f"💡 nExperts: {h.n_experts}\n💡 nActiveExperts: {h.n_active_experts}")
+    print(f"💡 Backend: {'MI355X HIP' if use_gpu else 'CPU'}  TP={cfg.world}")
+
+
+def run_inference(args) -> int:
+    engine, m, comm = load_engine(args)
+    if args.prompt is None:
+        print("error: missing --prompt", file=sys.stderr)
+        return 1
+    tok = engine.tokenizer
+    tokens = tok.encode(args.prompt) if tok else [int(t) for t in args.prompt.split()]
+    quiet = comm.rank != 0
+
+    pieces = []
+    if tok:
+        tok.reset_decoder()
+
+    def on_token(t):
+        if quiet or not tok:
+            return
+        piece = tok.decode(t)
+        if piece:
+            pieces.append(piece)
+            print(piece, end="", flush=True)
+
+    if not quiet:
+        print(args.prompt, end="", flush=True)
+    out, stats = engine.generate(tokens, args.steps, on_token=on_token,
+                                 stop_check=(tok.is_eos if tok else None))
+    if not quiet:
+        print()
+        # reference per-run stats format (dllama.cpp:104-115)
+        print(f"Evaluation\n   nBatches: {args.n_batches}\n   nTokens: {stats.prefill_tokens}\n"
+              f"   tokens/s: {stats.eval_tok_s:.2f} ({1000.0 / max(stats.eval_tok_s, 1e-9):.2f} ms/tok)")
+        print(f"Prediction\n   nTokens: {stats.decode_tokens}\n"
+              f"   tokens/s: {stats.pred_tok_s:.2f} ({1000.0 / max(stats.pred_tok_s, 1e-9):.2f} ms/tok)")
+    return 0
+
+
+def run_chat(args) -> int:
+    """Interactive chat REPL (reference dllama.cpp:174-258)."""
+    engine, m, comm = load_engine(args)
+    tok = engine.tokenizer
+    ttype = _TEMPLATE_NAMES.get(args.chat_template, TEMPLATE_UNKNOWN) \
+        if args.chat_template else TEMPLATE_UNKNOWN
+    eos_piece = tok.vocab[tok.eos_token_ids[0]].decode("utf-8", "replace") \
+        if tok.eos_token_ids else ""
+    gen = ChatTemplateGenerator(ttype, tok.chat_template, eos_piece)
+    stops = chat_stops(tok)
+    quiet = comm.rank != 0
+    is_start = True
+    while True:
+        try:
+            user = input("\n💻 > ") if not quiet else input()
+        except EOFError:
+            return 0
+        if not user.strip():
+            continue
+        items = [ChatItem("user", user)]
+        text = gen.generate(items, True).content
+        tokens = tok.encode(text, is_start=is_start)
+        is_start = False
+        detector = EosDetector(tok.eos_token_ids, stops)
+        tok.reset_decoder()
+        if not quiet:
+            print("🤖 ", end="", flush=True)
+
+        def on_token(t):
+            piece = tok.piece(t).decode("utf-8", "replace")
+            kind = detector.append(t, piece)
+            if not quiet:
+                delta = detector.get_delta()
+                if delta and kind != 0:  # not MAYBE_EOS
+                    print(delta, end="", flush=True)
+                    detector.reset()
+
+        engine.generate(tokens, args.steps, on_token=on_token,
+                        stop_check=lambda t: detector.is_eos(t)
+                        or detector.eos_pos >= 0)
+
+
+def run_perplexity(args) -> int:
+    """Perplexity over --prompt (reference dllama.cpp:132-172)."""
+    engine, m, comm = load_engine(args)
+    tok = engine.tokenizer
+    tokens = tok.encode(args.prompt) if tok else [int(t) for t in args.prompt.split()]
+    if len(tokens) < 2:
+        print("error: need at least 2 tokens", file=sys.stderr)
+        return 1
+    nll, count = 0.0, 0
+    pos = 0
+    for i in range(0, len(tokens) - 1, args.n_batches):
+        chunk = tokens[i: i + args.n_batches]
+        t = torch.tensor(chunk, dtype=torch.int64)
+        p = torch.arange(pos, pos + len(chunk), dtype=torch.int64)
+        logits = engine.model.forward(t, p)  # [B, vocab]
+        logits = logits.detach().float().cpu()
+        for j in range(len(chunk)):
+            nxt_idx = i + j + 1
+            if nxt_idx >= len(tokens):
+                break
+            row = logits[j]
+            logp = row - torch.logsumexp(row, dim=-1)
+            nll -= float(logp[tokens[nxt_idx]])
+            count += 1
+        pos += len(chunk)
+    ppl = float(np.exp(nll / count))
+    if comm.rank == 0:
+        print(f"Perplexity: {ppl:.4f}  (nll/token {nll / count:.4f}, {count} tokens)")
+    return 0
+
+
+def main(argv=None) -> int:
+    args = build_parser().parse_args(argv)
+    if args.mode == "worker":
+        print("ℹ️  On MI355X, workers are torchrun ranks on one node — run:\n"
+              "   torchrun --nproc-per-node N --master-addr 127.0.0.1 "
+              "-m dllama_amd.apps.main inference ...\n"
+              "   (replaces the reference's TCP root+worker mesh; "
+              "non-zero ranks stay silent)")
+        return 0
+    if args.mode == "inference":
+        return run_inference(args)
+    if args.mode == "chat":
+        return run_chat(args)
+    if args.mode == "perplexity":
+        return run_perplexity(args)
+    return 1
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -170,7 +170,7 @@ class Tokenizer:
     def decode(self, token: int) -> str | None:
         """Streaming decode of one token; returns printable text or None
         while mid-UTF-8-sequence (reference src/tokenizer.cpp:291-309)."""
-        if token == self.bos_id:
+        if token == self.bos_id or token >= self.vocab_size:
             return None
         if self.is_eos(token):
             if self._decode_buf:
@@ -197,7 +197,7 @@ class Tokenizer:
         return out if out else None
 
     def piece(self, token: int) -> bytes:
-        return self.vocab[token]
+        return self.vocab[token] if 0 <= token < self.vocab_size else b""
 
 
 def write_tokenizer(path: str, vocab: list[bytes], scores, bos_id: int,

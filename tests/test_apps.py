@@ -1,0 +1,151 @@
+"""End-to-end app tests on CPU with tiny synthetic model + byte tokenizer
+(the role of reference examples/macbeth.sh deterministic output check and
+the API server routes)."""
+
+import http.client
+import json
+import threading
+import time
+
+import pytest
+
+from dllama_amd.utils.testing import make_byte_tokenizer, make_tiny_llama
+
+
+@pytest.fixture(scope="module")
+def assets(tmp_path_factory):
+    d = tmp_path_factory.mktemp("app")
+    mp = str(d / "tiny.m")
+    tp = str(d / "tiny.t")
+    make_byte_tokenizer(tp)
+    from dllama_amd.tokenizer import Tokenizer
+    vocab = Tokenizer(tp).vocab_size
+    make_tiny_llama(mp, vocab_size=vocab + (32 - vocab % 32) % 32)
+    return mp, tp
+
+
+def test_cli_inference(assets, capsys):
+    from dllama_amd.apps.main import main
+    mp, tp = assets
+    rc = main(["inference", "--model", mp, "--tokenizer", tp,
+               "--prompt", "hello world", "--steps", "8",
+               "--temperature", "0", "--gpu-index", "-1"])
+    assert rc == 0
+    out = capsys.readouterr().out
+    assert "Prediction" in out and "tokens/s" in out
+
+
+def test_cli_inference_deterministic(assets, capsys):
+    from dllama_amd.apps.main import main
+    mp, tp = assets
+    outs = []
+    for _ in range(2):
+        main(["inference", "--model", mp, "--tokenizer", tp,
+              "--prompt", "abc", "--steps", "6", "--temperature", "0",
+              "--gpu-index", "-1"])
+        outs.append(capsys.readouterr().out.split("Evaluation")[0])
+    assert outs[0] == outs[1]  # greedy decode is deterministic (macbeth.sh role)
+
+
+def test_cli_perplexity(assets, capsys):
+    from dllama_amd.apps.main import main
+    mp, tp = assets
+    rc = main(["perplexity", "--model", mp, "--tokenizer", tp,
+               "--prompt", "the quick brown fox jumps", "--gpu-index", "-1"])
+    assert rc == 0
+    out = capsys.readouterr().out
+    assert "Perplexity:" in out
+
+
+def test_api_server(assets):
+    from dllama_amd.apps import api as api_mod
+    mp, tp = assets
+    from dllama_amd.apps.main import build_parser
+    args = build_parser().parse_args(
+        ["inference", "--model", mp, "--tokenizer", tp, "--temperature", "0",
+         "--gpu-index", "-1", "--port", "18931"])
+    api_mod.STATE = api_mod.ApiState(args)
+    from http.server import HTTPServer
+    server = HTTPServer(("127.0.0.1", 18931), api_mod.Handler)
+    t = threading.Thread(target=server.serve_forever, daemon=True)
+    t.start()
+    try:
+        conn = http.client.HTTPConnection("127.0.0.1", 18931, timeout=60)
+        conn.request("GET", "/v1/models")
+        r = conn.getresponse()
+        assert r.status == 200
+        assert json.loads(r.read())["data"][0]["id"] == "dllama"
+
+        body = json.dumps({"messages": [{"role": "user", "content": "hi"}],
+                           "max_tokens": 4})
+        conn.request("POST", "/v1/chat/completions", body,
+                     {"Content-Type": "application/json"})
+        r = conn.getresponse()
+        assert r.status == 200
+        data = json.loads(r.read())
+        assert data["object"] == "chat.completion"
+        assert data["usage"]["completion_tokens"] >= 1
+
+        # second request exercises the NaiveCache prefix path
+        body = json.dumps({"messages": [{"role": "user", "content": "hi"},
+                                        {"role": "assistant", "content": "x"},
+                                        {"role": "user", "content": "more"}],
+                           "max_tokens": 4, "stream": True})
+        conn.request("POST", "/v1/chat/completions", body,
+                     {"Content-Type": "application/json"})
+        r = conn.getresponse()
+        assert r.status == 200
+        payload = r.read().decode()
+        assert "data:" in payload and "[DONE]" in payload
+        conn.close()  # keep-alive would block the single-threaded server loop
+    finally:
+        server.shutdown()
+        server.server_close()
+
+
+def test_converter_roundtrip(tmp_path):
+    """Fabricated HF checkpoint -> convert_hf -> runtime load parity."""
+    import numpy as np
+    import torch
+    from safetensors.numpy import save_file
+    import sys, os
+    sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "converter"))
+    import convert_hf
+
+    dim, hidden, layers, heads, kv_heads, vocab = 64, 96, 2, 4, 2, 256
+    hd = dim // heads
+    cfg = {"model_type": "llama", "hidden_size": dim, "intermediate_size": hidden,
+           "num_hidden_layers": layers, "num_attention_heads": heads,
+           "num_key_value_heads": kv_heads, "vocab_size": vocab,
+           "max_position_embeddings": 128, "hidden_act": "silu",
+           "rope_theta": 10000.0, "rms_norm_eps": 1e-5}
+    (tmp_path / "config.json").write_text(json.dumps(cfg))
+    rng = np.random.default_rng(3)
+    tensors = {"model.embed_tokens.weight":
+               rng.standard_normal((vocab, dim)).astype(np.float32) * 0.02,
+               "model.norm.weight": np.ones(dim, dtype=np.float32)}
+    for l in range(layers):
+        p = f"model.layers.{l}"
+        tensors[f"{p}.self_attn.q_proj.weight"] = rng.standard_normal((dim, dim)).astype(np.float32) * 0.05
+        tensors[f"{p}.self_attn.k_proj.weight"] = rng.standard_normal((kv_heads * hd, dim)).astype(np.float32) * 0.05
+        tensors[f"{p}.self_attn.v_proj.weight"] = rng.standard_normal((kv_heads * hd, dim)).astype(np.float32) * 0.05
+        tensors[f"{p}.self_attn.o_proj.weight"] = rng.standard_normal((dim, dim)).astype(np.float32) * 0.05
+        tensors[f"{p}.mlp.gate_proj.weight"] = rng.standard_normal((hidden, dim)).astype(np.float32) * 0.05
+        tensors[f"{p}.mlp.down_proj.weight"] = rng.standard_normal((dim, hidden)).astype(np.float32) * 0.05
+        tensors[f"{p}.mlp.up_proj.weight"] = rng.standard_normal((hidden, dim)).astype(np.float32) * 0.05
+        tensors[f"{p}.input_layernorm.weight"] = np.ones(dim, dtype=np.float32)
+        tensors[f"{p}.post_attention_layernorm.weight"] = np.ones(dim, dtype=np.float32)
+    save_file(tensors, str(tmp_path / "model.safetensors"))
+
+    out = str(tmp_path / "out.m")
+    convert_hf.convert(str(tmp_path), convert_hf.FLOAT_TYPES["q40"], out)
+
+    from dllama_amd import model_file as mflib
+    from dllama_amd.models.config import ModelConfig
+    from dllama_amd.models.cpu_model import CpuTransformer
+    m = mflib.ModelFile(out)
+    assert m.header.dim == dim and m.header.n_layers == layers
+    model = CpuTransformer(m, ModelConfig.from_header(m.header))
+    logits = model.forward(torch.tensor([1, 2, 3]), torch.arange(3))
+    assert torch.isfinite(logits).all()
