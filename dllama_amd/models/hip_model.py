@@ -215,12 +215,17 @@ class HipTransformer:
                                   device=dev)
         self.attn_counter = torch.zeros(NB * c.n_heads0, dtype=torch.int32, device=dev)
         if c.world > 1:
-            self.logits_gather = torch.zeros(c.world, NB, c.vocab0, device=dev)
+            # per-batch-size contiguous gather buffers: collectives need a
+            # flat contiguous output (world, nb*...) — a [:, :nb] slice is not
+            nbs = [n for n in (1, 2, 4, 8, 16, 32) if n <= NB]
+            self.logits_gather = {n: torch.zeros(c.world, n, c.vocab0, device=dev)
+                                  for n in nbs}
             if c.sync_type == Q80:
                 row_bytes = c.dim + 2 * (c.dim // QB)
                 self.sync_out = torch.zeros(NB * row_bytes, dtype=torch.uint8, device=dev)
-                self.sync_in = torch.zeros(c.world, NB * row_bytes,
-                                           dtype=torch.uint8, device=dev)
+                self.sync_in = {n: torch.zeros(c.world, n * row_bytes,
+                                               dtype=torch.uint8, device=dev)
+                                for n in nbs}
         if c.is_moe:
             S = NB * c.n_active_experts
             self.moe_idx = torch.zeros(S, dtype=torch.int32, device=dev)
@@ -251,9 +256,10 @@ class HipTransformer:
             row_bytes = c.dim + 2 * nb_dim
             out = self.sync_out[: NB * row_bytes]
             self.k.sync_pack(q.q[:NB], q.s[:NB], out)
-            inb = self.sync_in[:, : NB * row_bytes]
+            inb = self.sync_in[NB]
             self.comm.all_gather(inb, out)
-            self.k.merge_add(self.x[:NB], inb, self.ssq[slot])
+            self.k.merge_add(self.x[:NB], inb.view(c.world, NB, row_bytes),
+                             self.ssq[slot])
         else:
             self.comm.allreduce_(self.partial[:NB])
             self.k.add_ssq(self.x[:NB], self.partial[:NB], self.ssq[slot], NB)
@@ -353,14 +359,14 @@ class HipTransformer:
         norm_gemv(self.wcls, self.final_norm, slot, self.logits0,
                   self.amax_scratch if use_amax else None)
         if c.world > 1:
-            self.comm.all_gather(self.logits_gather[:, :NB], self.logits0[:NB])
+            self.comm.all_gather(self.logits_gather[NB], self.logits0[:NB])
         if self.greedy_feedback and B == 1:
             # on-device greedy sampling feeding the next decode step (used by
             # the fully graph-captured bench loop; real serving samples on host)
             if use_amax:
                 k.token_from_argmax(self.tokens, self.amax_scratch, self.amax_blocks)
             else:
-                full = self.logits_gather[:, 0].reshape(-1)
+                full = self.logits_gather[1][:, 0].reshape(-1)
                 self.tokens[0].copy_(torch.argmax(full))
 
     def _moe_ffn(self, B: int, NB: int, lw: dict):
@@ -409,7 +415,8 @@ class HipTransformer:
             self.forward_buffers(B)
         c = self.cfg
         if c.world > 1:
-            return (self.logits_gather[:, :B].permute(1, 0, 2)
+            NBp = _pow2_batch(B)
+            return (self.logits_gather[NBp][:, :B].permute(1, 0, 2)
                     .reshape(B, c.vocab_size))
         return self.logits0[:B]
 
