@@ -206,8 +206,9 @@ class HipTransformer:
         self.logits0 = torch.zeros(NB, c.vocab0, device=dev)
         rpw = 2 if c.vocab0 >= 2048 else 1
         self.amax_blocks = -(-c.vocab0 // (4 * rpw))  # mirrors gemv RPW choice
-        # sum-of-squares accumulators: [slot, batch, 16-way spread]
-        self.ssq = torch.zeros(2 * c.n_layers + 1, NB, 16, device=dev)
+        # sum-of-squares accumulators: [slot, batch, 16 spread x 32 pad]
+        # (16 slots each on their own cacheline; atomics to one line serialize)
+        self.ssq = torch.zeros(2 * c.n_layers + 1, NB, 16 * 32, device=dev)
         self.amax_scratch = torch.zeros(self.amax_blocks, dtype=torch.int64, device=dev)
         self.attn_splits = 8  # S=16,32 measured slower (combine reads S partials)
         self.attn_ml = torch.zeros(NB * c.n_heads0 * self.attn_splits * 2, device=dev)
@@ -229,6 +230,8 @@ class HipTransformer:
         if c.is_moe:
             S = NB * c.n_active_experts
             self.moe_idx = torch.zeros(S, dtype=torch.int32, device=dev)
+            self.moe_wts = torch.zeros(NB, c.n_active_experts, device=dev)
+            self.moe_router = torch.zeros(NB, c.n_experts, device=dev)
             self.moe_out13 = torch.zeros(S, 2 * c.ff_dim0, device=dev)
             self.moe_dq = QuantBuf(S, c.ff_dim0, dev)
             self.moe_y = torch.zeros(S, c.dim, device=dev)
@@ -340,13 +343,11 @@ class HipTransformer:
 
             # ffn block
             if c.is_moe:
-                k.norm_f32(x[:NB], lw["norm1"], self.ssq[slot], self.t_norm[:NB],
-                           NB, c.norm_eps)
-                self._moe_ffn(B, NB, lw)
-                if c.world == 1:
-                    k.add_ssq(x[:NB], self.partial[:NB], self.ssq[slot + 1], NB)
-                else:
-                    self._sync_partial(NB, slot + 1)
+                # norm once: Q80 triple for the expert GEMVs + f32 for router
+                k.norm_quant(x[:NB], lw["norm1"], self.ssq[slot], self.xq.q[:NB],
+                             self.xq.s[:NB], self.xq.bs[:NB], NB, c.norm_eps,
+                             self.t_norm[:NB])
+                self._moe_ffn(B, NB, lw, slot + 1)
             else:
                 norm_gemv(lw["w13"], lw["norm1"], slot, self.ff_out)
                 k.swiglu_q80(self.ff_out, self.ff_out[:, c.ff_dim0:],
@@ -369,19 +370,15 @@ class HipTransformer:
                 full = self.logits_gather[1][:, 0].reshape(-1)
                 self.tokens[0].copy_(torch.argmax(full))
 
-    def _moe_ffn(self, B: int, NB: int, lw: dict):
-        """Router (torch) + grouped expert GEMVs (reference llm.cpp:450-487);
-        t_norm holds the f32 normed activations (router input)."""
+    def _moe_ffn(self, B: int, NB: int, lw: dict, slot: int):
+        """Router + grouped expert GEMVs (reference llm.cpp:450-487);
+        t_norm holds the f32 normed activations (router input), xq the same
+        values Q80-quantized (expert GEMV input = reference repeat_z)."""
         c, k = self.cfg, self.k
         ka = c.n_active_experts
-        router = self.t_norm[:NB] @ lw["gate"].t()
-        probs = torch.softmax(router.float(), dim=-1)
-        wts, idx = torch.topk(probs, ka, dim=-1)
-        wts = wts / wts.sum(dim=-1, keepdim=True)
-        self.moe_idx[: NB * ka].copy_(idx.reshape(-1).to(torch.int32))
         S = NB * ka
-        k.q80_quantize(self.t_norm[:NB], self.xq.q[:NB], self.xq.s[:NB],
-                       self.xq.bs[:NB])
+        torch.matmul(self.t_norm[:NB], lw["gate"].t(), out=self.moe_router[:NB])
+        k.moe_gate(self.moe_router[:NB], self.moe_idx, self.moe_wts, NB, ka)
         k.q40_gemv_grouped(lw["w13"].qs, lw["w13"].scales, self.xq.q, self.xq.s,
                            self.xq.bs, self.moe_idx[:S], self.moe_out13, ka)
         k.swiglu_q80(self.moe_out13, self.moe_out13[:, c.ff_dim0:],
@@ -390,8 +387,14 @@ class HipTransformer:
         k.q40_gemv_grouped(lw["w2"].qs, lw["w2"].scales, self.moe_dq.q,
                            self.moe_dq.s, self.moe_dq.bs, self.moe_idx[:S],
                            self.moe_y, 1)
-        y = self.moe_y[:S].reshape(NB, ka, c.dim)
-        torch.sum(y * wts.unsqueeze(-1), dim=1, out=self.partial[:NB])
+        if c.world == 1:
+            k.scale_merge_add(self.x[:NB], self.moe_y, self.moe_wts,
+                              self.ssq[slot], NB, ka)
+        else:
+            y = self.moe_y[:S].reshape(NB, ka, c.dim)
+            torch.sum(y * self.moe_wts[:NB].unsqueeze(-1), dim=1,
+                      out=self.partial[:NB])
+            self._sync_partial(NB, slot)
 
     # ------------------------------------------------------------ engine API
 

@@ -21,6 +21,7 @@
 #define WAVE 64
 #define QB 32  // quant block size
 #define SSQ_SPREAD 16
+#define SSQ_PAD 32  // floats between spread slots (own 128B cacheline)
 
 static inline int ceil_div(int64_t a, int64_t b) { return (int)((a + b - 1) / b); }
 
@@ -68,7 +69,8 @@ __device__ __forceinline__ float group16_reduce_sum(float v) {
 __device__ __forceinline__ float ssq_total(const float *ssq, int b) {
     float t = 0.0f;
     #pragma unroll
-    for (int k = 0; k < SSQ_SPREAD; k++) t += ssq[b * SSQ_SPREAD + k];
+    for (int k = 0; k < SSQ_SPREAD; k++)
+        t += ssq[(b * SSQ_SPREAD + k) * SSQ_PAD];
     return t;
 }
 
@@ -298,7 +300,7 @@ __global__ void k_embed_gather(const float *__restrict__ table,
         if (threadIdx.x == 0) {
             float t = 0.0f;
             for (int i = 0; i < blockDim.x / WAVE; i++) t += red[i];
-            atomicAdd(ssq + b * SSQ_SPREAD + (blockIdx.x & (SSQ_SPREAD - 1)), t);
+            atomicAdd(ssq + (b * SSQ_SPREAD + (blockIdx.x & (SSQ_SPREAD - 1))) * SSQ_PAD, t);
         }
     }
 }
@@ -313,6 +315,7 @@ __global__ void k_norm_quant(const float *__restrict__ x,
                              int8_t *__restrict__ q,
                              float *__restrict__ s,
                              float *__restrict__ bs,
+                             float *__restrict__ yout,
                              int n, float eps) {
     const int b = blockIdx.y;
     const float inv = rsqrtf(ssq_total(ssq, b) / n + eps);
@@ -323,6 +326,7 @@ __global__ void k_norm_quant(const float *__restrict__ x,
     if (blk >= nb) return;
     const int i = blk * QB + lane;
     const float v = x[(int64_t)b * n + i] * inv * w[i];
+    if (yout != nullptr) yout[(int64_t)b * n + i] = v;
     const float amax = group32_reduce_max(fabsf(v));
     const float dd = amax / 127.0f;
     const float qinv = dd > 0.0f ? 1.0f / dd : 0.0f;
@@ -374,7 +378,7 @@ __global__ void k_add_ssq(float *__restrict__ x,
     if (threadIdx.x == 0) {
         float t = 0.0f;
         for (int i = 0; i < blockDim.x / WAVE; i++) t += red[i];
-        atomicAdd(ssq + b * SSQ_SPREAD + (blockIdx.x & (SSQ_SPREAD - 1)), t);
+        atomicAdd(ssq + (b * SSQ_SPREAD + (blockIdx.x & (SSQ_SPREAD - 1))) * SSQ_PAD, t);
     }
 }
 
@@ -671,7 +675,7 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
         if (threadIdx.x < NB) {
             float t = 0.0f;
             for (int wv = 0; wv < wpb; wv++) t += sred[wv][threadIdx.x];
-            atomicAdd(ssq + threadIdx.x * SSQ_SPREAD + (blockIdx.x & (SSQ_SPREAD - 1)), t);
+            atomicAdd(ssq + (threadIdx.x * SSQ_SPREAD + (blockIdx.x & (SSQ_SPREAD - 1))) * SSQ_PAD, t);
         }
     }
 }
@@ -837,34 +841,147 @@ __global__ void k_q40_gemv_grouped(const uint8_t *__restrict__ qs,
                                    const int *__restrict__ expert_idx,
                                    float *__restrict__ y,
                                    int d, int n, int k_slots) {
-    const int row = blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
+    const int wpb = blockDim.x / WAVE;
+    const int row0 = (blockIdx.x * wpb + threadIdx.x / WAVE) * 2;  // 2 rows/wave
     const int slot = blockIdx.y;
-    if (row >= d) return;
+    if (row0 >= d) return;
     const int lane = threadIdx.x % WAVE;
     const int nb = n / QB;
+    const int nbp = nb >> 1;
     const int e = expert_idx[slot];
     const int b = slot / k_slots;
-    const uint4 *wrow = reinterpret_cast<const uint4 *>(
-        qs + ((int64_t)e * d + row) * (n >> 1));
-    const __half *srow = scales + ((int64_t)e * d + row) * nb;
-    float acc = 0.0f;
-    for (int j = lane; j < nb; j += WAVE) {
-        const uint4 wq = wrow[j];
-        const float sw = __half2float(srow[j]);
-        const uint32_t wv[4] = {wq.x, wq.y, wq.z, wq.w};
-        const int4 x0 = reinterpret_cast<const int4 *>(xq + (int64_t)b * n + j * QB)[0];
-        const int4 x1 = reinterpret_cast<const int4 *>(xq + (int64_t)b * n + j * QB)[1];
-        const int32_t xv[8] = {x0.x, x0.y, x0.z, x0.w, x1.x, x1.y, x1.z, x1.w};
-        int idot = 0;
-        #pragma unroll
-        for (int wi = 0; wi < 4; wi++) {
-            idot = dot4((int)(wv[wi] & 0x0F0F0F0Fu), xv[wi], idot);
-            idot = dot4((int)((wv[wi] >> 4) & 0x0F0F0F0Fu), xv[4 + wi], idot);
-        }
-        acc = fmaf(sw * xs[b * nb + j], (float)idot - 8.0f * xbs[b * nb + j], acc);
+    const int row1 = min(row0 + 1, d - 1);
+    const uint4 *wrow0 = reinterpret_cast<const uint4 *>(
+        qs + ((int64_t)e * d + row0) * (n >> 1));
+    const uint4 *wrow1 = reinterpret_cast<const uint4 *>(
+        qs + ((int64_t)e * d + row1) * (n >> 1));
+    const __half *srow0 = scales + ((int64_t)e * d + row0) * nb;
+    const __half *srow1 = scales + ((int64_t)e * d + row1) * nb;
+    float acc0 = 0.0f, acc1 = 0.0f;
+    for (int jp = lane; jp < nbp; jp += WAVE) {
+        const int j = jp << 1;
+        const uint4 a0 = wrow0[j], a1 = wrow0[j + 1];
+        const uint4 b0 = wrow1[j], b1 = wrow1[j + 1];
+        const float2 sw0 = __half22float2(*reinterpret_cast<const __half2 *>(srow0 + j));
+        const float2 sw1 = __half22float2(*reinterpret_cast<const __half2 *>(srow1 + j));
+        const int4 *xr = reinterpret_cast<const int4 *>(xq + (int64_t)b * n) + j * 2;
+        const int4 x0 = xr[0], x1 = xr[1], x2 = xr[2], x3 = xr[3];
+        const float2 sx = *reinterpret_cast<const float2 *>(xs + (int64_t)b * nb + j);
+        const float2 bsum = *reinterpret_cast<const float2 *>(xbs + (int64_t)b * nb + j);
+        acc0 = fmaf(sw0.x * sx.x, (float)q40_block_dot(a0, x0, x1) - 8.0f * bsum.x, acc0);
+        acc0 = fmaf(sw0.y * sx.y, (float)q40_block_dot(a1, x2, x3) - 8.0f * bsum.y, acc0);
+        acc1 = fmaf(sw1.x * sx.x, (float)q40_block_dot(b0, x0, x1) - 8.0f * bsum.x, acc1);
+        acc1 = fmaf(sw1.y * sx.y, (float)q40_block_dot(b1, x2, x3) - 8.0f * bsum.y, acc1);
     }
-    float r = wave_reduce_sum(acc);
-    if (lane == 0) y[(int64_t)slot * d + row] = r;
+    if ((nb & 1) && lane == 0) {  // odd trailing block
+        const int j = nb - 1;
+        const int4 *xb = reinterpret_cast<const int4 *>(xq + (int64_t)b * n) + j * 2;
+        acc0 = fmaf(__half2float(srow0[j]) * xs[(int64_t)b * nb + j],
+                    (float)q40_block_dot(wrow0[j], xb[0], xb[1])
+                    - 8.0f * xbs[(int64_t)b * nb + j], acc0);
+        acc1 = fmaf(__half2float(srow1[j]) * xs[(int64_t)b * nb + j],
+                    (float)q40_block_dot(wrow1[j], xb[0], xb[1])
+                    - 8.0f * xbs[(int64_t)b * nb + j], acc1);
+    }
+    const float r0 = wave_reduce_sum(acc0);
+    const float r1 = wave_reduce_sum(acc1);
+    if (lane == 0) {
+        y[(int64_t)slot * d + row0] = r0;
+        if (row0 + 1 < d) y[(int64_t)slot * d + row0 + 1] = r1;
+    }
+}
+
+// MoE router: softmax over n_experts logits, top-k (first-index ties),
+// normalized weights + int32 expert ids (reference OP_SOFTMAX + OP_MOE_GATE,
+// nn-cpu-ops.cpp:1443-1492). One wave per batch row; n_experts <= 1024.
+__global__ void k_moe_gate(const float *__restrict__ logits,
+                           int *__restrict__ idx,
+                           float *__restrict__ wts,
+                           int n_experts, int topk) {
+    const int b = blockIdx.x;
+    const int lane = threadIdx.x;  // blockDim == 64
+    const int per = (n_experts + WAVE - 1) / WAVE;
+    float v[16];  // per-lane expert logits (supports n_experts <= 1024)
+    float m = -1e30f;
+    for (int i = 0; i < per; i++) {
+        const int eidx = lane * per + i;
+        v[i] = eidx < n_experts ? logits[(int64_t)b * n_experts + eidx] : -1e30f;
+        m = fmaxf(m, v[i]);
+    }
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        m = fmaxf(m, __shfl_xor(m, off, WAVE));
+    float sum = 0.0f;
+    for (int i = 0; i < per; i++) {
+        v[i] = (lane * per + i) < n_experts ? __expf(v[i] - m) : 0.0f;
+        sum += v[i];
+    }
+    sum = wave_reduce_sum(sum);
+    const float inv = 1.0f / sum;
+    // iterative top-k: packed (prob, smallest-index-wins) max per round
+    float wsum = 0.0f;
+    for (int t = 0; t < topk; t++) {
+        float best = -1.0f;
+        int bi = -1;
+        for (int i = 0; i < per; i++) {
+            const int eidx = lane * per + i;
+            if (v[i] > best) { best = v[i]; bi = eidx; }
+        }
+        // wave argmax: (prob, -index) lexicographic via packed compare
+        #pragma unroll
+        for (int off = 32; off > 0; off >>= 1) {
+            const float ob = __shfl_xor(best, off, WAVE);
+            const int oi = __shfl_xor(bi, off, WAVE);
+            if (ob > best || (ob == best && oi >= 0 && (bi < 0 || oi < bi))) {
+                best = ob; bi = oi;
+            }
+        }
+        if (lane == 0) {
+            idx[(int64_t)b * topk + t] = bi;
+            wts[(int64_t)b * topk + t] = best * inv;
+        }
+        wsum += best * inv;
+        // clear the winner
+        const int wl = bi / per, wi = bi % per;
+        if (lane == wl) v[wi] = -1.0f;
+    }
+    // normalize by the top-k sum (reference normTopk)
+    if (lane == 0) {
+        float t = 0.0f;
+        for (int i = 0; i < topk; i++) t += wts[(int64_t)b * topk + i];
+        const float winv = 1.0f / t;
+        for (int i = 0; i < topk; i++) wts[(int64_t)b * topk + i] *= winv;
+    }
+}
+
+// weighted sum of expert outputs + residual fold + ssq (reference OP_SCALE +
+// OP_MERGE_SUM + merge_add fused): x[b] += sum_s wts[b,s] * y[b*k+s]
+__global__ void k_scale_merge_add(float *__restrict__ x,
+                                  const float *__restrict__ y,
+                                  const float *__restrict__ wts,
+                                  float *__restrict__ ssq,
+                                  int n, int topk) {
+    const int b = blockIdx.y;
+    float local = 0.0f;
+    for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * blockDim.x) {
+        float acc = x[(int64_t)b * n + i];
+        for (int s = 0; s < topk; s++)
+            acc = fmaf(wts[(int64_t)b * topk + s],
+                       y[((int64_t)b * topk + s) * n + i], acc);
+        x[(int64_t)b * n + i] = acc;
+        local += acc * acc;
+    }
+    local = wave_reduce_sum(local);
+    __shared__ float red[16];
+    const int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    if (lane == 0) red[wid] = local;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float t = 0.0f;
+        for (int i = 0; i < blockDim.x / WAVE; i++) t += red[i];
+        atomicAdd(ssq + (b * SSQ_SPREAD + (blockIdx.x & (SSQ_SPREAD - 1))) * SSQ_PAD, t);
+    }
 }
 
 // ------------------------------------------------------------------ rope
@@ -1069,7 +1186,7 @@ __global__ void k_merge_add(float *__restrict__ x,
         if (threadIdx.x == 0) {
             float t = 0.0f;
             for (int i = 0; i < blockDim.x / WAVE; i++) t += red[i];
-            atomicAdd(ssq + r * SSQ_SPREAD + (blockIdx.x & (SSQ_SPREAD - 1)), t);
+            atomicAdd(ssq + (r * SSQ_SPREAD + (blockIdx.x & (SSQ_SPREAD - 1))) * SSQ_PAD, t);
         }
     }
 }
@@ -1388,14 +1505,35 @@ void embed_gather(torch::Tensor table, torch::Tensor tokens, torch::Tensor x,
 
 void norm_quant(torch::Tensor x, torch::Tensor w, torch::Tensor ssq,
                 torch::Tensor q, torch::Tensor s, torch::Tensor bs,
-                int64_t batch, double eps) {
+                int64_t batch, double eps,
+                c10::optional<torch::Tensor> yout = c10::nullopt) {
     CHECK_CUDA(x);
     const int n = x.size(-1);
     const dim3 grid(ceil_div(n, 256), batch);
     hipLaunchKernelGGL(k_norm_quant, grid, dim3(256), 0, cur_stream(),
                        x.data_ptr<float>(), w.data_ptr<float>(), ssq.data_ptr<float>(),
                        q.data_ptr<int8_t>(), s.data_ptr<float>(), bs.data_ptr<float>(),
+                       yout.has_value() ? yout->data_ptr<float>() : nullptr,
                        n, (float)eps);
+}
+
+void moe_gate(torch::Tensor logits, torch::Tensor idx, torch::Tensor wts,
+              int64_t batch, int64_t topk) {
+    CHECK_CUDA(logits);
+    const int n_experts = logits.size(-1);
+    TORCH_CHECK(n_experts <= 1024, "moe_gate supports <=1024 experts");
+    hipLaunchKernelGGL(k_moe_gate, dim3(batch), dim3(64), 0, cur_stream(),
+                       logits.data_ptr<float>(), idx.data_ptr<int>(),
+                       wts.data_ptr<float>(), n_experts, (int)topk);
+}
+
+void scale_merge_add(torch::Tensor x, torch::Tensor y, torch::Tensor wts,
+                     torch::Tensor ssq, int64_t batch, int64_t topk) {
+    CHECK_CUDA(x);
+    const int n = x.size(-1);
+    hipLaunchKernelGGL(k_scale_merge_add, dim3(ceil_div(n, 1024), batch), dim3(256),
+                       0, cur_stream(), x.data_ptr<float>(), y.data_ptr<float>(),
+                       wts.data_ptr<float>(), ssq.data_ptr<float>(), n, (int)topk);
 }
 
 void norm_f32(torch::Tensor x, torch::Tensor w, torch::Tensor ssq,
@@ -1554,7 +1692,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("x"), py::arg("wnorm"), py::arg("ssq"), py::arg("eps"),
           py::arg("y"), py::arg("batch"), py::arg("amax_slot") = py::none());
     m.def("q40_gemv_nq_rope", &q40_gemv_nq_rope);
-    m.def("norm_quant", &norm_quant);
+    m.def("norm_quant", &norm_quant, py::arg("x"), py::arg("w"), py::arg("ssq"),
+          py::arg("q"), py::arg("s"), py::arg("bs"), py::arg("batch"),
+          py::arg("eps"), py::arg("yout") = py::none());
+    m.def("moe_gate", &moe_gate);
+    m.def("scale_merge_add", &scale_merge_add);
     m.def("norm_f32", &norm_f32);
     m.def("add_ssq", &add_ssq);
     m.def("q40_gemv_grouped", &q40_gemv_grouped);

@@ -299,3 +299,29 @@ def test_add_and_pos_inc(k):
     pos = torch.tensor([3], dtype=torch.int32, device=DEV)
     k.pos_inc(pos, 2)
     assert int(pos.item()) == 5
+
+
+def test_moe_gate(k):
+    B, E, topk = 3, 128, 8
+    logits = rand(B, E, seed=97)
+    idx = torch.zeros(B * topk, dtype=torch.int32, device=DEV)
+    wts = torch.zeros(B, topk, device=DEV)
+    k.moe_gate(logits, idx, wts, B, topk)
+    ridx, rwts = R.moe_gate(logits.cpu(), topk)
+    assert torch.equal(idx.cpu().reshape(B, topk).long(), ridx)
+    assert torch.allclose(wts.cpu(), rwts, atol=1e-5)
+
+
+def test_scale_merge_add(k):
+    B, n, topk = 2, 256, 4
+    x = rand(B, n, seed=98)
+    y = rand(B * topk, n, seed=99)
+    wts = rand(B, topk, seed=100).abs()
+    ssq = torch.zeros(B, 16 * 32, device=DEV)
+    got = x.clone()
+    k.scale_merge_add(got, y, wts, ssq, B, topk)
+    want = x.cpu() + (y.cpu().reshape(B, topk, n)
+                      * wts.cpu().unsqueeze(-1)).sum(1)
+    assert torch.allclose(got.cpu(), want, atol=1e-4)
+    ssq_tot = ssq.cpu().reshape(B, 16, 32)[:, :, 0].sum(-1)
+    assert torch.allclose(ssq_tot, (want * want).sum(-1), rtol=1e-4)
