@@ -268,14 +268,15 @@ class HipTransformer:
             self.k.add_ssq(self.x[:NB], self.partial[:NB], self.ssq[slot], NB)
 
     def _proj_merge(self, lin: Linear, qb: QuantBuf, slot: int, NB: int):
-        """Down-projection + residual fold. The GEMV RESID epilogue variant
-        measured +6.5us/call (dependent scalar RMW + barrier tail), so the
-        fold is a separate wide add_ssq kernel (~2us)."""
+        """Down-projection + residual fold: TP=1 fuses the add + ssq into the
+        GEMV epilogue (+1us with cacheline-strided ssq slots — the earlier
+        +6.5us was atomics serializing on one cacheline)."""
         k = self.k
-        k.q40_gemv(lin.qs, lin.scales, qb.q, qb.s, qb.bs, self.partial, NB)
         if self.cfg.world == 1:
-            k.add_ssq(self.x[:NB], self.partial[:NB], self.ssq[slot], NB)
+            k.q40_gemv_resid(lin.qs, lin.scales, qb.q, qb.s, qb.bs,
+                             self.x, self.ssq[slot], NB)
         else:
+            k.q40_gemv(lin.qs, lin.scales, qb.q, qb.s, qb.bs, self.partial, NB)
             self._sync_partial(NB, slot)
 
     def forward_buffers(self, B: int):
