@@ -16,6 +16,7 @@
 #include <hip/hip_runtime.h>
 #include <hip/hip_fp16.h>
 #include <torch/extension.h>
+#include <cstdlib>
 #include <ATen/hip/HIPContext.h>
 
 #define WAVE 64
@@ -1307,13 +1308,17 @@ static void gemv_launch(torch::Tensor &qs, torch::Tensor &scales, torch::Tensor 
         return;
     }
     switch (batch) {
-        case 1:
-            // RPW=2 is the sweet spot (RPW=4 halves wave count and loses)
-            if (d >= 2048 || e.force_rpw2)
+        case 1: {
+            // RPW=2 halves the wave count for 2x per-lane loads in flight;
+            // tunable crossover (DLLAMA_RPW1_MAX: use RPW1 up to that d)
+            static const int rpw1_max =
+                std::getenv("DLLAMA_RPW1_MAX") ? atoi(std::getenv("DLLAMA_RPW1_MAX")) : 0;
+            if (e.force_rpw2 || (d >= 2048 && d > rpw1_max))
                 launch(std::integral_constant<int, 1>{}, r2);
             else
                 launch(std::integral_constant<int, 1>{}, r1);
             break;
+        }
         case 2: launch(std::integral_constant<int, 2>{}, r2); break;
         case 4: launch(std::integral_constant<int, 4>{}, r2); break;
         case 8: launch(std::integral_constant<int, 8>{}, r1); break;

@@ -88,3 +88,21 @@ print("attn(pos=64)  fused+quant: %6.2f us" %
 xr = torch.randn(1, dim, device=dev, generator=g)
 pr = torch.randn(1, dim, device=dev, generator=g)
 print("add_ssq [1,4096]: %6.2f us" % bench(lambda: k.add_ssq(xr, pr, ssq[0], 1)))
+
+# PRO (fused norm+quant prologue) vs separate norm_quant + gemv, NB=1
+for d, n, tag in ((6144, 4096, "qkv"), (28672, 4096, "w13"), (128256, 4096, "cls")):
+    qs, sc = mklin(d, n)
+    xq, xs, xbs = mkx(1, n)
+    y = torch.zeros(1, d, device=dev)
+    q8b = torch.zeros(1, n, dtype=torch.int8, device=dev)
+    s8b = torch.zeros(1, n // 32, device=dev)
+    bs8b = torch.zeros(1, n // 32, device=dev)
+    wn = torch.rand(n, device=dev, generator=g)
+    xf = torch.randn(1, n, device=dev, generator=g)
+    sq = torch.rand(1, 16 * 32, device=dev, generator=g)
+    def sep():
+        k.norm_quant(xf, wn, sq, q8b, s8b, bs8b, 1, 1e-5)
+        k.q40_gemv(qs, sc, q8b, s8b, bs8b, y, 1)
+    t_sep = bench(sep, n=64)
+    t_pro = bench(lambda: k.q40_gemv_nq(qs, sc, xf, wn, sq, 1e-5, y, 1), n=64)
+    print(f"PRO {tag:4s}: norm_quant+gemv={t_sep:6.2f}us  fused-nq={t_pro:6.2f}us")
