@@ -1680,7 +1680,117 @@ void pos_inc(torch::Tensor pos, int64_t delta) {
                        pos.data_ptr<int>(), (int)delta);
 }
 
+// ===================================================== native BPE encoder
+// Host-side tokenizer encode (the reference's tokenizer is native C++,
+// src/tokenizer.cpp:311-390): exact-match byte accumulation with special
+// token scan, then greedy highest-score pair merging. The Python Tokenizer
+// delegates here when the extension is loaded.
+#include <algorithm>
+#include <string>
+#include <unordered_map>
+#include <vector>
+
+struct BpeEncoder {
+    std::vector<std::string> vocab;
+    std::vector<float> scores;
+    std::unordered_map<std::string, int> regular;  // first-id wins
+    std::vector<int> special_ids;                  // ids >= regular_size
+    int regular_size;
+
+    BpeEncoder(const std::vector<std::string> &vocab_,
+               const std::vector<double> &scores_, int64_t regular_size_)
+        : vocab(vocab_), regular_size((int)regular_size_) {
+        scores.reserve(scores_.size());
+        for (double s : scores_) scores.push_back((float)s);
+        for (int i = 0; i < regular_size && i < (int)vocab.size(); i++)
+            regular.emplace(vocab[i], i);  // emplace keeps the first id
+        for (int i = regular_size; i < (int)vocab.size(); i++)
+            special_ids.push_back(i);
+    }
+
+    std::vector<int> encode(const std::string &data, bool add_special) const {
+        std::vector<int> tokens;
+        std::string buf;
+        size_t i = 0;
+        while (i < data.size()) {
+            if (add_special) {
+                int sp = -1;
+                for (int sid : special_ids) {
+                    const std::string &p = vocab[sid];
+                    if (!p.empty() && data.compare(i, p.size(), p) == 0) {
+                        sp = sid;
+                        break;
+                    }
+                }
+                if (sp >= 0) {
+                    if (!buf.empty())
+                        throw std::runtime_error("unencodable byte run before special");
+                    tokens.push_back(sp);
+                    i += vocab[sp].size();
+                    continue;
+                }
+            }
+            buf.push_back(data[i++]);
+            auto it = regular.find(buf);
+            if (it != regular.end()) {
+                tokens.push_back(it->second);
+                buf.clear();
+            }
+        }
+        if (!buf.empty())
+            throw std::runtime_error("cannot encode byte run");
+
+        // greedy merge: globally best-score adjacent pair each round with
+        // leftmost tie-break — identical semantics to the reference's O(n^2)
+        // rescan (tokenizer.cpp:352-379) via heap + linked list + lazy
+        // invalidation (O(n log n)).
+        const int n = (int)tokens.size();
+        if (n == 0) return tokens;
+        std::vector<int> tok(tokens), prev(n), next(n), ver(n, 0);
+        for (int j = 0; j < n; j++) { prev[j] = j - 1; next[j] = j + 1 < n ? j + 1 : -1; }
+        struct Cand { float score; int pos, id, vl, vr, right; };
+        auto worse = [](const Cand &a, const Cand &b) {
+            if (a.score != b.score) return a.score < b.score;  // max-heap
+            return a.pos > b.pos;                              // leftmost wins
+        };
+        std::vector<Cand> heap;
+        auto push_pair = [&](int l) {
+            const int r = next[l];
+            if (l < 0 || r < 0) return;
+            auto it = regular.find(vocab[tok[l]] + vocab[tok[r]]);
+            if (it == regular.end()) return;
+            heap.push_back({scores[it->second], l, it->second, ver[l], ver[r], r});
+            std::push_heap(heap.begin(), heap.end(), worse);
+        };
+        for (int j = 0; j + 1 < n; j++) push_pair(j);
+        while (!heap.empty()) {
+            std::pop_heap(heap.begin(), heap.end(), worse);
+            const Cand c = heap.back();
+            heap.pop_back();
+            const int l = c.pos, r = c.right;
+            if (ver[l] != c.vl || ver[r] != c.vr || next[l] != r) continue;
+            tok[l] = c.id;
+            ver[l]++;
+            ver[r]++;
+            next[l] = next[r];
+            if (next[r] >= 0) prev[next[r]] = l;
+            push_pair(prev[l] >= 0 ? prev[l] : -1);
+            push_pair(l);
+        }
+        std::vector<int> out;
+        for (int j = 0; j >= 0; j = next[j]) out.push_back(tok[j]);
+        return out;
+    }
+};
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    py::class_<BpeEncoder>(m, "BpeEncoder")
+        .def(py::init<const std::vector<std::string> &,
+                      const std::vector<double> &, int64_t>())
+        .def("encode", [](const BpeEncoder &e, const py::bytes &data,
+                          bool add_special) {
+            return e.encode(std::string(data), add_special);
+        });
     m.def("q80_quantize", &q80_quantize);
     m.def("rmsnorm", &rmsnorm);
     m.def("rmsnorm_q80", &rmsnorm_q80);

@@ -105,6 +105,17 @@ class Tokenizer:
         self._special = [(self.vocab[t], t)
                          for t in range(self.regular_vocab_size, self.vocab_size)]
         self._decode_buf = bytearray()
+        # native C++ encoder from the extension (reference tokenizer is C++);
+        # fall back to the pure-Python path when the extension isn't built
+        self._native = None
+        try:
+            from .ops import hip_ops
+            k = hip_ops()
+            self._native = k.BpeEncoder(list(self.vocab),
+                                        [float(s) for s in self.scores],
+                                        self.regular_vocab_size)
+        except Exception:  # noqa: BLE001
+            self._native = None
 
     # ---------------------------------------------------------- encode
 
@@ -114,6 +125,10 @@ class Tokenizer:
         tokens: list[int] = []
         if is_start and self.add_bos and self.bos_id >= 0:
             tokens.append(self.bos_id)
+
+        if self._native is not None:
+            tokens.extend(self._native.encode(data, add_special_tokens))
+            return tokens
 
         buf = bytearray()
         i = 0
