@@ -61,11 +61,18 @@ class InferenceEngine:
         if start_pos is not None:
             self.pos = start_pos
         logits = None
-        for i in range(0, len(tokens), self.n_batches):
+        starts = list(range(0, len(tokens), self.n_batches))
+        for i in starts:
             chunk = tokens[i: i + self.n_batches]
             t = torch.tensor(chunk, dtype=torch.int64)
             p = torch.arange(self.pos, self.pos + len(chunk), dtype=torch.int64)
-            logits = self.model.forward(t, p)
+            # intermediate prefill chunks don't need logits (the reference
+            # computes them anyway; real serving shouldn't)
+            self.model.skip_logits = i != starts[-1]
+            try:
+                logits = self.model.forward(t, p)
+            finally:
+                self.model.skip_logits = False
             self.pos += len(chunk)
         return logits[-1]
 
@@ -94,7 +101,7 @@ class InferenceEngine:
         out: list[int] = []
         sampler = self.sampler or Sampler(logits.shape[-1], 0.0, 0.9, 12345)
         t0 = time.perf_counter()
-        token = sampler.sample(_to_numpy(logits))
+        token = sampler.sample(logits)
         out.append(token)
         if on_token:
             on_token(token)
@@ -102,7 +109,7 @@ class InferenceEngine:
             if stop_check and stop_check(token):
                 break
             logits = self.decode_one(token)
-            token = sampler.sample(_to_numpy(logits))
+            token = sampler.sample(logits)
             out.append(token)
             if on_token:
                 on_token(token)

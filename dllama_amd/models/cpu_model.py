@@ -33,6 +33,7 @@ class CpuTransformer:
         self.cfg = config
         self.comm = comm or SingleComm()
         self.activation_quant = activation_quant  # False = pure f32 (debugging)
+        self.skip_logits = False  # accepted for engine compat; CPU always computes
         c = self.cfg
         r, w = c.rank, c.world
 

@@ -87,6 +87,7 @@ class HipTransformer:
         self._graph = None
         self._graph_pos = None
         self.greedy_feedback = False
+        self.skip_logits = False  # prefill chunks before the last skip wcls
         self._alloc_buffers()
 
     # ------------------------------------------------------------ weights
@@ -357,6 +358,8 @@ class HipTransformer:
                 self._proj_merge(lw["w2"], self.dq, slot + 1, NB)
             slot += 1
 
+        if self.skip_logits and B > 1:
+            return
         use_amax = (self.greedy_feedback and B == 1 and c.world == 1)
         norm_gemv(self.wcls, self.final_norm, slot, self.logits0,
                   self.amax_scratch if use_amax else None)

@@ -274,7 +274,34 @@ class Sampler:
         u, self.state = _xorshift_u32(self.state)
         return (u >> 8) / 16777216.0
 
-    def sample(self, logits: np.ndarray) -> int:
+    def sample_torch(self, logits) -> int:
+        """Device-side sampling for GPU logits (temperature/softmax/top-p all
+        on device; only the chosen token id crosses PCIe). Same xorshift coin
+        as the CPU path."""
+        import torch
+        logits = logits.reshape(-1)[: self.vocab_size]
+        if self.temperature == 0.0:
+            return int(torch.argmax(logits).item())
+        p = torch.softmax(logits.float() / self.temperature, dim=-1)
+        coin = self._random_f32()
+        if self.topp <= 0 or self.topp >= 1:
+            cdf = torch.cumsum(p, dim=-1)
+            return int(torch.searchsorted(cdf, torch.tensor(coin, device=p.device))
+                       .clamp(0, self.vocab_size - 1).item())
+        probs, order = torch.sort(p, descending=True)
+        c = torch.cumsum(probs, dim=-1)
+        last = int(torch.searchsorted(c, torch.tensor(self.topp, device=p.device))
+                   .item())
+        last = min(last, probs.numel() - 1)
+        r = coin * float(c[last].item())
+        pick = int(torch.searchsorted(c[: last + 1],
+                                      torch.tensor(r, device=p.device)).item())
+        return int(order[min(pick, last)].item())
+
+    def sample(self, logits) -> int:
+        import torch
+        if isinstance(logits, torch.Tensor) and logits.is_cuda:
+            return self.sample_torch(logits)
         logits = np.asarray(logits, dtype=np.float32).reshape(-1)[: self.vocab_size]
         if self.temperature == 0.0:
             return int(np.argmax(logits))
