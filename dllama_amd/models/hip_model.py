@@ -268,16 +268,27 @@ class HipTransformer:
             self.comm.allreduce_(self.partial[:NB])
             self.k.add_ssq(self.x[:NB], self.partial[:NB], self.ssq[slot], NB)
 
+    def _mm(self, lin: Linear, qb: QuantBuf, out, NB: int, amax=None):
+        """Batched matmul dispatch: decode batches use the dot4 GEMV,
+        prefill batches (>=8) the int8-MFMA GEMM."""
+        if NB >= 8:
+            self.k.q40_gemm(lin.qs, lin.scales, qb.q, qb.s, out, NB)
+        else:
+            self.k.q40_gemv(lin.qs, lin.scales, qb.q, qb.s, qb.bs, out, NB, amax)
+
     def _proj_merge(self, lin: Linear, qb: QuantBuf, slot: int, NB: int):
-        """Down-projection + residual fold: TP=1 fuses the add + ssq into the
-        GEMV epilogue (+1us with cacheline-strided ssq slots — the earlier
-        +6.5us was atomics serializing on one cacheline)."""
+        """Down-projection + residual fold: TP=1 decode fuses the add + ssq
+        into the GEMV epilogue (+1us with cacheline-strided ssq slots — the
+        earlier +6.5us was atomics serializing on one cacheline)."""
         k = self.k
-        if self.cfg.world == 1:
+        if self.cfg.world == 1 and NB < 8:
             k.q40_gemv_resid(lin.qs, lin.scales, qb.q, qb.s, qb.bs,
                              self.x, self.ssq[slot], NB)
+            return
+        self._mm(lin, qb, self.partial, NB)
+        if self.cfg.world == 1:
+            k.add_ssq(self.x[:NB], self.partial[:NB], self.ssq[slot], NB)
         else:
-            k.q40_gemv(lin.qs, lin.scales, qb.q, qb.s, qb.bs, self.partial, NB)
             self._sync_partial(NB, slot)
 
     def forward_buffers(self, B: int):
@@ -299,15 +310,14 @@ class HipTransformer:
         slot = 0
 
         def norm_gemv(lin, wn, slot, out, amax=None):
-            """normed+quantized x -> GEMV (fused prologue for decode)."""
+            """normed+quantized x -> matmul (fused prologue for decode)."""
             if fused_norm:
                 k.q40_gemv_nq(lin.qs, lin.scales, x, wn, self.ssq[slot],
                               c.norm_eps, out, NB, amax)
             else:
                 k.norm_quant(x[:NB], wn, self.ssq[slot], self.xq.q[:NB],
                              self.xq.s[:NB], self.xq.bs[:NB], NB, c.norm_eps)
-                k.q40_gemv(lin.qs, lin.scales, self.xq.q, self.xq.s,
-                           self.xq.bs, out, NB, amax)
+                self._mm(lin, self.xq, out, NB, amax)
 
         for l, lw in enumerate(self.layers):
             # attention block
@@ -317,7 +327,7 @@ class HipTransformer:
                                    self.qkv_out, NB, self.rope_cache, self.pos,
                                    self.k_cache[l], self.v_cache[l],
                                    c.q_dim0, c.kv_dim0, c.head_dim)
-            elif fused_rope:
+            elif fused_rope and NB < 8:
                 k.norm_quant(x[:NB], lw["norm0"], self.ssq[slot], self.xq.q[:NB],
                              self.xq.s[:NB], self.xq.bs[:NB], NB, c.norm_eps)
                 k.q40_gemv_rope(lw["qkv"].qs, lw["qkv"].scales, self.xq.q,

@@ -892,6 +892,81 @@ __global__ void k_q40_gemv_grouped(const uint8_t *__restrict__ qs,
     }
 }
 
+// ------------------------------------------------- int8 MFMA prefill GEMM
+// Batched (prefill) Q40xQ80 matmul on the matrix cores:
+// C[b][m] = sum_j sw[m,j]*sx[b,j] * (x_i8[b, j*32..] . w_i4[m, j*32..])
+// One v_mfma_i32_32x32x32_i8 per (32-row, 32-batch, 1-block) tile — K=32
+// matches the quant block exactly, so per-block scales stay exact: the i32
+// tile is descaled into f32 accumulators after every MFMA.
+// Fragment layouts verified on-device (tools/mfma_probe.hip):
+//   A[m][k]: lane=(m&31)|((k>>4)<<5), byte=k&15 ; B[k][n] mirrored on n;
+//   C[m][n]: col=lane&31, row=(r&3)+8*(r>>2)+4*(lane>>5).
+// Weights ride as the B operand so each lane needs only ONE weight-row
+// scale per block; the 16 batch-row x-scales come from wave-private LDS.
+typedef int v4i32_t __attribute__((ext_vector_type(4)));
+typedef int v16i32_t __attribute__((ext_vector_type(16)));
+
+__global__ void __launch_bounds__(256)
+k_q40_gemm(const uint8_t *__restrict__ qs,
+           const __half *__restrict__ scales,
+           const int8_t *__restrict__ xq,
+           const float *__restrict__ xs,
+           float *__restrict__ y,
+           int d, int n, int batch) {
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int mbase = (blockIdx.x * 4 + wave) * 32;  // 32 weight rows per wave
+    if (mbase >= d) return;
+    const int nb = n / QB;
+    const int khi = lane >> 5;       // 0: elems 0..15 (lo nibbles), 1: 16..31 (hi)
+    const int bcol = lane & 31;      // x batch row for the A fragment
+    const int mcol = lane & 31;      // weight row within the tile (B operand)
+    const int mrow = min(mbase + mcol, d - 1);
+    const uint4 *wrow = reinterpret_cast<const uint4 *>(qs + (int64_t)mrow * (n >> 1));
+    const __half *srow = scales + (int64_t)mrow * nb;
+
+    float facc[16];
+    #pragma unroll
+    for (int r = 0; r < 16; r++) facc[r] = 0.0f;
+
+    for (int j = 0; j < nb; j++) {
+        // lanes 0..31 hold this block's x-scale for batch row `lane`;
+        // the descale fetches them cross-lane via shfl (no LDS, no barrier)
+        const float sxv = (lane < 32) ? xs[(int64_t)(lane & 31) * nb + j] : 0.0f;
+        // A: x int8, batch row bcol, 16 bytes at k = khi*16
+        v4i32_t a = *reinterpret_cast<const v4i32_t *>(
+            xq + (int64_t)bcol * n + j * QB + khi * 16);
+        // B: weight nibbles -> signed i8 (nib - 8), bytewise:
+        //   s = nib ^ 8 gives |result| bits; negative bytes (nib<8) OR in 0xF0
+        const uint4 wq = wrow[j];
+        const uint32_t wv[4] = {wq.x, wq.y, wq.z, wq.w};
+        v4i32_t b;
+        #pragma unroll
+        for (int t = 0; t < 4; t++) {
+            uint32_t s = khi ? ((wv[t] >> 4) & 0x0F0F0F0Fu)
+                             : (wv[t] & 0x0F0F0F0Fu);
+            s ^= 0x08080808u;
+            b[t] = (int)(s | (((s >> 3) & 0x01010101u) * 0xF0u));
+        }
+        v16i32_t iacc = {};
+        iacc = __builtin_amdgcn_mfma_i32_32x32x32_i8(a, b, iacc, 0, 0, 0);
+        const float sw = __half2float(srow[j]);
+        #pragma unroll
+        for (int r = 0; r < 16; r++) {
+            const int brow = (r & 3) + 8 * (r >> 2) + 4 * khi;
+            facc[r] = fmaf((float)iacc[r], sw * __shfl(sxv, brow, WAVE), facc[r]);
+        }
+    }
+    if (mbase + mcol < d) {
+        #pragma unroll
+        for (int r = 0; r < 16; r++) {
+            const int brow = (r & 3) + 8 * (r >> 2) + 4 * khi;
+            if (brow < batch)
+                y[(int64_t)brow * d + mbase + mcol] = facc[r];
+        }
+    }
+}
+
 // MoE router: softmax over n_experts logits, top-k (first-index ties),
 // normalized weights + int32 expert ids (reference OP_SOFTMAX + OP_MOE_GATE,
 // nn-cpu-ops.cpp:1443-1492). One wave per batch row; n_experts <= 1024.
@@ -1407,6 +1482,22 @@ void q40_gemv_nq_rope(torch::Tensor qs, torch::Tensor scales, torch::Tensor x,
     gemv_launch<EPI_ROPE, 1>(qs, scales, x, x, x, y.data_ptr<float>(), batch, e);
 }
 
+void q40_gemm(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
+              torch::Tensor xs, torch::Tensor y, int64_t batch) {
+    // int8-MFMA batched matmul (prefill path); xq/xs must have >=32 rows
+    CHECK_CUDA(qs); CHECK_CONT(qs); CHECK_CONT(xq);
+    const int d = qs.size(0);
+    const int n = qs.size(1) * 2;
+    TORCH_CHECK(xq.size(-1) == n, "x width mismatch");
+    TORCH_CHECK(xq.size(0) >= 32, "gemm needs 32 padded batch rows");
+    TORCH_CHECK(n % QB == 0, "n must be a multiple of 32");
+    hipLaunchKernelGGL(k_q40_gemm, dim3(ceil_div(d, 128)), dim3(256), 0,
+                       cur_stream(), qs.data_ptr<uint8_t>(),
+                       reinterpret_cast<const __half *>(scales.data_ptr<at::Half>()),
+                       xq.data_ptr<int8_t>(), xs.data_ptr<float>(),
+                       y.data_ptr<float>(), d, n, (int)batch);
+}
+
 int64_t q40_gemv_argmax_blocks(int64_t d) {
     // stage-1 argmax scratch entries for a batch-1 GEMV over d rows;
     // must mirror the RPW selection above
@@ -1802,6 +1893,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("xs"), py::arg("xbs"), py::arg("y"), py::arg("batch"),
           py::arg("amax_slot") = py::none());
     m.def("q40_gemv_resid", &q40_gemv_resid);
+    m.def("q40_gemm", &q40_gemm);
     m.def("q40_gemv_rope", &q40_gemv_rope);
     m.def("q40_gemv_nq", &q40_gemv_nq, py::arg("qs"), py::arg("scales"),
           py::arg("x"), py::arg("wnorm"), py::arg("ssq"), py::arg("eps"),

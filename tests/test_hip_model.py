@@ -94,3 +94,21 @@ def test_native_extension_is_loaded():
     from dllama_amd.ops import hip_ops
     k = hip_ops()
     assert "dllama_amd/ops/_build" in k.__file__, k.__file__
+
+
+def test_hip_prefill_gemm_path(tiny_path):
+    """Prefill batches >=8 run the int8-MFMA GEMM; decode==prefill must
+    still hold across the GEMV/GEMM boundary."""
+    from dllama_amd.models.hip_model import HipTransformer
+    m = mf.ModelFile(tiny_path)
+    cfg = ModelConfig.from_header(m.header)
+    hip = HipTransformer.from_file(m, cfg)
+    tokens = list(range(3, 23))  # B=20 -> padded 32 -> GEMM path
+    batch = hip.forward(torch.tensor(tokens), torch.arange(len(tokens))).cpu().clone()
+    hip2 = HipTransformer.from_file(m, cfg)
+    for i, t in enumerate(tokens):
+        one = hip2.forward(torch.tensor([t]), torch.tensor([i])).cpu()
+    assert _rel_err(one[0], batch[-1]) < 0.02
+    cpu = CpuTransformer(m, cfg)
+    want = cpu.forward(torch.tensor(tokens), torch.arange(len(tokens)))
+    assert _rel_err(batch, want) < 0.02

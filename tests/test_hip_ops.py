@@ -325,3 +325,23 @@ def test_scale_merge_add(k):
     assert torch.allclose(got.cpu(), want, atol=1e-4)
     ssq_tot = ssq.cpu().reshape(B, 16, 32)[:, :, 0].sum(-1)
     assert torch.allclose(ssq_tot, (want * want).sum(-1), rtol=1e-4)
+
+
+def test_q40_gemm_matches_gemv(k):
+    """int8-MFMA prefill GEMM vs the dot4 GEMV on identical inputs."""
+    d, n, B = 160, 1024, 20  # d not a multiple of 128 exercises row masking
+    qs, sc, wref = _mk_linear(d, n, 55)
+    x = rand(32, n, seed=56, scale=0.5)
+    q = torch.zeros(32, n, dtype=torch.int8, device=DEV)
+    s = torch.zeros(32, n // 32, device=DEV)
+    bs = torch.zeros(32, n // 32, device=DEV)
+    k.q80_quantize(x, q, s, bs)
+    y_gemv = torch.zeros(32, d, device=DEV)
+    k.q40_gemv(qs, sc, q, s, bs, y_gemv, 32)
+    y_gemm = torch.zeros(32, d, device=DEV)
+    k.q40_gemm(qs, sc, q, s, y_gemm, B)
+    assert torch.allclose(y_gemm[:B], y_gemv[:B], atol=1e-3, rtol=1e-4), \
+        (y_gemm[:B] - y_gemv[:B]).abs().max().item()
+    want = R.q40_matmul(x[:B].cpu(), wref)
+    tol = want.abs().max().item() * 0.02 + 1e-3
+    assert torch.allclose(y_gemm[:B].cpu(), want, atol=tol)
