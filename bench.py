@@ -77,7 +77,11 @@ def main():
 
     # short prefill so the decode attends over a non-trivial context
     prompt = torch.randint(0, cfg.vocab_size, (args.prefill,))
-    model.forward(prompt, torch.arange(args.prefill))
+    for i in range(0, args.prefill, 32):
+        chunk = prompt[i: i + 32]
+        model.skip_logits = i + 32 < args.prefill
+        model.forward(chunk, torch.arange(i, i + len(chunk)))
+        model.skip_logits = False
 
     use_graph = not args.no_graph
     if use_graph:
