@@ -88,6 +88,8 @@ class HipTransformer:
         self._graph_pos = None
         self.greedy_feedback = False
         self.skip_logits = False  # prefill chunks before the last skip wcls
+        self._pf_graphs = {}      # skip_logits -> captured 32-token graph
+        self._pf_failed = False
         self._alloc_buffers()
 
     # ------------------------------------------------------------ weights
@@ -426,6 +428,18 @@ class HipTransformer:
                 self.pos.fill_(p0)
             self._graph.replay()
             self._graph_pos = p0 + 1  # the graph's pos_inc advanced it
+        elif B == self.n_batches and not self._pf_failed:
+            # full prefill chunks replay a captured graph (eager chunk cost
+            # is host-launch-bound: ~400 python kernel launches)
+            g = self._pf_graphs.get(self.skip_logits)
+            if g is None:
+                g = self._capture_prefill_graph(self.skip_logits)
+            self.pos.fill_(p0)
+            self._graph_pos = None
+            if g is not None:
+                g.replay()
+            else:
+                self.forward_buffers(B)
         else:
             self.pos.fill_(p0)
             self._graph_pos = None
@@ -438,6 +452,28 @@ class HipTransformer:
         return self.logits0[:B]
 
     # ------------------------------------------------------------ graphs
+
+    def _capture_prefill_graph(self, skip_logits: bool):
+        try:
+            torch.cuda.synchronize(self.device)
+            saved = self.skip_logits
+            self.skip_logits = skip_logits
+            s = torch.cuda.Stream(self.device)
+            s.wait_stream(torch.cuda.current_stream(self.device))
+            with torch.cuda.stream(s):
+                self.forward_buffers(self.n_batches)
+            torch.cuda.current_stream(self.device).wait_stream(s)
+            torch.cuda.synchronize(self.device)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self.forward_buffers(self.n_batches)
+            self.skip_logits = saved
+            self._pf_graphs[skip_logits] = g
+            return g
+        except Exception:  # noqa: BLE001
+            self._pf_failed = True
+            self.skip_logits = skip_logits
+            return None
 
     def capture_decode_graph(self):
         """Capture the whole B=1 decode step (forward + pos advance) as a
