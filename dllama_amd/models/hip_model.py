@@ -209,6 +209,7 @@ class HipTransformer:
         self.logits0 = torch.zeros(NB, c.vocab0, device=dev)
         rpw = 2 if c.vocab0 >= 2048 else 1
         self.amax_blocks = -(-c.vocab0 // (4 * rpw))  # mirrors gemv RPW choice
+        self.gemm_part = torch.zeros(16 * 32 * 8192, device=dev)  # K-split partials
         # sum-of-squares accumulators: [slot, batch, 16 spread x 32 pad]
         # (16 slots each on their own cacheline; atomics to one line serialize)
         self.ssq = torch.zeros(2 * c.n_layers + 1, NB, 16 * 32, device=dev)
@@ -274,7 +275,8 @@ class HipTransformer:
         """Batched matmul dispatch: decode batches use the dot4 GEMV,
         prefill batches (>=8) the int8-MFMA GEMM."""
         if NB >= 8:
-            self.k.q40_gemm(lin.qs, lin.scales, qb.q, qb.s, out, NB)
+            self.k.q40_gemm(lin.qs, lin.scales, qb.q, qb.s, out, NB,
+                            self.gemm_part)
         else:
             self.k.q40_gemv(lin.qs, lin.scales, qb.q, qb.s, qb.bs, out, NB, amax)
 

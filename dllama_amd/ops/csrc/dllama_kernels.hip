@@ -912,12 +912,19 @@ k_q40_gemm(const uint8_t *__restrict__ qs,
            const int8_t *__restrict__ xq,
            const float *__restrict__ xs,
            float *__restrict__ y,
+           float *__restrict__ part,
            int d, int n, int batch) {
+    // gridDim.y = K-splits: a pure M decomposition leaves <1 workgroup/CU
+    // for mid-size d (224 wg for the 28672-row W13) — K-split partials
+    // restore occupancy, combined by k_gemm_reduce
     const int wave = threadIdx.x / WAVE;
     const int lane = threadIdx.x % WAVE;
     const int mbase = (blockIdx.x * 4 + wave) * 32;  // 32 weight rows per wave
     if (mbase >= d) return;
     const int nb = n / QB;
+    const int ksplit = gridDim.y;
+    const int j0 = (int)((int64_t)nb * blockIdx.y / ksplit);
+    const int j1 = (int)((int64_t)nb * (blockIdx.y + 1) / ksplit);
     const int khi = lane >> 5;       // 0: elems 0..15 (lo nibbles), 1: 16..31 (hi)
     const int bcol = lane & 31;      // x batch row for the A fragment
     const int mcol = lane & 31;      // weight row within the tile (B operand)
@@ -929,7 +936,7 @@ k_q40_gemm(const uint8_t *__restrict__ qs,
     #pragma unroll
     for (int r = 0; r < 16; r++) facc[r] = 0.0f;
 
-    for (int j = 0; j < nb; j++) {
+    for (int j = j0; j < j1; j++) {
         // lanes 0..31 hold this block's x-scale for batch row `lane`;
         // the descale fetches them cross-lane via shfl (no LDS, no barrier)
         const float sxv = (lane < 32) ? xs[(int64_t)(lane & 31) * nb + j] : 0.0f;
@@ -961,9 +968,27 @@ k_q40_gemm(const uint8_t *__restrict__ qs,
         #pragma unroll
         for (int r = 0; r < 16; r++) {
             const int brow = (r & 3) + 8 * (r >> 2) + 4 * khi;
-            if (brow < batch)
-                y[(int64_t)brow * d + mbase + mcol] = facc[r];
+            if (ksplit == 1) {
+                if (brow < batch)
+                    y[(int64_t)brow * d + mbase + mcol] = facc[r];
+            } else {
+                part[(((int64_t)blockIdx.y * 32) + brow) * d + mbase + mcol] = facc[r];
+            }
         }
+    }
+}
+
+__global__ void k_gemm_reduce(const float *__restrict__ part,
+                              float *__restrict__ y,
+                              int d, int batch, int ksplit) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         i < (int64_t)batch * d; i += (int64_t)gridDim.x * blockDim.x) {
+        const int b = (int)(i / d);
+        const int m = (int)(i % d);
+        float acc = 0.0f;
+        for (int k = 0; k < ksplit; k++)
+            acc += part[(((int64_t)k * 32) + b) * d + m];
+        y[(int64_t)b * d + m] = acc;
     }
 }
 
@@ -1483,7 +1508,8 @@ void q40_gemv_nq_rope(torch::Tensor qs, torch::Tensor scales, torch::Tensor x,
 }
 
 void q40_gemm(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
-              torch::Tensor xs, torch::Tensor y, int64_t batch) {
+              torch::Tensor xs, torch::Tensor y, int64_t batch,
+              c10::optional<torch::Tensor> part = c10::nullopt) {
     // int8-MFMA batched matmul (prefill path); xq/xs must have >=32 rows
     CHECK_CUDA(qs); CHECK_CONT(qs); CHECK_CONT(xq);
     const int d = qs.size(0);
@@ -1491,11 +1517,23 @@ void q40_gemm(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
     TORCH_CHECK(xq.size(-1) == n, "x width mismatch");
     TORCH_CHECK(xq.size(0) >= 32, "gemm needs 32 padded batch rows");
     TORCH_CHECK(n % QB == 0, "n must be a multiple of 32");
-    hipLaunchKernelGGL(k_q40_gemm, dim3(ceil_div(d, 128)), dim3(256), 0,
+    const int mtiles = ceil_div(d, 128);
+    int ksplit = 1;
+    if (part.has_value()) {
+        ksplit = std::max(1, std::min(512 / mtiles, 16));
+        ksplit = std::min<int>(ksplit, n / QB);
+        while (ksplit > 1 && (int64_t)ksplit * 32 * d > part->numel()) ksplit--;
+    }
+    float *pp = ksplit > 1 ? part->data_ptr<float>() : nullptr;
+    hipLaunchKernelGGL(k_q40_gemm, dim3(mtiles, ksplit), dim3(256), 0,
                        cur_stream(), qs.data_ptr<uint8_t>(),
                        reinterpret_cast<const __half *>(scales.data_ptr<at::Half>()),
                        xq.data_ptr<int8_t>(), xs.data_ptr<float>(),
-                       y.data_ptr<float>(), d, n, (int)batch);
+                       y.data_ptr<float>(), pp, d, n, (int)batch);
+    if (ksplit > 1)
+        hipLaunchKernelGGL(k_gemm_reduce, dim3(ceil_div((int64_t)batch * d, 1024)),
+                           dim3(256), 0, cur_stream(), pp, y.data_ptr<float>(),
+                           d, (int)batch, ksplit);
 }
 
 int64_t q40_gemv_argmax_blocks(int64_t d) {
@@ -1893,7 +1931,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("xs"), py::arg("xbs"), py::arg("y"), py::arg("batch"),
           py::arg("amax_slot") = py::none());
     m.def("q40_gemv_resid", &q40_gemv_resid);
-    m.def("q40_gemm", &q40_gemm);
+    m.def("q40_gemm", &q40_gemm, py::arg("qs"), py::arg("scales"),
+          py::arg("xq"), py::arg("xs"), py::arg("y"), py::arg("batch"),
+          py::arg("part") = py::none());
     m.def("q40_gemv_rope", &q40_gemv_rope);
     m.def("q40_gemv_nq", &q40_gemv_nq, py::arg("qs"), py::arg("scales"),
           py::arg("x"), py::arg("wnorm"), py::arg("ssq"), py::arg("eps"),
