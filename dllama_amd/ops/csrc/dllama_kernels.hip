@@ -1520,7 +1520,7 @@ void q40_gemm(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
     const int mtiles = ceil_div(d, 128);
     int ksplit = 1;
     if (part.has_value()) {
-        ksplit = std::max(1, std::min(512 / mtiles, 16));
+        ksplit = std::max(1, std::min(1024 / mtiles, 16));
         ksplit = std::min<int>(ksplit, n / QB);
         while (ksplit > 1 && (int64_t)ksplit * 32 * d > part->numel()) ksplit--;
     }
@@ -1531,7 +1531,7 @@ void q40_gemm(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
                        xq.data_ptr<int8_t>(), xs.data_ptr<float>(),
                        y.data_ptr<float>(), pp, d, n, (int)batch);
     if (ksplit > 1)
-        hipLaunchKernelGGL(k_gemm_reduce, dim3(ceil_div((int64_t)batch * d, 1024)),
+        hipLaunchKernelGGL(k_gemm_reduce, dim3(ceil_div((int64_t)batch * d, 256)),
                            dim3(256), 0, cur_stream(), pp, y.data_ptr<float>(),
                            d, (int)batch, ksplit);
 }
