@@ -85,6 +85,13 @@ def load_engine(args):
         model = CpuTransformer(m, cfg, comm)
     tok = Tokenizer(args.tokenizer) if args.tokenizer else None
     seed = args.seed if args.seed is not None else int(time.time())
+    if comm.world > 1:
+        # all ranks must sample identically (lockstep decode): share rank 0's seed
+        st = torch.tensor([seed], dtype=torch.int64)
+        if torch.cuda.is_available():
+            st = st.cuda()
+        comm.broadcast_(st, src=0)
+        seed = int(st.item())
     sampler = Sampler(m.header.vocab_size, args.temperature, args.topp, seed)
     return InferenceEngine(model, tok, sampler, n_batches=args.n_batches), m, comm
 

@@ -71,3 +71,54 @@ def test_tp2_gloo_matches_tp1(tmp_path):
         assert not isinstance(val, str), val
         assert np.allclose(val, ref, atol=1e-4, rtol=1e-4), \
             f"rank {rank}: {np.abs(val - ref).max()}"
+
+
+def _worker_moe(rank, world, path, port, out_q):
+    try:
+        os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                          WORLD_SIZE=str(world), RANK=str(rank))
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from dllama_amd.models.config import ModelConfig
+        from dllama_amd.models.cpu_model import CpuTransformer
+        from dllama_amd.parallel.comm import DistComm
+        m = mf.ModelFile(path, sync_type=F32)
+        cfg = ModelConfig.from_header(m.header, world, rank)
+        cfg.sync_type = F32
+        model = CpuTransformer(m, cfg, DistComm())
+        logits = model.forward(torch.tensor([1, 5, 9]), torch.arange(3))
+        out_q.put((rank, logits.numpy()))
+        dist.destroy_process_group()
+    except Exception as e:  # noqa: BLE001
+        out_q.put((rank, f"ERROR: {e!r}"))
+
+
+@pytest.mark.timeout(180)
+def test_tp2_gloo_moe(tmp_path):
+    """Qwen3-MoE TP=2 over gloo matches TP=1 (expert slices + redundant
+    gate, reference SURVEY §2.2 EP row)."""
+    from dllama_amd.utils.testing import make_tiny_qwen3
+    path = str(tmp_path / "moe.m")
+    make_tiny_qwen3(path, moe=True)
+    from dllama_amd.models.config import ModelConfig
+    from dllama_amd.models.cpu_model import CpuTransformer
+    m = mf.ModelFile(path, sync_type=F32)
+    cfg = ModelConfig.from_header(m.header)
+    cfg.sync_type = F32
+    ref = CpuTransformer(m, cfg).forward(torch.tensor([1, 5, 9]),
+                                         torch.arange(3)).numpy()
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_moe, args=(r, 2, path, 29573, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, val = q.get(timeout=150)
+        results[rank] = val
+    for p in procs:
+        p.join(timeout=60)
+    for rank, val in results.items():
+        assert not isinstance(val, str), val
+        assert np.allclose(val, ref, atol=1e-4, rtol=1e-4)
