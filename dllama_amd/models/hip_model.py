@@ -209,7 +209,7 @@ class HipTransformer:
         self.logits0 = torch.zeros(NB, c.vocab0, device=dev)
         rpw = 2 if c.vocab0 >= 2048 else 1
         self.amax_blocks = -(-c.vocab0 // (4 * rpw))  # mirrors gemv RPW choice
-        self.gemm_part = torch.zeros(16 * 32 * 8192, device=dev)  # K-split partials
+        self.gemm_part = torch.zeros(32 * 32 * 8192, device=dev)  # K-split partials
         # sum-of-squares accumulators: [slot, batch, 16 spread x 32 pad]
         # (16 slots each on their own cacheline; atomics to one line serialize)
         self.ssq = torch.zeros(2 * c.n_layers + 1, NB, 16 * 32, device=dev)
