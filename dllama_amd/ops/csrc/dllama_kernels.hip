@@ -1520,7 +1520,7 @@ void q40_gemm(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
     const int mtiles = ceil_div(d, 128);
     int ksplit = 1;
     if (part.has_value()) {
-        ksplit = std::max(1, std::min(2048 / mtiles, 32));
+        ksplit = std::max(1, std::min(1024 / mtiles, 16));
         ksplit = std::min<int>(ksplit, n / QB);
         while (ksplit > 1 && (int64_t)ksplit * 32 * d > part->numel()) ksplit--;
     }
