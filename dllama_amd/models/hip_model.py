@@ -395,7 +395,7 @@ class HipTransformer:
         c, k = self.cfg, self.k
         ka = c.n_active_experts
         S = NB * ka
-        torch.matmul(self.t_norm[:NB], lw["gate"].t(), out=self.moe_router[:NB])
+        k.router_gemv(lw["gate"], self.t_norm, self.moe_router, NB)
         k.moe_gate(self.moe_router[:NB], self.moe_idx, self.moe_wts, NB, ka)
         k.q40_gemv_grouped(lw["w13"].qs, lw["w13"].scales, self.xq.q, self.xq.s,
                            self.xq.bs, self.moe_idx[:S], self.moe_out13, ka)

@@ -992,6 +992,27 @@ __global__ void k_gemm_reduce(const float *__restrict__ part,
     }
 }
 
+// f32 router GEMV: logits[b][e] = gate[e,:] . t[b,:]
+__global__ void k_router_gemv(const float *__restrict__ gate,
+                              const float *__restrict__ t,
+                              float *__restrict__ logits,
+                              int n_experts, int dim) {
+    const int e = blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
+    const int b = blockIdx.y;
+    if (e >= n_experts) return;
+    const int lane = threadIdx.x % WAVE;
+    const float4 *g4 = reinterpret_cast<const float4 *>(gate + (int64_t)e * dim);
+    const float4 *t4 = reinterpret_cast<const float4 *>(t + (int64_t)b * dim);
+    float acc = 0.0f;
+    for (int i = lane; i < dim / 4; i += WAVE) {
+        const float4 gv = g4[i];
+        const float4 tv = t4[i];
+        acc += gv.x * tv.x + gv.y * tv.y + gv.z * tv.z + gv.w * tv.w;
+    }
+    acc = wave_reduce_sum(acc);
+    if (lane == 0) logits[(int64_t)b * n_experts + e] = acc;
+}
+
 // MoE router: softmax over n_experts logits, top-k (first-index ties),
 // normalized weights + int32 expert ids (reference OP_SOFTMAX + OP_MOE_GATE,
 // nn-cpu-ops.cpp:1443-1492). One wave per batch row; n_experts <= 1024.
@@ -1021,6 +1042,7 @@ __global__ void k_moe_gate(const float *__restrict__ logits,
     const float inv = 1.0f / sum;
     // iterative top-k: packed (prob, smallest-index-wins) max per round
     float wsum = 0.0f;
+    float chosen[16];  // topk <= 16
     for (int t = 0; t < topk; t++) {
         float best = -1.0f;
         int bi = -1;
@@ -1037,21 +1059,18 @@ __global__ void k_moe_gate(const float *__restrict__ logits,
                 best = ob; bi = oi;
             }
         }
-        if (lane == 0) {
-            idx[(int64_t)b * topk + t] = bi;
-            wts[(int64_t)b * topk + t] = best * inv;
-        }
+        if (lane == 0) idx[(int64_t)b * topk + t] = bi;
+        chosen[t] = best * inv;
         wsum += best * inv;
         // clear the winner
         const int wl = bi / per, wi = bi % per;
         if (lane == wl) v[wi] = -1.0f;
     }
-    // normalize by the top-k sum (reference normTopk)
+    // normalize by the top-k sum (reference normTopk); single write pass
     if (lane == 0) {
-        float t = 0.0f;
-        for (int i = 0; i < topk; i++) t += wts[(int64_t)b * topk + i];
-        const float winv = 1.0f / t;
-        for (int i = 0; i < topk; i++) wts[(int64_t)b * topk + i] *= winv;
+        const float winv = 1.0f / wsum;
+        for (int t = 0; t < topk; t++)
+            wts[(int64_t)b * topk + t] = chosen[t] * winv;
     }
 }
 
@@ -1665,9 +1684,20 @@ void scale_merge_add(torch::Tensor x, torch::Tensor y, torch::Tensor wts,
                      torch::Tensor ssq, int64_t batch, int64_t topk) {
     CHECK_CUDA(x);
     const int n = x.size(-1);
-    hipLaunchKernelGGL(k_scale_merge_add, dim3(ceil_div(n, 1024), batch), dim3(256),
+    hipLaunchKernelGGL(k_scale_merge_add, dim3(ceil_div(n, 256), batch), dim3(256),
                        0, cur_stream(), x.data_ptr<float>(), y.data_ptr<float>(),
                        wts.data_ptr<float>(), ssq.data_ptr<float>(), n, (int)topk);
+}
+
+void router_gemv(torch::Tensor gate, torch::Tensor t, torch::Tensor logits,
+                 int64_t batch) {
+    CHECK_CUDA(gate);
+    const int n_experts = gate.size(0);
+    const int dim = gate.size(1);
+    hipLaunchKernelGGL(k_router_gemv, dim3(ceil_div(n_experts, 4), batch),
+                       dim3(256), 0, cur_stream(), gate.data_ptr<float>(),
+                       t.data_ptr<float>(), logits.data_ptr<float>(),
+                       n_experts, dim);
 }
 
 void norm_f32(torch::Tensor x, torch::Tensor w, torch::Tensor ssq,
@@ -1944,6 +1974,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("eps"), py::arg("yout") = py::none());
     m.def("moe_gate", &moe_gate);
     m.def("scale_merge_add", &scale_merge_add);
+    m.def("router_gemv", &router_gemv);
     m.def("norm_f32", &norm_f32);
     m.def("add_ssq", &add_ssq);
     m.def("q40_gemv_grouped", &q40_gemv_grouped);
