@@ -936,25 +936,50 @@ k_q40_gemm(const uint8_t *__restrict__ qs,
     #pragma unroll
     for (int r = 0; r < 16; r++) facc[r] = 0.0f;
 
-    for (int j = j0; j < j1; j++) {
-        // lanes 0..31 hold this block's x-scale for batch row `lane`;
-        // the descale fetches them cross-lane via shfl (no LDS, no barrier)
-        const float sxv = (lane < 32) ? xs[(int64_t)(lane & 31) * nb + j] : 0.0f;
-        // A: x int8, batch row bcol, 16 bytes at k = khi*16
-        v4i32_t a = *reinterpret_cast<const v4i32_t *>(
-            xq + (int64_t)bcol * n + j * QB + khi * 16);
-        // B: weight nibbles -> signed i8 (nib - 8), bytewise:
-        //   s = nib ^ 8 gives |result| bits; negative bytes (nib<8) OR in 0xF0
-        const uint4 wq = wrow[j];
+    // 2-block unroll: two independent load->MFMA->descale chains keep two
+    // 16B weight loads in flight per lane (single-chain was latency-bound:
+    // MfmaUtil 1.5%)
+    auto extract = [&](const uint4 &wq, v4i32_t &b) {
         const uint32_t wv[4] = {wq.x, wq.y, wq.z, wq.w};
-        v4i32_t b;
         #pragma unroll
         for (int t = 0; t < 4; t++) {
-            uint32_t s = khi ? ((wv[t] >> 4) & 0x0F0F0F0Fu)
-                             : (wv[t] & 0x0F0F0F0Fu);
-            s ^= 0x08080808u;
-            b[t] = (int)(s | (((s >> 3) & 0x01010101u) * 0xF0u));
+            uint32_t sx_ = khi ? ((wv[t] >> 4) & 0x0F0F0F0Fu)
+                               : (wv[t] & 0x0F0F0F0Fu);
+            sx_ ^= 0x08080808u;
+            b[t] = (int)(sx_ | (((sx_ >> 3) & 0x01010101u) * 0xF0u));
         }
+    };
+    int j = j0;
+    for (; j + 1 < j1; j += 2) {
+        const float sxv0 = (lane < 32) ? xs[(int64_t)(lane & 31) * nb + j] : 0.0f;
+        const float sxv1 = (lane < 32) ? xs[(int64_t)(lane & 31) * nb + j + 1] : 0.0f;
+        const uint4 wq0 = wrow[j];
+        const uint4 wq1 = wrow[j + 1];
+        v4i32_t a0 = *reinterpret_cast<const v4i32_t *>(
+            xq + (int64_t)bcol * n + j * QB + khi * 16);
+        v4i32_t a1 = *reinterpret_cast<const v4i32_t *>(
+            xq + (int64_t)bcol * n + (j + 1) * QB + khi * 16);
+        v4i32_t b0, b1;
+        extract(wq0, b0);
+        extract(wq1, b1);
+        v16i32_t i0 = {}, i1 = {};
+        i0 = __builtin_amdgcn_mfma_i32_32x32x32_i8(a0, b0, i0, 0, 0, 0);
+        i1 = __builtin_amdgcn_mfma_i32_32x32x32_i8(a1, b1, i1, 0, 0, 0);
+        const float sw0 = __half2float(srow[j]);
+        const float sw1 = __half2float(srow[j + 1]);
+        #pragma unroll
+        for (int r = 0; r < 16; r++) {
+            const int brow = (r & 3) + 8 * (r >> 2) + 4 * khi;
+            facc[r] = fmaf((float)i0[r], sw0 * __shfl(sxv0, brow, WAVE), facc[r]);
+            facc[r] = fmaf((float)i1[r], sw1 * __shfl(sxv1, brow, WAVE), facc[r]);
+        }
+    }
+    for (; j < j1; j++) {
+        const float sxv = (lane < 32) ? xs[(int64_t)(lane & 31) * nb + j] : 0.0f;
+        v4i32_t a = *reinterpret_cast<const v4i32_t *>(
+            xq + (int64_t)bcol * n + j * QB + khi * 16);
+        v4i32_t b;
+        extract(wrow[j], b);
         v16i32_t iacc = {};
         iacc = __builtin_amdgcn_mfma_i32_32x32x32_i8(a, b, iacc, 0, 0, 0);
         const float sw = __half2float(srow[j]);
