@@ -149,3 +149,33 @@ def test_converter_roundtrip(tmp_path):
     model = CpuTransformer(m, ModelConfig.from_header(m.header))
     logits = model.forward(torch.tensor([1, 2, 3]), torch.arange(3))
     assert torch.isfinite(logits).all()
+
+
+def test_api_prefix_cache_correctness(assets):
+    """KV prefix reuse must not change outputs: a fresh engine and a
+    prefix-cached engine must produce identical greedy continuations."""
+    import torch
+    from dllama_amd import model_file as mflib
+    from dllama_amd.models.config import ModelConfig
+    from dllama_amd.models.cpu_model import CpuTransformer
+    from dllama_amd.engine import InferenceEngine
+    from dllama_amd.tokenizer import Sampler
+    mp_, tp_ = assets
+    m = mflib.ModelFile(mp_)
+    cfg = ModelConfig.from_header(m.header)
+
+    prompt_a = [1, 2, 3, 4, 5]
+    prompt_b = prompt_a + [6, 7, 8]  # shares a 5-token prefix
+
+    # fresh engine on prompt_b
+    e1 = InferenceEngine(CpuTransformer(m, cfg),
+                         sampler=Sampler(m.header.vocab_size, 0.0, 0.9, 1))
+    out_fresh, _ = e1.generate(prompt_b, 6)
+
+    # cached engine: run prompt_a first, then reuse the prefix for prompt_b
+    e2 = InferenceEngine(CpuTransformer(m, cfg),
+                         sampler=Sampler(m.header.vocab_size, 0.0, 0.9, 1))
+    e2.generate(prompt_a, 2)
+    e2.reset(len(prompt_a))  # NaiveCache-style: prefix of prompt_b is cached
+    out_cached, _ = e2.generate(prompt_b[len(prompt_a):], 6)
+    assert out_fresh == out_cached
