@@ -1048,19 +1048,23 @@ __global__ void k_moe_gate(const float *__restrict__ logits,
     const int b = blockIdx.x;
     const int lane = threadIdx.x;  // blockDim == 64
     const int per = (n_experts + WAVE - 1) / WAVE;
+    // fixed-trip unrolled loops: runtime-indexed arrays go to scratch
     float v[16];  // per-lane expert logits (supports n_experts <= 1024)
     float m = -1e30f;
-    for (int i = 0; i < per; i++) {
+    #pragma unroll
+    for (int i = 0; i < 16; i++) {
         const int eidx = lane * per + i;
-        v[i] = eidx < n_experts ? logits[(int64_t)b * n_experts + eidx] : -1e30f;
+        v[i] = (i < per && eidx < n_experts)
+                   ? logits[(int64_t)b * n_experts + eidx] : -1e30f;
         m = fmaxf(m, v[i]);
     }
     #pragma unroll
     for (int off = 32; off > 0; off >>= 1)
         m = fmaxf(m, __shfl_xor(m, off, WAVE));
     float sum = 0.0f;
-    for (int i = 0; i < per; i++) {
-        v[i] = (lane * per + i) < n_experts ? __expf(v[i] - m) : 0.0f;
+    #pragma unroll
+    for (int i = 0; i < 16; i++) {
+        v[i] = (i < per && (lane * per + i) < n_experts) ? __expf(v[i] - m) : 0.0f;
         sum += v[i];
     }
     sum = wave_reduce_sum(sum);
@@ -1068,12 +1072,14 @@ __global__ void k_moe_gate(const float *__restrict__ logits,
     // iterative top-k: packed (prob, smallest-index-wins) max per round
     float wsum = 0.0f;
     float chosen[16];  // topk <= 16
-    for (int t = 0; t < topk; t++) {
+    #pragma unroll
+    for (int t = 0; t < 16; t++) {
+        if (t >= topk) break;
         float best = -1.0f;
         int bi = -1;
-        for (int i = 0; i < per; i++) {
-            const int eidx = lane * per + i;
-            if (v[i] > best) { best = v[i]; bi = eidx; }
+        #pragma unroll
+        for (int i = 0; i < 16; i++) {
+            if (i < per && v[i] > best) { best = v[i]; bi = lane * per + i; }
         }
         // wave argmax: (prob, -index) lexicographic via packed compare
         #pragma unroll
@@ -1087,15 +1093,19 @@ __global__ void k_moe_gate(const float *__restrict__ logits,
         if (lane == 0) idx[(int64_t)b * topk + t] = bi;
         chosen[t] = best * inv;
         wsum += best * inv;
-        // clear the winner
-        const int wl = bi / per, wi = bi % per;
-        if (lane == wl) v[wi] = -1.0f;
+        // clear the winner (static-index scan, see scratch note above)
+        #pragma unroll
+        for (int i = 0; i < 16; i++)
+            if (lane * per + i == bi) v[i] = -1.0f;
     }
     // normalize by the top-k sum (reference normTopk); single write pass
     if (lane == 0) {
         const float winv = 1.0f / wsum;
-        for (int t = 0; t < topk; t++)
+        #pragma unroll
+        for (int t = 0; t < 16; t++) {
+            if (t >= topk) break;
             wts[(int64_t)b * topk + t] = chosen[t] * winv;
+        }
     }
 }
 
