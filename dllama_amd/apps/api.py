@@ -72,7 +72,10 @@ class ApiState:
             self.engine.sampler.set_temp(float(body["temperature"]))
         if body.get("seed") is not None:
             self.engine.sampler.set_seed(int(body["seed"]))
-        stops = chat_stops(self.tok) + (body.get("stop") or [])
+        user_stop = body.get("stop") or []
+        if isinstance(user_stop, str):  # OpenAI allows string or list
+            user_stop = [user_stop]
+        stops = chat_stops(self.tok) + user_stop
         detector = EosDetector(self.tok.eos_token_ids, stops)
         out_text = []
 
@@ -158,7 +161,11 @@ class Handler(BaseHTTPRequestHandler):
                 self._chunk("data: [DONE]\n\n")
                 self.wfile.write(b"0\r\n\r\n")
         else:
-            text, n_prompt, n_gen = STATE.complete(body, lambda d: None)
+            try:
+                text, n_prompt, n_gen = STATE.complete(body, lambda d: None)
+            except Exception as e:  # noqa: BLE001
+                self._json(500, {"error": {"message": str(e), "type": "server_error"}})
+                return
             self._json(200, {
                 "id": rid, "object": "chat.completion", "created": created,
                 "model": STATE.model_name,
