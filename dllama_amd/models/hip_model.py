@@ -97,6 +97,12 @@ class HipTransformer:
     @classmethod
     def from_file(cls, m: ModelFile, config: ModelConfig, device=None,
                   comm: Comm | None = None, n_batches: int = 32) -> "HipTransformer":
+        from ..quants import Q40
+        if m.header.weight_type != Q40:
+            raise ValueError(
+                "the MI355X HIP backend runs Q40 weights (the reference's "
+                "shipped format); f32/q80 .m files run on the CPU backend "
+                "(--gpu-index -1) or can be re-quantized with convert_hf.py")
         self = cls(config, device, comm, n_batches)
         c, dev = config, self.device
         r, w = c.rank, c.world

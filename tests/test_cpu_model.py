@@ -214,3 +214,36 @@ def test_engine_greedy_deterministic(tiny):
     assert out1 == out2
     assert len(out1) == 8
     assert stats.prefill_tokens == 3 and stats.decode_tokens == 8
+
+
+def test_q80_weight_model_runs(tmp_path):
+    """The reference supports q80 (and f32) weight files; the CPU backend
+    runs them (HIP backend raises a clear error, guarded in hip_model)."""
+    from dllama_amd.quants import Q80 as Q80T
+    h = mf.LlmHeader(arch_type=mf.ARCH_LLAMA, dim=64, hidden_dim=128, n_layers=1,
+                     n_heads=4, n_kv_heads=2, head_dim=64, vocab_size=128,
+                     seq_len=64, rope_theta=10000, rope_type=mf.ROPE_LLAMA,
+                     weight_type=Q80T)
+    h.finalize()
+    path = str(tmp_path / "q80.m")
+    mf.write_synthetic_model(path, h)
+    m = mf.ModelFile(path)
+    assert m.header.weight_type == Q80T
+    model = CpuTransformer(m, ModelConfig.from_header(m.header))
+    logits = model.forward(torch.tensor([1, 2]), torch.arange(2))
+    assert torch.isfinite(logits).all()
+
+
+def test_f32_weight_model_runs(tmp_path):
+    from dllama_amd.quants import F32 as F32T
+    h = mf.LlmHeader(arch_type=mf.ARCH_LLAMA, dim=64, hidden_dim=128, n_layers=1,
+                     n_heads=4, n_kv_heads=2, head_dim=64, vocab_size=128,
+                     seq_len=64, rope_theta=10000, rope_type=mf.ROPE_LLAMA,
+                     weight_type=F32T)
+    h.finalize()
+    path = str(tmp_path / "f32.m")
+    mf.write_synthetic_model(path, h)
+    m = mf.ModelFile(path)
+    model = CpuTransformer(m, ModelConfig.from_header(m.header))
+    logits = model.forward(torch.tensor([1, 2]), torch.arange(2))
+    assert torch.isfinite(logits).all()
