@@ -77,10 +77,12 @@ class ApiState:
             user_stop = [user_stop]
         stops = chat_stops(self.tok) + user_stop
         detector = EosDetector(self.tok.eos_token_ids, stops)
+        self.tok.reset_decoder()
         out_text = []
 
         def on_token(t):
-            piece = self.tok.piece(t).decode("utf-8", "replace")
+            # streaming-decoder first: UTF-8-safe pieces into the detector
+            piece = self.tok.decode(t)
             kind = detector.append(t, piece)
             if kind != 0:  # not MAYBE_EOS
                 delta = detector.get_delta()

@@ -170,7 +170,9 @@ def run_chat(args) -> int:
             print("🤖 ", end="", flush=True)
 
         def on_token(t):
-            piece = tok.piece(t).decode("utf-8", "replace")
+            # route through the streaming decoder first (UTF-8-safe pieces;
+            # reference feeds tokenizer->decode output to the detector)
+            piece = tok.decode(t)
             kind = detector.append(t, piece)
             if not quiet:
                 delta = detector.get_delta()

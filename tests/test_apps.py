@@ -179,3 +179,24 @@ def test_api_prefix_cache_correctness(assets):
     e2.reset(len(prompt_a))  # NaiveCache-style: prefix of prompt_b is cached
     out_cached, _ = e2.generate(prompt_b[len(prompt_a):], 6)
     assert out_fresh == out_cached
+
+
+def test_cli_chat(assets, capsys, monkeypatch):
+    """Chat REPL smoke: one user turn then EOF exits cleanly."""
+    from dllama_amd.apps import main as main_mod
+    mp, tp = assets
+    answers = iter(["hello there"])
+
+    def fake_input(prompt=""):
+        try:
+            return next(answers)
+        except StopIteration:
+            raise EOFError
+
+    monkeypatch.setattr("builtins.input", fake_input)
+    rc = main_mod.main(["chat", "--model", mp, "--tokenizer", tp,
+                        "--steps", "8", "--temperature", "0",
+                        "--gpu-index", "-1"])
+    assert rc == 0
+    out = capsys.readouterr().out
+    assert "🤖" in out
