@@ -105,9 +105,12 @@ class InferenceEngine:
         out.append(token)
         if on_token:
             on_token(token)
+        seq_len = getattr(getattr(self.model, "cfg", None), "seq_len", None)
         for _ in range(max_tokens - 1):
             if stop_check and stop_check(token):
                 break
+            if seq_len is not None and self.pos >= seq_len:
+                break  # context window exhausted (reference clamps via seqLen)
             logits = self.decode_one(token)
             token = sampler.sample(logits)
             out.append(token)

@@ -89,6 +89,9 @@ class CpuTransformer:
         """tokens, positions: int64 [B] -> logits f32 [B, vocab] (all ranks)."""
         c = self.cfg
         B = tokens.shape[0]
+        if int(positions[-1]) >= c.seq_len:
+            raise ValueError(
+                f"position {int(positions[-1])} exceeds seq_len {c.seq_len}")
         x = self.embedding[tokens.long()].clone()  # [B, dim], replicated
 
         for l, lw in enumerate(self.layers):

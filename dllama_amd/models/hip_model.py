@@ -430,6 +430,10 @@ class HipTransformer:
         B = tokens.shape[0]
         assert B <= self.n_batches
         p0 = int(positions[0])
+        if p0 + B > self.cfg.seq_len:
+            raise ValueError(
+                f"position {p0}+{B} exceeds seq_len {self.cfg.seq_len} "
+                "(rebuild with a larger --max-seq-len / seq_len)")
         self.tokens[:B].copy_(tokens.to(self.device), non_blocking=True)
         if B == 1 and self._graph is not None:
             if p0 != self._graph_pos:

@@ -247,3 +247,20 @@ def test_f32_weight_model_runs(tmp_path):
     model = CpuTransformer(m, ModelConfig.from_header(m.header))
     logits = model.forward(torch.tensor([1, 2]), torch.arange(2))
     assert torch.isfinite(logits).all()
+
+
+def test_seq_len_guard(tiny):
+    cfg = ModelConfig.from_header(tiny.header)
+    model = CpuTransformer(tiny, cfg)
+    with pytest.raises(ValueError):
+        model.forward(torch.tensor([1]), torch.tensor([cfg.seq_len]))
+
+
+def test_engine_stops_at_seq_len(tmp_path):
+    path = str(tmp_path / "short.m")
+    make_tiny_llama(path, vocab_size=256, seq_len=16)
+    m = mf.ModelFile(path)
+    cfg = ModelConfig.from_header(m.header)
+    eng = InferenceEngine(CpuTransformer(m, cfg))
+    out, _ = eng.generate([1, 2, 3, 4], 64)  # asks for more than fits
+    assert len(out) <= 16 - 4 + 1
