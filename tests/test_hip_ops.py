@@ -345,3 +345,43 @@ def test_q40_gemm_matches_gemv(k):
     want = R.q40_matmul(x[:B].cpu(), wref)
     tol = want.abs().max().item() * 0.02 + 1e-3
     assert torch.allclose(y_gemm[:B].cpu(), want, atol=tol)
+
+
+def test_attn_very_long_context(k):
+    """pos ~2000: many tiles per split; exercises the 4-t-per-wave loop."""
+    B, H0, hd, n_kv0, seq = 1, 4, 128, 2, 2048
+    kv_dim0 = n_kv0 * hd
+    kc = rand(seq, kv_dim0, seed=170, scale=0.3)
+    vc = rand(seq, kv_dim0, seed=171)
+    q = rand(B, H0 * hd, seed=172)
+    y = torch.zeros(B, H0 * hd, device=DEV)
+    pos = torch.tensor([2000], dtype=torch.int32, device=DEV)
+    S = 8
+    ml = torch.zeros(B * H0 * S * 2, device=DEV)
+    osc = torch.zeros(B * H0 * S * hd, device=DEV)
+    cnt = torch.zeros(B * H0, dtype=torch.int32, device=DEV)
+    k.attn(q, H0 * hd, kc, vc, y, pos, B, H0, H0 // n_kv0, hd, S, ml, osc, cnt)
+    want = R.attention(q.cpu(), kc.cpu(), vc.cpu(), torch.tensor([2000]), H0, hd)
+    assert torch.allclose(y.cpu(), want, atol=1e-4, rtol=1e-3)
+
+
+def test_router_gemv(k):
+    E, dim, B = 128, 2048, 2
+    gate = rand(E, dim, seed=180, scale=0.1)
+    t = rand(32, dim, seed=181)
+    out = torch.zeros(B, E, device=DEV)
+    k.router_gemv(gate, t, out, B)
+    want = t[:B].cpu() @ gate.cpu().t()
+    assert torch.allclose(out.cpu(), want, atol=1e-3, rtol=1e-4)
+
+
+def test_moe_gate_ties_prefer_first(k):
+    """Equal probabilities: the smaller expert index must win (reference
+    topk_F32 stable sort semantics)."""
+    B, E, topk = 1, 64, 4
+    logits = torch.zeros(B, E, device=DEV)  # all equal
+    idx = torch.zeros(B * topk, dtype=torch.int32, device=DEV)
+    wts = torch.zeros(B, topk, device=DEV)
+    k.moe_gate(logits, idx, wts, B, topk)
+    assert idx.cpu().tolist() == [0, 1, 2, 3]
+    assert torch.allclose(wts.cpu(), torch.full((B, topk), 0.25))
