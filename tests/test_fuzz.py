@@ -1,0 +1,67 @@
+"""Property-based tests (hypothesis): quant round-trips and tokenizer
+encode/decode over arbitrary inputs."""
+
+import numpy as np
+from hypothesis import given, settings, strategies as st
+
+from dllama_amd import quants
+
+
+@settings(max_examples=50, deadline=None)
+@given(st.lists(st.floats(min_value=-1e4, max_value=1e4, allow_nan=False,
+                           width=32), min_size=32, max_size=256))
+def test_q80_roundtrip_bounded_error(vals):
+    x = np.array(vals[: len(vals) // 32 * 32], dtype=np.float32)
+    if x.size == 0:
+        return
+    y = quants.q80_roundtrip(x)
+    blocks = x.reshape(-1, 32)
+    amax = np.abs(blocks).max(axis=1, keepdims=True)
+    # per-element error bounded by half a quantization step per block
+    assert np.all(np.abs(y.reshape(-1, 32) - blocks) <= amax * (0.51 / 127.0 + 2**-10) + 1e-6)
+
+
+@settings(max_examples=50, deadline=None)
+@given(st.lists(st.floats(min_value=-1e4, max_value=1e4, allow_nan=False,
+                           width=32), min_size=32, max_size=256))
+def test_q40_roundtrip_bounded_error(vals):
+    x = np.array(vals[: len(vals) // 32 * 32], dtype=np.float32)
+    if x.size == 0:
+        return
+    y = quants.q40_roundtrip(x)
+    blocks = x.reshape(-1, 32)
+    amax = np.abs(blocks).max(axis=1, keepdims=True)
+    # 4-bit: |err| <= step (amax/8) * (0.5 + f16-scale slack)
+    # full-step slack: the writer's trunc-based nibble rounding
+    assert np.all(np.abs(y.reshape(-1, 32) - blocks)
+                  <= amax * (1.2 / 8.0 + 2**-9) + 1e-6)
+
+
+_TOK = None
+
+
+def _get_tok():
+    global _TOK
+    if _TOK is None:
+        import tempfile, os
+        from dllama_amd.utils.testing import make_byte_tokenizer
+        from dllama_amd.tokenizer import Tokenizer
+        d = tempfile.mkdtemp()
+        p = os.path.join(d, "t.t")
+        make_byte_tokenizer(p)
+        _TOK = Tokenizer(p)
+    return _TOK
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.text(min_size=0, max_size=200))
+def test_tokenizer_encode_decode_roundtrip(text):
+    tok = _get_tok()
+    ids = tok.encode(text, is_start=False, add_special_tokens=False)
+    tok.reset_decoder()
+    out = "".join(p for p in (tok.decode(t) for t in ids) if p)
+    # flush any buffered partial sequence via an eos-style drain
+    if tok._decode_buf:
+        out += bytes(tok._decode_buf).decode("utf-8", "replace")
+        tok.reset_decoder()
+    assert out == text
