@@ -1074,7 +1074,7 @@ __global__ void k_moe_gate(const float *__restrict__ logits,
     float chosen[16];  // topk <= 16
     #pragma unroll
     for (int t = 0; t < 16; t++) {
-        if (t >= topk) break;
+        if (t >= topk) continue;  // guarded full unroll (break blocks unrolling)
         float best = -1.0f;
         int bi = -1;
         #pragma unroll
@@ -1103,8 +1103,8 @@ __global__ void k_moe_gate(const float *__restrict__ logits,
         const float winv = 1.0f / wsum;
         #pragma unroll
         for (int t = 0; t < 16; t++) {
-            if (t >= topk) break;
-            wts[(int64_t)b * topk + t] = chosen[t] * winv;
+            if (t < topk)
+                wts[(int64_t)b * topk + t] = chosen[t] * winv;
         }
     }
 }
