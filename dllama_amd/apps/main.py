@@ -53,6 +53,8 @@ def build_parser() -> argparse.ArgumentParser:
                    help="max prompt tokens per prefill step (reference nBatches)")
     p.add_argument("--no-graph", action="store_true",
                    help="disable hipGraph decode capture")
+    p.add_argument("--show-timing", action="store_true",
+                   help="per-token generation time (reference 🔶 lines)")
     # accepted for reference CLI parity; meaningless on one xGMI node
     p.add_argument("--workers", nargs="*", default=None,
                    help="ignored: TP ranks come from torchrun (xGMI, not TCP)")
@@ -119,8 +121,16 @@ def run_inference(args) -> int:
     if tok:
         tok.reset_decoder()
 
+    last = [time.perf_counter()]
+
     def on_token(t):
         if quiet or not tok:
+            return
+        if args.show_timing:
+            now = time.perf_counter()
+            print(f"🔶 P {1000 * (now - last[0]):6.2f} ms - {t}", flush=True)
+            last[0] = now
+            tok.decode(t)
             return
         piece = tok.decode(t)
         if piece:
