@@ -374,7 +374,8 @@ class HipTransformer:
                 norm_gemv(lw["w13"], lw["norm1"], slot, self.ff_out)
                 k.swiglu_q80(self.ff_out, self.ff_out[:, c.ff_dim0:],
                              2 * c.ff_dim0, c.ff_dim0, NB, self.dq.q[:NB],
-                             self.dq.s[:NB], self.dq.bs[:NB])
+                             self.dq.s[:NB], self.dq.bs[:NB],
+                             c.hidden_act == HIDDEN_ACT_GELU)
                 self._proj_merge(lw["w2"], self.dq, slot + 1, NB)
             slot += 1
 

@@ -1249,6 +1249,7 @@ __global__ void k_kv_append(const float *__restrict__ k,
 // ------------------------------------------------------------------ swiglu (+q80)
 // d = silu(a) * g, quantized straight to Q80 (reference OP_SILU + OP_MUL +
 // OP_CAST fused; silu nn-cpu-ops.cpp:462-500).
+template <bool GELU>
 __global__ void k_swiglu_q80(const float *__restrict__ a,
                              const float *__restrict__ g,
                              int lda, int n,
@@ -1264,7 +1265,11 @@ __global__ void k_swiglu_q80(const float *__restrict__ a,
     const int i = (gid % nb) * QB + lane;
     float av = a[(int64_t)r * lda + i];
     float gv = g[(int64_t)r * lda + i];
-    float v = av / (1.0f + __expf(-av)) * gv;
+    // silu (reference nn-cpu-ops.cpp:462) or tanh-approx gelu (:454)
+    float act = GELU
+        ? 0.5f * av * (1.0f + tanhf(0.797884560802865f * (av + 0.044715f * av * av * av)))
+        : av / (1.0f + __expf(-av));
+    float v = act * gv;
     float amax = group32_reduce_max(fabsf(v));
     float d = amax / 127.0f;
     float inv = d > 0.0f ? 1.0f / d : 0.0f;
@@ -1755,14 +1760,22 @@ void add_ssq(torch::Tensor x, torch::Tensor p, torch::Tensor ssq, int64_t batch)
 }
 
 void swiglu_q80(torch::Tensor a, torch::Tensor g, int64_t lda, int64_t n,
-                int64_t rows, torch::Tensor q, torch::Tensor s, torch::Tensor bs) {
+                int64_t rows, torch::Tensor q, torch::Tensor s, torch::Tensor bs,
+                bool gelu = false) {
     CHECK_CUDA(a);
     const int64_t blocks = rows * (n / QB);
-    hipLaunchKernelGGL(k_swiglu_q80, dim3(ceil_div(blocks * 32, 256)), dim3(256), 0,
-                       cur_stream(), a.data_ptr<float>(), g.data_ptr<float>(),
-                       (int)lda, (int)n,
-                       q.data_ptr<int8_t>(), s.data_ptr<float>(), bs.data_ptr<float>(),
-                       (int)blocks);
+    if (gelu)
+        hipLaunchKernelGGL(k_swiglu_q80<true>, dim3(ceil_div(blocks * 32, 256)),
+                           dim3(256), 0, cur_stream(), a.data_ptr<float>(),
+                           g.data_ptr<float>(), (int)lda, (int)n,
+                           q.data_ptr<int8_t>(), s.data_ptr<float>(),
+                           bs.data_ptr<float>(), (int)blocks);
+    else
+        hipLaunchKernelGGL(k_swiglu_q80<false>, dim3(ceil_div(blocks * 32, 256)),
+                           dim3(256), 0, cur_stream(), a.data_ptr<float>(),
+                           g.data_ptr<float>(), (int)lda, (int)n,
+                           q.data_ptr<int8_t>(), s.data_ptr<float>(),
+                           bs.data_ptr<float>(), (int)blocks);
 }
 
 void rmsnorm_rows_s(torch::Tensor buf, int64_t ld, int64_t off, int64_t heads,
@@ -2023,7 +2036,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("zq") = py::none(), py::arg("zs") = py::none(),
           py::arg("zbs") = py::none());
     m.def("embed_gather", &embed_gather, py::arg("table"), py::arg("tokens"), py::arg("x"), py::arg("batch"), py::arg("ssq") = py::none());
-    m.def("swiglu_q80", &swiglu_q80);
+    m.def("swiglu_q80", &swiglu_q80, py::arg("a"), py::arg("g"),
+          py::arg("lda"), py::arg("n"), py::arg("rows"), py::arg("q"),
+          py::arg("s"), py::arg("bs"), py::arg("gelu") = false);
     m.def("silu_mul", &silu_mul);
     m.def("sync_pack", &sync_pack);
     m.def("merge_add", &merge_add, py::arg("x"), py::arg("bufs"), py::arg("ssq") = py::none());

@@ -200,3 +200,46 @@ def test_cli_chat(assets, capsys, monkeypatch):
     assert rc == 0
     out = capsys.readouterr().out
     assert "🤖" in out
+
+
+def test_convert_llama_pth(tmp_path):
+    """Fabricated Meta consolidated.pth -> convert_llama -> runtime load."""
+    import numpy as np
+    import torch
+    import sys, os
+    sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "converter"))
+    import convert_llama
+
+    dim, hidden, layers, heads, kv = 64, 96, 2, 4, 2
+    vocab = 128
+    hd = dim // heads
+    g = torch.Generator().manual_seed(5)
+    sd = {"tok_embeddings.weight": torch.randn(vocab, dim, generator=g) * 0.02,
+          "norm.weight": torch.ones(dim),
+          "output.weight": torch.randn(vocab, dim, generator=g) * 0.05}
+    for l in range(layers):
+        p = f"layers.{l}"
+        sd[f"{p}.attention.wq.weight"] = torch.randn(dim, dim, generator=g) * 0.05
+        sd[f"{p}.attention.wk.weight"] = torch.randn(kv * hd, dim, generator=g) * 0.05
+        sd[f"{p}.attention.wv.weight"] = torch.randn(kv * hd, dim, generator=g) * 0.05
+        sd[f"{p}.attention.wo.weight"] = torch.randn(dim, dim, generator=g) * 0.05
+        sd[f"{p}.feed_forward.w1.weight"] = torch.randn(hidden, dim, generator=g) * 0.05
+        sd[f"{p}.feed_forward.w2.weight"] = torch.randn(dim, hidden, generator=g) * 0.05
+        sd[f"{p}.feed_forward.w3.weight"] = torch.randn(hidden, dim, generator=g) * 0.05
+        sd[f"{p}.attention_norm.weight"] = torch.ones(dim)
+        sd[f"{p}.ffn_norm.weight"] = torch.ones(dim)
+    torch.save(sd, str(tmp_path / "consolidated.00.pth"))
+    (tmp_path / "params.json").write_text(json.dumps(
+        {"dim": dim, "n_heads": heads, "n_kv_heads": kv, "n_layers": layers,
+         "vocab_size": vocab, "norm_eps": 1e-5, "max_seq_len": 128}))
+    out = str(tmp_path / "out.m")
+    convert_llama.convert(str(tmp_path), convert_llama.FLOAT_TYPES["q40"], out)
+
+    from dllama_amd import model_file as mflib
+    from dllama_amd.models.config import ModelConfig
+    from dllama_amd.models.cpu_model import CpuTransformer
+    m = mflib.ModelFile(out)
+    model = CpuTransformer(m, ModelConfig.from_header(m.header))
+    logits = model.forward(torch.tensor([1, 2, 3]), torch.arange(3))
+    assert torch.isfinite(logits).all()
