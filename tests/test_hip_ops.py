@@ -385,3 +385,15 @@ def test_moe_gate_ties_prefer_first(k):
     k.moe_gate(logits, idx, wts, B, topk)
     assert idx.cpu().tolist() == [0, 1, 2, 3]
     assert torch.allclose(wts.cpu(), torch.full((B, topk), 0.25))
+
+
+def test_swiglu_q80_gelu(k):
+    a = rand(2, 256, seed=190)
+    g = rand(2, 256, seed=191)
+    q = torch.zeros(2, 256, dtype=torch.int8, device=DEV)
+    s = torch.zeros(2, 8, device=DEV)
+    bs = torch.zeros(2, 8, device=DEV)
+    k.swiglu_q80(a, g, 256, 256, 2, q, s, bs, True)
+    want = R.gelu(a.cpu()) * g.cpu()
+    got = R.q80_dequantize(q.cpu(), s.cpu())
+    assert torch.allclose(got, want, atol=want.abs().max().item() / 80)
