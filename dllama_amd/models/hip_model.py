@@ -220,7 +220,9 @@ class HipTransformer:
         # (16 slots each on their own cacheline; atomics to one line serialize)
         self.ssq = torch.zeros(2 * c.n_layers + 1, NB, 16 * 32, device=dev)
         self.amax_scratch = torch.zeros(self.amax_blocks, dtype=torch.int64, device=dev)
-        self.attn_splits = 8  # S=16,32 measured slower (combine reads S partials)
+        import os as _os
+        self.attn_splits = int(_os.environ.get("DLLAMA_ATTN_SPLITS", "8"))
+        # S=8 measured best at decode (16/32: combine reads S partials)
         self.attn_ml = torch.zeros(NB * c.n_heads0 * self.attn_splits * 2, device=dev)
         self.attn_o = torch.zeros(NB * c.n_heads0 * self.attn_splits * c.head_dim,
                                   device=dev)
