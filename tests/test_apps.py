@@ -243,3 +243,67 @@ def test_convert_llama_pth(tmp_path):
     model = CpuTransformer(m, ModelConfig.from_header(m.header))
     logits = model.forward(torch.tensor([1, 2, 3]), torch.arange(3))
     assert torch.isfinite(logits).all()
+
+
+def test_convert_tokenizer_hf(tmp_path):
+    """Fabricated HF byte-level BPE tokenizer.json -> .t -> encode parity
+    with the HF fast tokenizer."""
+    import sys, os
+    sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "converter"))
+    import convert_tokenizer_hf as ctok
+
+    # minimal byte-level BPE: 256 byte-alphabet symbols + a few merges
+    utb = ctok.unicode_to_bytes()
+    btu = {v: k for k, v in utb.items()}
+    vocab = {btu[b]: b for b in range(256)}
+    merges = []
+    nxt = 256
+    for pair in [("h", "e"), ("l", "l"), ("he", "ll"), ("hell", "o")]:
+        merges.append(f"{pair[0]} {pair[1]}")
+        vocab[pair[0] + pair[1]] = nxt
+        nxt += 1
+    vocab["<|bos|>"] = nxt
+    vocab["<|eos|>"] = nxt + 1
+    tok_json = {
+        "version": "1.0",
+        "truncation": None, "padding": None,
+        "added_tokens": [
+            {"id": nxt, "content": "<|bos|>", "special": True,
+             "single_word": False, "lstrip": False, "rstrip": False,
+             "normalized": False},
+            {"id": nxt + 1, "content": "<|eos|>", "special": True,
+             "single_word": False, "lstrip": False, "rstrip": False,
+             "normalized": False}],
+        "normalizer": None,
+        "pre_tokenizer": {"type": "ByteLevel", "add_prefix_space": False,
+                          "trim_offsets": True, "use_regex": True},
+        "post_processor": None,
+        "decoder": {"type": "ByteLevel", "add_prefix_space": True,
+                    "trim_offsets": True, "use_regex": True},
+        "model": {"type": "BPE", "dropout": None, "unk_token": None,
+                  "continuing_subword_prefix": None,
+                  "end_of_word_suffix": None, "fuse_unk": False,
+                  "byte_fallback": False,
+                  "vocab": vocab, "merges": merges},
+    }
+    (tmp_path / "tokenizer.json").write_text(json.dumps(tok_json))
+    (tmp_path / "tokenizer_config.json").write_text(json.dumps(
+        {"tokenizer_class": "PreTrainedTokenizerFast", "add_bos_token": False,
+         "bos_token": "<|bos|>", "eos_token": "<|eos|>",
+         "chat_template": "{{'<|im_start|>'}}"}))
+    (tmp_path / "config.json").write_text(json.dumps(
+        {"bos_token_id": nxt, "eos_token_id": nxt + 1}))
+
+    out = str(tmp_path / "out.t")
+    ctok.convert(str(tmp_path), out)
+
+    from dllama_amd.tokenizer import Tokenizer
+    t = Tokenizer(out)
+    assert t.vocab_size == nxt + 2
+    ids = t.encode("hello", is_start=False)
+    assert b"".join(t.piece(i) for i in ids) == b"hello"
+    # HF parity on the merge result
+    from transformers import PreTrainedTokenizerFast
+    hf = PreTrainedTokenizerFast(tokenizer_file=str(tmp_path / "tokenizer.json"))
+    assert ids == hf.encode("hello")
