@@ -397,3 +397,11 @@ def test_swiglu_q80_gelu(k):
     want = R.gelu(a.cpu()) * g.cpu()
     got = R.q80_dequantize(q.cpu(), s.cpu())
     assert torch.allclose(got, want, atol=want.abs().max().item() / 80)
+
+
+def test_silu_mul(k):
+    a = rand(2, 256, seed=200)
+    g = rand(2, 256, seed=201)
+    out = torch.zeros_like(a)
+    k.silu_mul(a, g, out)
+    assert torch.allclose(out.cpu(), R.swiglu(a.cpu(), g.cpu()), atol=1e-5)
