@@ -21,7 +21,7 @@ def test_greedy_parity_with_reference(tmp_path_factory):
     workdir = "/tmp/dllama_parity_cache"
     os.makedirs(workdir, exist_ok=True)
     binary = rp.build_reference(workdir)
-    for arch in ("llama", "qwen3"):
+    for arch in ("llama", "qwen3", "qwen3_moe"):
         model, tok = rp.make_ascii_assets(workdir, arch=arch)
         ref = rp.run_reference(binary, model, tok, "hello world, this is", 48)
         ours = rp.run_ours(model, tok, "hello world, this is", 48)

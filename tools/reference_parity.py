@@ -49,7 +49,13 @@ def make_ascii_assets(workdir: str, vocab_size: int = 128, arch: str = "llama"):
     tk.write_tokenizer(tok_path, vocab, scores, bos_id, True, [bos_id + 1],
                        "{{<|start_header_id|>}}")
 
-    if arch == "qwen3":
+    if arch == "qwen3_moe":
+        h = mf.LlmHeader(arch_type=mf.ARCH_QWEN3_MOE, dim=64, hidden_dim=128,
+                         n_layers=2, n_heads=4, n_kv_heads=2, head_dim=64,
+                         n_experts=4, n_active_experts=2, moe_hidden_dim=64,
+                         vocab_size=vocab_size, seq_len=256, rope_theta=10000,
+                         norm_epsilon=1e-6)
+    elif arch == "qwen3":
         h = mf.LlmHeader(arch_type=mf.ARCH_QWEN3, dim=64, hidden_dim=128,
                          n_layers=2, n_heads=4, n_kv_heads=2, head_dim=64,
                          vocab_size=vocab_size, seq_len=256, rope_theta=10000,
@@ -133,7 +139,7 @@ def main():
     os.makedirs(workdir, exist_ok=True)
     binary = build_reference(workdir)
     rc = 0
-    for arch in ("llama", "qwen3"):
+    for arch in ("llama", "qwen3", "qwen3_moe"):
         model, tok = make_ascii_assets(workdir, arch=arch)
         prompt = "hello world, this is"
         steps = 48
