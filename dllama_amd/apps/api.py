@@ -19,8 +19,6 @@ import time
 import uuid
 from http.server import BaseHTTPRequestHandler, HTTPServer
 
-import torch
-
 from ..tokenizer import (ChatItem, ChatTemplateGenerator, EosDetector,
                          TEMPLATE_UNKNOWN, chat_stops)
 from .main import build_parser, load_engine

@@ -24,7 +24,7 @@ import torch
 from .. import model_file as mf
 from ..engine import InferenceEngine
 from ..models.config import ModelConfig
-from ..parallel.comm import SingleComm, init_dist_comm
+from ..parallel.comm import init_dist_comm
 from ..quants import F32, Q80
 from ..tokenizer import (ChatItem, ChatTemplateGenerator, EosDetector, Sampler,
                          TEMPLATE_UNKNOWN, Tokenizer, _TEMPLATE_NAMES, chat_stops)

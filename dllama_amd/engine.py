@@ -14,7 +14,7 @@ broadcast, app.cpp:197-230).
 from __future__ import annotations
 
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 import torch
 

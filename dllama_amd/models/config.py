@@ -11,8 +11,7 @@ from __future__ import annotations
 
 from dataclasses import dataclass
 
-from ..model_file import (ARCH_LLAMA, ARCH_QWEN3, ARCH_QWEN3_MOE, HIDDEN_ACT_GELU,
-                          LlmHeader, ROPE_FALCON, ROPE_LLAMA, ROPE_LLAMA3_1)
+from ..model_file import (ARCH_QWEN3, ARCH_QWEN3_MOE, LlmHeader, ROPE_LLAMA3_1)
 
 
 @dataclass

@@ -19,11 +19,10 @@ from __future__ import annotations
 import numpy as np
 import torch
 
-from .. import model_file as mf
 from ..model_file import HIDDEN_ACT_GELU, ModelFile, ROPE_FALCON
 from ..ops import reference as R
 from ..parallel.comm import Comm, SingleComm
-from ..quants import F32, Q80
+from ..quants import Q80
 from .config import ModelConfig
 
 

@@ -20,7 +20,6 @@ from __future__ import annotations
 import numpy as np
 import torch
 
-from .. import model_file as mf
 from ..model_file import HIDDEN_ACT_GELU, ModelFile, ROPE_FALCON
 from ..ops import hip_ops
 from ..ops import reference as R

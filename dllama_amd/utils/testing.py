@@ -3,8 +3,6 @@
 
 from __future__ import annotations
 
-import numpy as np
-
 from .. import tokenizer as tok
 from .. import model_file as mf
 
