@@ -15,7 +15,6 @@ import sys
 
 sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 
-import numpy as np
 import torch
 
 from dllama_amd import model_file as mf

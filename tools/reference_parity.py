@@ -118,7 +118,6 @@ def run_ours(model, tok_path, prompt, steps) -> str:
     To compare trajectories we reproduce exactly that: prefill
     tokens[:-1] + [0], then sample steps-n+1 tokens.
     (dllama_amd's own engine uses the correct handoff.)"""
-    import torch
     from dllama_amd.engine import InferenceEngine
     from dllama_amd.models.config import ModelConfig
     from dllama_amd.models.cpu_model import CpuTransformer
