@@ -73,6 +73,46 @@ def test_tp2_gloo_matches_tp1(tmp_path):
             f"rank {rank}: {np.abs(val - ref).max()}"
 
 
+@pytest.mark.timeout(240)
+def test_tp4_gloo_matches_tp1(tmp_path):
+    """World=4 slicing (q/kv/ff/vocab all split 4-way) matches the
+    single-rank oracle — exercises the deeper shard math the driver's
+    round-end 8-GPU scaling run depends on (reference nn-core.cpp:211-285
+    slicers at higher world sizes)."""
+    path = str(tmp_path / "tiny4.m")
+    h = mf.LlmHeader(arch_type=mf.ARCH_LLAMA, dim=64, hidden_dim=128,
+                     n_layers=2, n_heads=4, n_kv_heads=4, head_dim=64,
+                     vocab_size=256, seq_len=128, rope_theta=10000,
+                     rope_type=mf.ROPE_LLAMA)
+    h.finalize()
+    mf.write_synthetic_model(path, h, seed=13)
+
+    from dllama_amd.models.config import ModelConfig
+    from dllama_amd.models.cpu_model import CpuTransformer
+    m = mf.ModelFile(path, sync_type=F32)
+    cfg = ModelConfig.from_header(m.header)
+    cfg.sync_type = F32
+    ref = CpuTransformer(m, cfg).forward(torch.tensor([3, 17, 101]),
+                                         torch.arange(3)).numpy()
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker, args=(r, 4, path, 29577, q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(4):
+        rank, val = q.get(timeout=200)
+        results[rank] = val
+    for p in procs:
+        p.join(timeout=60)
+    for rank, val in results.items():
+        assert not isinstance(val, str), val
+        assert np.allclose(val, ref, atol=1e-4, rtol=1e-4), \
+            f"rank {rank}: {np.abs(val - ref).max()}"
+
+
 def _worker_moe(rank, world, path, port, out_q):
     try:
         os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
