@@ -32,6 +32,30 @@ BASELINE_TOKS = {1: 1000.0 / 1312.50, 2: 1000.0 / 793.69,
                  4: 1000.0 / 494.00, 8: 1000.0 / 588.19}
 
 
+def _build_cpu_smoke(args, world, comm):
+    """Tiny CPU model wired through the SAME TP plumbing (torchrun env,
+    comm collectives, sync type) — used by tests to validate the driver's
+    N>1 launch contract on machines without a GPU."""
+    import tempfile
+
+    from dllama_amd import model_file as mf
+    from dllama_amd.models.config import ModelConfig
+    from dllama_amd.models.cpu_model import CpuTransformer
+    from dllama_amd.quants import F32, Q80
+    from dllama_amd.utils.testing import make_tiny_llama
+
+    path = os.path.join(tempfile.gettempdir(),
+                        f"dllama_bench_smoke_{os.getppid()}.m")
+    if comm.rank == 0 and not os.path.exists(path):
+        make_tiny_llama(path, vocab_size=256)
+    comm.barrier()
+    sync = Q80 if args.sync == "q80" else F32
+    m = mf.ModelFile(path, sync_type=sync)
+    cfg = ModelConfig.from_header(m.header, world=world, rank=comm.rank)
+    cfg.sync_type = sync
+    return CpuTransformer(m, cfg, comm)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -43,7 +67,13 @@ def main():
                     help="prompt tokens evaluated before the timed decode")
     ap.add_argument("--no-graph", action="store_true")
     ap.add_argument("--sync", choices=["q80", "f32"], default="q80")
+    # hidden: --device cpu runs a tiny model through the SAME distributed
+    # plumbing (torchrun env, init_dist_comm, barriers, max-over-ranks,
+    # rank-0 JSON) so tests can validate the driver contract without a GPU
+    ap.add_argument("--device", choices=["cuda", "cpu"], default="cuda",
+                    help=argparse.SUPPRESS)
     args = ap.parse_args()
+    use_cpu = args.device == "cpu"
 
     sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
     from dllama_amd import model_file as mf
@@ -60,22 +90,35 @@ def main():
     n_gpus = world if world > 1 else 1
     comm = init_dist_comm() if world > 1 else SingleComm()
     rank = comm.rank
-    device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", "0")))
-    torch.cuda.set_device(device)
+    if use_cpu:
+        device = torch.device("cpu")
+    else:
+        device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", "0")))
+        torch.cuda.set_device(device)
 
-    header = mf.preset_header(args.model, seq_len=args.seq_len)
-    header.sync_type = Q80 if args.sync == "q80" else F32
-    cfg = ModelConfig.from_header(header, world=n_gpus, rank=rank)
+    def sync_device():
+        if not use_cpu:
+            torch.cuda.synchronize(device)
 
     t0 = time.time()
-    model = HipTransformer.synthetic(cfg, device=device, comm=comm)
-    model.greedy_feedback = True
-    torch.cuda.synchronize(device)
+    if use_cpu:
+        model = _build_cpu_smoke(args, n_gpus, comm)
+    else:
+        header = mf.preset_header(args.model, seq_len=args.seq_len)
+        header.sync_type = Q80 if args.sync == "q80" else F32
+        cfg = ModelConfig.from_header(header, world=n_gpus, rank=rank)
+        model = HipTransformer.synthetic(cfg, device=device, comm=comm)
+        model.greedy_feedback = True
+    cfg = model.cfg
+    sync_device()
     if rank == 0:
-        print(f"# built synthetic {args.model} TP={n_gpus} in {time.time()-t0:.1f}s",
+        built = "tiny-llama-cpu-smoke" if use_cpu else f"synthetic {args.model}"
+        print(f"# built {built} TP={n_gpus} in {time.time()-t0:.1f}s",
               file=sys.stderr)
 
     # short prefill so the decode attends over a non-trivial context
+    # (seeded so every TP rank feeds identical tokens)
+    torch.manual_seed(1234)
     prompt = torch.randint(0, cfg.vocab_size, (args.prefill,))
     for i in range(0, args.prefill, 32):
         chunk = prompt[i: i + 32]
@@ -83,7 +126,7 @@ def main():
         model.forward(chunk, torch.arange(i, i + len(chunk)))
         model.skip_logits = False
 
-    use_graph = not args.no_graph
+    use_graph = not args.no_graph and not use_cpu
     if use_graph:
         try:
             model.capture_decode_graph()
@@ -94,26 +137,34 @@ def main():
                 print(f"# graph capture failed ({e}); running eager", file=sys.stderr)
             use_graph = False
 
-    def step():
-        if use_graph:
-            model._graph.replay()
-        else:
-            # eager decode: same kernels, per-op launches
-            model.forward_buffers(1)
-            model.k.pos_inc(model.pos, 1)
+    if use_cpu:
+        pos_h = [args.prefill]
+        tok_t = torch.tensor([7])
 
-    if not use_graph:
-        model.pos.fill_(args.prefill)
-    model.tokens[0] = 7
+        def step():
+            model.forward(tok_t, torch.tensor([pos_h[0]]))
+            pos_h[0] += 1
+    else:
+        def step():
+            if use_graph:
+                model._graph.replay()
+            else:
+                # eager decode: same kernels, per-op launches
+                model.forward_buffers(1)
+                model.k.pos_inc(model.pos, 1)
+
+        if not use_graph:
+            model.pos.fill_(args.prefill)
+        model.tokens[0] = 7
 
     for _ in range(args.warmup):
         step()
     comm.barrier()
-    torch.cuda.synchronize(device)
+    sync_device()
     t0 = time.perf_counter()
     for _ in range(args.steps):
         step()
-    torch.cuda.synchronize(device)
+    sync_device()
     elapsed = time.perf_counter() - t0
     # max over ranks
     if world > 1:
@@ -137,11 +188,11 @@ def main():
             "ms_per_step": round(ms_per_step, 4),
             "higher_is_better": True,
             "scaling": "strong",
-            "vs_baseline": round(toks / base, 1) if base else None,
+            "vs_baseline": round(toks / base, 1) if base and not use_cpu else None,
             "dtype": "f32-accum/int8-dot (Q40 weights, Q80 activations)",
             "data": "synthetic (random-init weights, random prompt; no network for checkpoints)",
             "config": {
-                "model": args.model,
+                "model": args.model if not use_cpu else "tiny-llama-cpu-smoke",
                 "global_batch": 1,
                 "seq_len": args.seq_len,
                 "prefill": args.prefill,

@@ -84,9 +84,11 @@ def init_dist_comm(backend: str | None = None, timeout_s: int = 300) -> Comm:
         return SingleComm()
     if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
+    if backend == "nccl":
+        # bind the device BEFORE init so the first collective (and barrier's
+        # device guess) never lands every rank on GPU 0
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
     if not dist.is_initialized():
         dist.init_process_group(backend=backend,
                                 timeout=datetime.timedelta(seconds=timeout_s))
-    if backend == "nccl":
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
     return DistComm()

@@ -162,3 +162,30 @@ def test_tp2_gloo_moe(tmp_path):
     for rank, val in results.items():
         assert not isinstance(val, str), val
         assert np.allclose(val, ref, atol=1e-4, rtol=1e-4)
+
+
+@pytest.mark.timeout(240)
+def test_bench_driver_contract_tp2_cpu():
+    """Launch bench.py EXACTLY the way the round-end driver does
+    (torch.distributed.run --nnodes=1 --nproc-per-node N ... bench.py
+    --gpus N) using the hidden --device cpu smoke mode: validates torchrun
+    env parsing, init_dist_comm, Q80 wire sync over collectives,
+    max-over-ranks timing and the rank-0 JSON line without a GPU."""
+    import json
+    import subprocess
+    import sys as _sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [_sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29583", "bench.py", "--gpus", "2",
+         "--device", "cpu", "--steps", "4", "--warmup", "1",
+         "--prefill", "8"],
+        cwd=repo, capture_output=True, text=True, timeout=220)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [ln for ln in out.stdout.splitlines() if ln.startswith("{")]
+    assert len(line) == 1, out.stdout  # exactly ONE JSON line (rank 0 only)
+    j = json.loads(line[0])
+    assert j["n_gpus"] == 2 and j["steps"] == 4
+    assert j["config"]["parallelism"] == "tp2"
+    assert j["value"] > 0 and j["ms_per_step"] > 0
