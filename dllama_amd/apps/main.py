@@ -15,6 +15,7 @@ from __future__ import annotations
 
 import argparse
 import os
+import signal
 import sys
 import time
 
@@ -227,6 +228,10 @@ def run_perplexity(args) -> int:
 
 
 def main(argv=None) -> int:
+    # die silently when stdout's reader goes away (e.g. `dllama ... | head`),
+    # like any Unix CLI, instead of a BrokenPipeError traceback
+    if hasattr(signal, "SIGPIPE"):
+        signal.signal(signal.SIGPIPE, signal.SIG_DFL)
     args = build_parser().parse_args(argv)
     if args.mode == "worker":
         print("ℹ️  On MI355X, workers are torchrun ranks on one node — run:\n"
