@@ -65,3 +65,27 @@ def test_tokenizer_encode_decode_roundtrip(text):
         out += bytes(tok._decode_buf).decode("utf-8", "replace")
         tok.reset_decoder()
     assert out == text
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.text(min_size=0, max_size=300),
+       st.booleans())
+def test_native_bpe_matches_python(text, with_special):
+    """The C++ BpeEncoder (extension) and the pure-Python fallback must
+    tokenize identically — same greedy-merge order, same leftmost-max
+    tie-break, same special-token scan (reference tokenizer.cpp:311-390)."""
+    tok = _get_tok()
+    if tok._native is None:
+        import pytest
+        pytest.skip("extension not built")
+    if with_special:
+        text = "<|start_header_id|>" + text + "<|eot_id|>"
+    data = text.encode("utf-8")
+    native = tok._native.encode(data, True)
+    saved = tok._native
+    try:
+        tok._native = None
+        python = tok.encode(data, is_start=False, add_special_tokens=True)
+    finally:
+        tok._native = saved
+    assert native == python
