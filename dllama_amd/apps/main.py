@@ -42,6 +42,9 @@ def build_parser() -> argparse.ArgumentParser:
                    help="TP sync buffer quantization (reference --buffer-float-type)")
     p.add_argument("--nthreads", type=int, default=os.cpu_count(),
                    help="CPU backend threads (reference --nthreads)")
+    p.add_argument("--cpu-dtype", default="f32", choices=["f32", "f16"],
+                   help="CPU backend weight dtype: f16 halves weight memory "
+                        "traffic; f32 is the exact-oracle default")
     p.add_argument("--temperature", type=float, default=0.8)
     p.add_argument("--topp", type=float, default=0.9)
     p.add_argument("--seed", type=int, default=None)
@@ -85,7 +88,8 @@ def load_engine(args):
     else:
         from ..models.cpu_model import CpuTransformer
         torch.set_num_threads(max(1, args.nthreads))
-        model = CpuTransformer(m, cfg, comm)
+        wdt = torch.float16 if args.cpu_dtype == "f16" else torch.float32
+        model = CpuTransformer(m, cfg, comm, weight_dtype=wdt)
     tok = Tokenizer(args.tokenizer) if args.tokenizer else None
     seed = args.seed if args.seed is not None else int(time.time())
     if comm.world > 1:

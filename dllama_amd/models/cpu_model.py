@@ -28,17 +28,21 @@ from .config import ModelConfig
 
 class CpuTransformer:
     def __init__(self, m: ModelFile, config: ModelConfig, comm: Comm | None = None,
-                 activation_quant: bool = True):
+                 activation_quant: bool = True, weight_dtype: torch.dtype = torch.float32):
         self.cfg = config
         self.comm = comm or SingleComm()
         self.activation_quant = activation_quant  # False = pure f32 (debugging)
         self.skip_logits = False  # accepted for engine compat; CPU always computes
+        # f16 weights halve the memory traffic of the bandwidth-bound CPU
+        # decode. f32 stays the default: it is the numerics oracle and the
+        # reference-parity path.
+        self.weight_dtype = weight_dtype
         c = self.cfg
         r, w = c.rank, c.world
 
         def t(name, layer=-1, expert=-1):
             return torch.from_numpy(
-                np.array(m.slice_f32(name, layer, r, w, expert)))
+                np.array(m.slice_f32(name, layer, r, w, expert))).to(weight_dtype)
 
         self.embedding = torch.from_numpy(np.array(m.f32("embedding")))
         self.final_norm = torch.from_numpy(np.array(m.f32("final_norm")))
@@ -77,6 +81,12 @@ class CpuTransformer:
         return self.comm.allreduce_(partial)
 
     def _matmul(self, x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+        if self.weight_dtype != torch.float32:
+            # fast serving path: Q80-roundtripped activations (same wire
+            # semantics), f16 weight stream, f32 output
+            if self.activation_quant:
+                x = R.q80_roundtrip(x)
+            return (x.to(self.weight_dtype) @ w.t()).float()
         return R.q40_matmul(x, w, quantize_x=self.activation_quant)
 
     def _rope(self, x: torch.Tensor, positions: torch.Tensor) -> torch.Tensor:
@@ -142,14 +152,16 @@ class CpuTransformer:
         router = t1 @ lw["gate"].t()
         idx, wts = R.moe_gate(router, c.n_active_experts)  # [B,k]
         tq = R.q80_roundtrip(t1) if self.activation_quant else t1
+        tq = tq.to(self.weight_dtype)
         partial = torch.zeros(B, c.dim)
         for b in range(B):
             for s in range(c.n_active_experts):
                 e = int(idx[b, s])
-                a = tq[b] @ lw["w1"][e].t()
-                g = tq[b] @ lw["w3"][e].t()
+                a = (tq[b] @ lw["w1"][e].t()).float()
+                g = (tq[b] @ lw["w3"][e].t()).float()
                 d = R.swiglu(a, g)
                 if self.activation_quant:
                     d = R.q80_roundtrip(d)
-                partial[b] += wts[b, s] * (d @ lw["w2"][e].t())
+                partial[b] += wts[b, s] * (d.to(self.weight_dtype)
+                                           @ lw["w2"][e].t()).float()
         return partial

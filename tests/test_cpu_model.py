@@ -264,3 +264,27 @@ def test_engine_stops_at_seq_len(tmp_path):
     eng = InferenceEngine(CpuTransformer(m, cfg))
     out, _ = eng.generate([1, 2, 3, 4], 64)  # asks for more than fits
     assert len(out) <= 16 - 4 + 1
+
+
+def test_cpu_f16_weights_close_to_f32(tmp_path):
+    """--cpu-dtype f16 fast path: same Q80 activation semantics, f16 weight
+    stream; logits must track the f32 oracle closely (weights originate from
+    4-bit Q40, so f16 rounding is far below the quantization noise)."""
+    import torch
+    from dllama_amd import model_file as mf
+    from dllama_amd.models.config import ModelConfig
+    from dllama_amd.models.cpu_model import CpuTransformer
+    from dllama_amd.utils.testing import make_tiny_llama
+
+    path = str(tmp_path / "tiny.m")
+    make_tiny_llama(path, vocab_size=256)
+    m = mf.ModelFile(path)
+    cfg = ModelConfig.from_header(m.header)
+    toks = torch.tensor([3, 17, 101, 9])
+    pos = torch.arange(4)
+    ref = CpuTransformer(m, cfg).forward(toks, pos)
+    f16 = CpuTransformer(m, cfg, weight_dtype=torch.float16).forward(toks, pos)
+    assert f16.dtype == torch.float32
+    # tight relative agreement and same greedy tokens on a decisive model
+    err = (f16 - ref).abs().max() / ref.abs().max()
+    assert float(err) < 0.05, float(err)
