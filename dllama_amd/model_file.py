@@ -450,6 +450,18 @@ def preset_header(name: str, seq_len: int | None = None) -> LlmHeader:
                            n_layers=28, n_heads=16, n_kv_heads=8, head_dim=128,
                            vocab_size=151936, seq_len=40960, rope_theta=1000000,
                            norm_epsilon=1e-6),
+        "qwen3-1.7b": dict(arch_type=ARCH_QWEN3, dim=2048, hidden_dim=6144,
+                           n_layers=28, n_heads=16, n_kv_heads=8, head_dim=128,
+                           vocab_size=151936, seq_len=40960, rope_theta=1000000,
+                           norm_epsilon=1e-6),
+        "qwen3-8b": dict(arch_type=ARCH_QWEN3, dim=4096, hidden_dim=12288,
+                         n_layers=36, n_heads=32, n_kv_heads=8, head_dim=128,
+                         vocab_size=151936, seq_len=40960, rope_theta=1000000,
+                         norm_epsilon=1e-6),
+        "qwen3-14b": dict(arch_type=ARCH_QWEN3, dim=5120, hidden_dim=17408,
+                          n_layers=40, n_heads=40, n_kv_heads=8, head_dim=128,
+                          vocab_size=151936, seq_len=40960, rope_theta=1000000,
+                          norm_epsilon=1e-6),
     }
     if name not in presets:
         raise KeyError(f"unknown preset {name}; have {sorted(presets)}")
