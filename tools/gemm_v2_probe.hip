@@ -121,8 +121,13 @@ k_gemm_v1(const uint8_t *__restrict__ qs, const __half *__restrict__ scales,
 // 4 waves/wg, one 32-row m-tile per wave; activations + x-scales staged in
 // LDS per CHUNK blocks; weight uint4 tiles prefetched RING deep.
 #define CHUNK 8
+typedef __attribute__((address_space(3))) void *lds_ptr_t;
 
-template <int RING>
+// DMA=true stages fragments with global_load_lds_dwordx4 (direct
+// global->LDS, no VGPR round-trip): one instruction moves 1 KB per wave
+// (lane i's 16 B land at lds_base + i*16, so the source address is arranged
+// so lane i reads batch row i/2, half i%2).
+template <int RING, bool DMA>
 __global__ void __launch_bounds__(256)
 k_gemm_v2(const uint8_t *__restrict__ qs, const __half *__restrict__ scales,
           const int8_t *__restrict__ xq, const float *__restrict__ xs,
@@ -147,15 +152,26 @@ k_gemm_v2(const uint8_t *__restrict__ qs, const __half *__restrict__ scales,
     // 8 KB of fragments = 32 B/thread (2x uint4), 256 scale floats
     auto stage = [&](int buf, int jc) {
         const int nblk = min(CHUNK, j1 - jc);
-        // 256 threads copy 32 batch rows x nblk blocks, one (block, batch-row)
-        // 32 B fragment per thread per step
-        for (int u = tid; u < nblk * 32; u += 256) {
-            const int b = u & 31;
-            const int blk = u >> 5;
-            *reinterpret_cast<uint4 *>(&lds_a[buf][blk][b][0]) =
-                *reinterpret_cast<const uint4 *>(xq + (int64_t)b * n + (jc + blk) * QB);
-            *reinterpret_cast<uint4 *>(&lds_a[buf][blk][b][16]) =
-                *reinterpret_cast<const uint4 *>(xq + (int64_t)b * n + (jc + blk) * QB + 16);
+        if constexpr (DMA) {
+            // wave w stages blocks w, w+4, ...: one dwordx4 LDS-DMA moves
+            // the whole 1 KB fragment (lane i -> batch row i/2, half i%2)
+            for (int blk = wave; blk < nblk; blk += 4) {
+                const int8_t *src = xq + (int64_t)(lane >> 1) * n +
+                                    (jc + blk) * QB + (lane & 1) * 16;
+                __builtin_amdgcn_load_to_lds((void *)src,
+                                             (lds_ptr_t)&lds_a[buf][blk][0][0],
+                                             16, 0, 0);
+            }
+        } else {
+            // 256 threads copy 32 batch rows x nblk blocks, 32 B each
+            for (int u = tid; u < nblk * 32; u += 256) {
+                const int b = u & 31;
+                const int blk = u >> 5;
+                *reinterpret_cast<uint4 *>(&lds_a[buf][blk][b][0]) =
+                    *reinterpret_cast<const uint4 *>(xq + (int64_t)b * n + (jc + blk) * QB);
+                *reinterpret_cast<uint4 *>(&lds_a[buf][blk][b][16]) =
+                    *reinterpret_cast<const uint4 *>(xq + (int64_t)b * n + (jc + blk) * QB + 16);
+            }
         }
         for (int u = tid; u < nblk * 32; u += 256) {
             const int b = u & 31;
@@ -177,6 +193,7 @@ k_gemm_v2(const uint8_t *__restrict__ qs, const __half *__restrict__ scales,
         const int j = j0 + p;
         if (live && j < j1) { ring[p] = wrow[j]; ring_s[p] = srow[j]; }
     }
+    if constexpr (DMA) __builtin_amdgcn_s_waitcnt(0);  // LDS-DMA uses vmcnt
     __syncthreads();
 
     int buf = 0;
@@ -216,6 +233,7 @@ k_gemm_v2(const uint8_t *__restrict__ qs, const __half *__restrict__ scales,
                 }
             }
         }
+        if constexpr (DMA) __builtin_amdgcn_s_waitcnt(0);  // LDS-DMA uses vmcnt
         __syncthreads();
     }
 
@@ -333,12 +351,16 @@ int main(int argc, char **argv) {
                            dqs, dsw, dxq, dxs, dy, d, n, batch);
     });
     bench("v2 ring4", [&] {
-        hipLaunchKernelGGL((k_gemm_v2<4>), dim3(mtiles, ksplit), dim3(256), 0, 0,
-                           dqs, dsw, dxq, dxs, dy, d, n, batch);
+        hipLaunchKernelGGL((k_gemm_v2<4, false>), dim3(mtiles, ksplit), dim3(256),
+                           0, 0, dqs, dsw, dxq, dxs, dy, d, n, batch);
     });
     bench("v2 ring8", [&] {
-        hipLaunchKernelGGL((k_gemm_v2<8>), dim3(mtiles, ksplit), dim3(256), 0, 0,
-                           dqs, dsw, dxq, dxs, dy, d, n, batch);
+        hipLaunchKernelGGL((k_gemm_v2<8, false>), dim3(mtiles, ksplit), dim3(256),
+                           0, 0, dqs, dsw, dxq, dxs, dy, d, n, batch);
+    });
+    bench("v2 ring4+ldsdma", [&] {
+        hipLaunchKernelGGL((k_gemm_v2<4, true>), dim3(mtiles, ksplit), dim3(256),
+                           0, 0, dqs, dsw, dxq, dxs, dy, d, n, batch);
     });
     return 0;
 }
