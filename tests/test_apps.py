@@ -307,3 +307,51 @@ def test_convert_tokenizer_hf(tmp_path):
     from transformers import PreTrainedTokenizerFast
     hf = PreTrainedTokenizerFast(tokenizer_file=str(tmp_path / "tokenizer.json"))
     assert ids == hf.encode("hello")
+
+
+def test_api_error_isolation(assets):
+    """Bad requests (invalid JSON, unknown route, empty messages) must not
+    kill the server: each gets an error status and the NEXT request still
+    succeeds (reference dllama-api stays up across malformed requests)."""
+    from dllama_amd.apps import api as api_mod
+    from dllama_amd.apps.main import build_parser
+    mp, tp = assets
+    args = build_parser().parse_args(
+        ["inference", "--model", mp, "--tokenizer", tp, "--temperature", "0",
+         "--gpu-index", "-1", "--port", "18933"])
+    api_mod.STATE = api_mod.ApiState(args)
+    from http.server import HTTPServer
+    server = HTTPServer(("127.0.0.1", 18933), api_mod.Handler)
+    t = threading.Thread(target=server.serve_forever, daemon=True)
+    t.start()
+    try:
+        def req(method, path, body=None):
+            conn = http.client.HTTPConnection("127.0.0.1", 18933, timeout=60)
+            conn.request(method, path, body,
+                         {"Content-Type": "application/json"} if body else {})
+            r = conn.getresponse()
+            out = (r.status, r.read())
+            conn.close()
+            return out
+
+        status, _ = req("POST", "/v1/chat/completions", "{not json")
+        assert status == 400
+        status, _ = req("POST", "/v1/nope", "{}")
+        assert status == 404
+        status, _ = req("GET", "/nope")
+        assert status == 404
+        # empty messages -> server error, not a crash
+        status, _ = req("POST", "/v1/chat/completions",
+                        json.dumps({"messages": [], "max_tokens": 4}))
+        assert status in (200, 500)
+        # server still alive and serving
+        status, body = req("GET", "/health")
+        assert status == 200 and json.loads(body)["status"] == "ok"
+        status, _ = req("POST", "/v1/chat/completions",
+                        json.dumps({"messages": [{"role": "user",
+                                                  "content": "ok?"}],
+                                    "max_tokens": 2}))
+        assert status == 200
+    finally:
+        server.shutdown()
+        server.server_close()
