@@ -221,6 +221,7 @@ class HipTransformer:
         self.amax_scratch = torch.zeros(self.amax_blocks, dtype=torch.int64, device=dev)
         import os as _os
         self.attn_splits = int(_os.environ.get("DLLAMA_ATTN_SPLITS", "8"))
+        self.fused_sync = _os.environ.get("DLLAMA_FUSED_SYNC", "0") == "1"
         # S=8 measured best at decode (16/32: combine reads S partials)
         self.attn_ml = torch.zeros(NB * c.n_heads0 * self.attn_splits * 2, device=dev)
         self.attn_o = torch.zeros(NB * c.n_heads0 * self.attn_splits * c.head_dim,
@@ -265,11 +266,16 @@ class HipTransformer:
         c = self.cfg
         if c.sync_type == Q80:
             nb_dim = c.dim // QB
-            q = self.xq  # reuse dim-sized quant buffer
-            self.k.q80_quantize(self.partial[:NB], q.q[:NB], q.s[:NB], q.bs[:NB])
             row_bytes = c.dim + 2 * nb_dim
             out = self.sync_out[: NB * row_bytes]
-            self.k.sync_pack(q.q[:NB], q.s[:NB], out)
+            if getattr(self, "fused_sync", False):
+                # round-2 experimental: quantize straight into the wire
+                # buffer, one pass (DLLAMA_FUSED_SYNC=1)
+                self.k.sync_quant_pack(self.partial[:NB], out)
+            else:
+                q = self.xq  # reuse dim-sized quant buffer
+                self.k.q80_quantize(self.partial[:NB], q.q[:NB], q.s[:NB], q.bs[:NB])
+                self.k.sync_pack(q.q[:NB], q.s[:NB], out)
             inb = self.sync_in[NB]
             self.comm.all_gather(inb, out)
             self.k.merge_add(self.x[:NB], inb.view(c.world, NB, row_bytes),

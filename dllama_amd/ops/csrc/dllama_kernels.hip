@@ -1294,6 +1294,31 @@ __global__ void k_silu_mul(const float *__restrict__ a,
 // wire layout per row: n int8 payload then n/32 f16 scales
 // (role of cast-forward-f32-q80.comp; consumed after RCCL all-gather by
 //  k_merge_add, the reference merge-add-forward-q80-f32.comp equivalent).
+// fused quantize-into-wire: one pass replaces k_q80_quantize + k_sync_pack
+// for the TP sync (the wire consumer k_merge_add needs no blocksum).
+// EXPERIMENTAL round-2 path: dispatched only under DLLAMA_FUSED_SYNC=1.
+__global__ void k_sync_quant_pack(const float *__restrict__ x,
+                                  uint8_t *__restrict__ buf,
+                                  int n, int n_blocks_total) {
+    const int gid = (blockIdx.x * blockDim.x + threadIdx.x) / 32;
+    const int lane = threadIdx.x & 31;
+    if (gid >= n_blocks_total) return;
+    const int nb = n / QB;
+    const int r = gid / nb, jb = gid % nb;
+    const int row_bytes = n + 2 * nb;
+    const float v = x[(int64_t)gid * QB + lane];
+    const float amax = group32_reduce_max(fabsf(v));
+    const float d = amax / 127.0f;
+    const float inv = d > 0.0f ? 1.0f / d : 0.0f;
+    buf[(int64_t)r * row_bytes + jb * QB + lane] = (uint8_t)(int8_t)rintf(v * inv);
+    if (lane == 0) {
+        const __half h = __float2half(d);
+        const uint16_t u = *reinterpret_cast<const uint16_t *>(&h);
+        buf[(int64_t)r * row_bytes + n + 2 * jb] = (uint8_t)(u & 0xFF);
+        buf[(int64_t)r * row_bytes + n + 2 * jb + 1] = (uint8_t)(u >> 8);
+    }
+}
+
 __global__ void k_sync_pack(const int8_t *__restrict__ q,
                             const float *__restrict__ s,
                             uint8_t *__restrict__ buf,
@@ -1864,6 +1889,16 @@ void sync_pack(torch::Tensor q, torch::Tensor s, torch::Tensor buf) {
                        buf.data_ptr<uint8_t>(), n, rows);
 }
 
+void sync_quant_pack(torch::Tensor x, torch::Tensor buf) {
+    CHECK_CUDA(x);
+    const int n = x.size(-1);
+    const int rows = x.numel() / n;
+    const int blocks = rows * (n / QB);
+    hipLaunchKernelGGL(k_sync_quant_pack, dim3(ceil_div(blocks * 32, 256)),
+                       dim3(256), 0, cur_stream(), x.data_ptr<float>(),
+                       buf.data_ptr<uint8_t>(), n, blocks);
+}
+
 void merge_add(torch::Tensor x, torch::Tensor bufs,
                c10::optional<torch::Tensor> ssq = c10::nullopt) {
     CHECK_CUDA(x);
@@ -2041,6 +2076,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("s"), py::arg("bs"), py::arg("gelu") = false);
     m.def("silu_mul", &silu_mul);
     m.def("sync_pack", &sync_pack);
+    m.def("sync_quant_pack", &sync_quant_pack);
     m.def("merge_add", &merge_add, py::arg("x"), py::arg("bufs"), py::arg("ssq") = py::none());
     m.def("add_", &add_);
     m.def("pos_inc", &pos_inc);

@@ -405,3 +405,20 @@ def test_silu_mul(k):
     out = torch.zeros_like(a)
     k.silu_mul(a, g, out)
     assert torch.allclose(out.cpu(), R.swiglu(a.cpu(), g.cpu()), atol=1e-5)
+
+
+def test_sync_quant_pack_matches_two_kernel_path(k):
+    """Fused quantize-into-wire (round-2 DLLAMA_FUSED_SYNC path) must be
+    bit-identical to k_q80_quantize + k_sync_pack: same rounding, same wire
+    layout (int8 row then f16 scales)."""
+    B, n = 3, 512
+    x = rand(B, n, seed=97)
+    q = torch.zeros(B, n, dtype=torch.int8, device=DEV)
+    s = torch.zeros(B, n // 32, device=DEV)
+    bs = torch.zeros(B, n // 32, device=DEV)
+    k.q80_quantize(x, q, s, bs)
+    want = torch.zeros(B * (n + 2 * (n // 32)), dtype=torch.uint8, device=DEV)
+    k.sync_pack(q, s, want)
+    got = torch.zeros_like(want)
+    k.sync_quant_pack(x, got)
+    assert torch.equal(got, want)
