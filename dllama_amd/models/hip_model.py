@@ -222,6 +222,10 @@ class HipTransformer:
         import os as _os
         self.attn_splits = int(_os.environ.get("DLLAMA_ATTN_SPLITS", "8"))
         self.fused_sync = _os.environ.get("DLLAMA_FUSED_SYNC", "0") == "1"
+        # round-2 experimental: past this decode position, recapture the
+        # graph with S=16 splits (S=16 measured better at ~1.1k ctx);
+        # 0 = disabled
+        self.adaptive_thresh = int(_os.environ.get("DLLAMA_ADAPTIVE_SPLITS", "0"))
         # S=8 measured best at decode (16/32: combine reads S partials)
         self.attn_ml = torch.zeros(NB * c.n_heads0 * self.attn_splits * 2, device=dev)
         self.attn_o = torch.zeros(NB * c.n_heads0 * self.attn_splits * c.head_dim,
@@ -444,6 +448,9 @@ class HipTransformer:
                 "(rebuild with a larger --max-seq-len / seq_len)")
         self.tokens[:B].copy_(tokens.to(self.device), non_blocking=True)
         if B == 1 and self._graph is not None:
+            if (self.adaptive_thresh and p0 >= self.adaptive_thresh
+                    and self.attn_splits < 16):
+                self._set_attn_splits(16)  # recapture with more K-splits
             if p0 != self._graph_pos:
                 self.pos.fill_(p0)
             self._graph.replay()
@@ -472,6 +479,25 @@ class HipTransformer:
         return self.logits0[:B]
 
     # ------------------------------------------------------------ graphs
+
+    def _set_attn_splits(self, s: int):
+        """Switch the flash-decode K-split count mid-stream (long-context
+        adaptivity): resize the split scratch and recapture the decode
+        graph. Decode state (KV caches, device pos) is untouched; the
+        capture warmup's KV write at the current pos is overwritten by the
+        next real step."""
+        c = self.cfg
+        self.attn_splits = s
+        NB = self.n_batches
+        dev = self.device
+        self.attn_ml = torch.zeros(NB * c.n_heads0 * s * 2, device=dev)
+        self.attn_o = torch.zeros(NB * c.n_heads0 * s * c.head_dim, device=dev)
+        if self._graph is not None:
+            pos_saved = int(self.pos.item())
+            self.capture_decode_graph()
+            self.pos.fill_(pos_saved)
+            self._graph_pos = pos_saved
+        self._pf_graphs.clear()  # prefill graphs also reference the scratch
 
     def _capture_prefill_graph(self, skip_logits: bool):
         try:

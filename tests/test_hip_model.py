@@ -112,3 +112,27 @@ def test_hip_prefill_gemm_path(tiny_path):
     cpu = CpuTransformer(m, cfg)
     want = cpu.forward(torch.tensor(tokens), torch.arange(len(tokens)))
     assert _rel_err(batch, want) < 0.02
+
+
+@pytest.mark.skipif(not __import__("os").environ.get("DLLAMA_EXPERIMENTAL"),
+                    reason="round-2 experimental path (DLLAMA_EXPERIMENTAL=1)")
+def test_adaptive_splits_recapture(tiny_path):
+    """DLLAMA_ADAPTIVE_SPLITS: crossing the position threshold re-captures
+    the decode graph with S=16; logits must keep matching eager decode."""
+    from dllama_amd.models.hip_model import HipTransformer
+    m = mf.ModelFile(tiny_path)
+    cfg = ModelConfig.from_header(m.header)
+    eager = HipTransformer.from_file(m, cfg)
+    adaptive = HipTransformer.from_file(m, cfg)
+    adaptive.adaptive_thresh = 6
+    prompt = [1, 2, 3]
+    eager.forward(torch.tensor(prompt), torch.arange(3))
+    adaptive.forward(torch.tensor(prompt), torch.arange(3))
+    adaptive.capture_decode_graph()
+    for step in range(8):  # crosses the threshold at pos 6
+        t = torch.tensor([20 + step])
+        p = torch.tensor([3 + step])
+        want = eager.forward(t, p).cpu().clone()
+        got = adaptive.forward(t, p).cpu().clone()
+        assert _rel_err(got[0], want[0]) < 1e-4, f"step {step}"
+    assert adaptive.attn_splits == 16
