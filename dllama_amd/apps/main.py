@@ -65,6 +65,9 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--host", default="0.0.0.0")
     p.add_argument("--port", type=int, default=9990)
     p.add_argument("--net-turbo", type=int, default=1, help="ignored (no TCP mesh)")
+    p.add_argument("--gpu-segments", default=None,
+                   help="ignored: reference hybrid CPU/GPU device placement "
+                        "(app.cpp:119-124); every rank here is one whole GPU")
     return p
 
 
