@@ -115,6 +115,12 @@ def _print_header(h, cfg, use_gpu):
     if h.n_experts:
         print(f"💡 nExperts: {h.n_experts}\n💡 nActiveExperts: {h.n_active_experts}")
     print(f"💡 Backend: {'MI355X HIP' if use_gpu else 'CPU'}  TP={cfg.world}")
+    # memory accounting (reference 📀 prints, nn-core.cpp:175-189): per-rank
+    # resident weight shard + dense f32 KV cache
+    kv_bytes = 2 * cfg.n_layers * cfg.seq_len * cfg.kv_dim0 * 4
+    weights = h.file_size - h.header_size  # .m = header + packed weights
+    print(f"📀 Weights/rank: {weights / max(1, cfg.world) / 1e9:.2f} GB"
+          f"  KV cache/rank: {kv_bytes / 1e9:.2f} GB (seq {cfg.seq_len})")
 
 
 def run_inference(args) -> int:
