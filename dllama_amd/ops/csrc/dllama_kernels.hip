@@ -1003,6 +1003,127 @@ k_q40_gemm(const uint8_t *__restrict__ qs,
     }
 }
 
+// EXPERIMENTAL round-2 GEMM (DLLAMA_GEMM_V2=1; see tools/gemm_v2_probe.hip
+// for the standalone A/B harness). Differences vs k_q40_gemm:
+//   - activation fragments + x-scales staged in LDS per 8-block chunk,
+//     loaded once per workgroup and shared by all 4 waves,
+//   - weight uint4 tiles prefetched through a 4-deep register ring,
+//   - descale as float2 pairs (adjacent C rows) with broadcast ds_reads.
+// 71 VGPR + 16 AGPR (v1: 126+30) -> 5 waves/SIMD, no spills.
+#define GEMM_V2_CHUNK 8
+
+__global__ void __launch_bounds__(256)
+k_q40_gemm_v2(const uint8_t *__restrict__ qs,
+              const __half *__restrict__ scales,
+              const int8_t *__restrict__ xq,
+              const float *__restrict__ xs,
+              float *__restrict__ y,
+              float *__restrict__ part,
+              int d, int n, int batch) {
+    constexpr int RING = 4;
+    __shared__ int8_t lds_a[2][GEMM_V2_CHUNK][32][QB];
+    __shared__ float lds_s[2][GEMM_V2_CHUNK][32];
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int tid = threadIdx.x;
+    const int mbase = (blockIdx.x * 4 + wave) * 32;
+    const int nb = n / QB;
+    const int ksplit = gridDim.y;
+    const int j0 = (int)((int64_t)nb * blockIdx.y / ksplit);
+    const int j1 = (int)((int64_t)nb * (blockIdx.y + 1) / ksplit);
+    const int khi = lane >> 5;
+    const int mcol = lane & 31;
+    const int mrow = min(mbase + mcol, d - 1);
+    const uint4 *wrow = reinterpret_cast<const uint4 *>(qs + (int64_t)mrow * (n >> 1));
+    const __half *srow = scales + (int64_t)mrow * nb;
+    const bool live = mbase < d;
+
+    auto stage = [&](int buf, int jc) {
+        const int nblk = min(GEMM_V2_CHUNK, j1 - jc);
+        for (int u = tid; u < nblk * 32; u += 256) {
+            const int b = u & 31;
+            const int blk = u >> 5;
+            *reinterpret_cast<uint4 *>(&lds_a[buf][blk][b][0]) =
+                *reinterpret_cast<const uint4 *>(xq + (int64_t)b * n + (jc + blk) * QB);
+            *reinterpret_cast<uint4 *>(&lds_a[buf][blk][b][16]) =
+                *reinterpret_cast<const uint4 *>(xq + (int64_t)b * n + (jc + blk) * QB + 16);
+            lds_s[buf][blk][b] = xs[(int64_t)b * nb + (jc + blk)];
+        }
+    };
+    auto extract = [&](const uint4 &wq, v4i32_t &b) {
+        const uint32_t wv[4] = {wq.x, wq.y, wq.z, wq.w};
+        #pragma unroll
+        for (int t = 0; t < 4; t++) {
+            uint32_t s = khi ? ((wv[t] >> 4) & 0x0F0F0F0Fu) : (wv[t] & 0x0F0F0F0Fu);
+            s ^= 0x08080808u;
+            b[t] = (int)(s | (((s >> 3) & 0x01010101u) * 0xF0u));
+        }
+    };
+
+    float2 facc[8];
+    #pragma unroll
+    for (int r = 0; r < 8; r++) facc[r] = make_float2(0.0f, 0.0f);
+
+    stage(0, j0);
+    uint4 ring[RING];
+    __half ring_s[RING];
+    #pragma unroll
+    for (int p = 0; p < RING; p++) {
+        const int j = j0 + p;
+        if (live && j < j1) { ring[p] = wrow[j]; ring_s[p] = srow[j]; }
+    }
+    __syncthreads();
+
+    int buf = 0;
+    for (int jc = j0; jc < j1; jc += GEMM_V2_CHUNK, buf ^= 1) {
+        const int nblk = min(GEMM_V2_CHUNK, j1 - jc);
+        if (jc + GEMM_V2_CHUNK < j1) stage(buf ^ 1, jc + GEMM_V2_CHUNK);
+        if (live) {
+            #pragma unroll
+            for (int jj = 0; jj < GEMM_V2_CHUNK; jj++) {
+                if (jj >= nblk) break;
+                const int j = jc + jj;
+                const int slot = (j - j0) % RING;  // CHUNK % RING == 0: static
+                const uint4 wq = ring[slot];
+                const float sw = __half2float(ring_s[slot]);
+                const int jn = j + RING;
+                if (jn < j1) { ring[slot] = wrow[jn]; ring_s[slot] = srow[jn]; }
+                v4i32_t a = *reinterpret_cast<const v4i32_t *>(
+                    &lds_a[buf][jj][lane & 31][khi * 16]);
+                v4i32_t b;
+                extract(wq, b);
+                v16i32_t iacc = {};
+                iacc = __builtin_amdgcn_mfma_i32_32x32x32_i8(a, b, iacc, 0, 0, 0);
+                #pragma unroll
+                for (int r2 = 0; r2 < 8; r2++) {
+                    const int brow = ((2 * r2) & 3) + 8 * (r2 >> 1) + 4 * khi;
+                    const float2 sx2 = *reinterpret_cast<const float2 *>(
+                        &lds_s[buf][jj][brow]);
+                    facc[r2].x = fmaf((float)iacc[2 * r2], sw * sx2.x, facc[r2].x);
+                    facc[r2].y = fmaf((float)iacc[2 * r2 + 1], sw * sx2.y, facc[r2].y);
+                }
+            }
+        }
+        __syncthreads();
+    }
+
+    if (live && mbase + mcol < d) {
+        #pragma unroll
+        for (int r2 = 0; r2 < 8; r2++) {
+            const int brow = ((2 * r2) & 3) + 8 * (r2 >> 1) + 4 * khi;
+            if (ksplit == 1) {
+                if (brow < batch)
+                    y[(int64_t)brow * d + mbase + mcol] = facc[r2].x;
+                if (brow + 1 < batch)
+                    y[(int64_t)(brow + 1) * d + mbase + mcol] = facc[r2].y;
+            } else {
+                part[(((int64_t)blockIdx.y * 32) + brow) * d + mbase + mcol] = facc[r2].x;
+                part[(((int64_t)blockIdx.y * 32) + brow + 1) * d + mbase + mcol] = facc[r2].y;
+            }
+        }
+    }
+}
+
 __global__ void k_gemm_reduce(const float *__restrict__ part,
                               float *__restrict__ y,
                               int d, int batch, int ksplit) {
@@ -1593,7 +1714,8 @@ void q40_gemv_nq_rope(torch::Tensor qs, torch::Tensor scales, torch::Tensor x,
 
 void q40_gemm(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
               torch::Tensor xs, torch::Tensor y, int64_t batch,
-              c10::optional<torch::Tensor> part = c10::nullopt) {
+              c10::optional<torch::Tensor> part = c10::nullopt,
+              bool use_v2 = false) {
     // int8-MFMA batched matmul (prefill path); xq/xs must have >=32 rows
     CHECK_CUDA(qs); CHECK_CONT(qs); CHECK_CONT(xq);
     const int d = qs.size(0);
@@ -1609,7 +1731,10 @@ void q40_gemm(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
         while (ksplit > 1 && (int64_t)ksplit * 32 * d > part->numel()) ksplit--;
     }
     float *pp = ksplit > 1 ? part->data_ptr<float>() : nullptr;
-    hipLaunchKernelGGL(k_q40_gemm, dim3(mtiles, ksplit), dim3(256), 0,
+    static const bool env_v2 =
+        std::getenv("DLLAMA_GEMM_V2") && atoi(std::getenv("DLLAMA_GEMM_V2")) == 1;
+    auto *kern = (use_v2 || env_v2) ? k_q40_gemm_v2 : k_q40_gemm;
+    hipLaunchKernelGGL(kern, dim3(mtiles, ksplit), dim3(256), 0,
                        cur_stream(), qs.data_ptr<uint8_t>(),
                        reinterpret_cast<const __half *>(scales.data_ptr<at::Half>()),
                        xq.data_ptr<int8_t>(), xs.data_ptr<float>(),
@@ -2046,7 +2171,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("q40_gemv_resid", &q40_gemv_resid);
     m.def("q40_gemm", &q40_gemm, py::arg("qs"), py::arg("scales"),
           py::arg("xq"), py::arg("xs"), py::arg("y"), py::arg("batch"),
-          py::arg("part") = py::none());
+          py::arg("part") = py::none(), py::arg("use_v2") = false);
     m.def("q40_gemv_rope", &q40_gemv_rope);
     m.def("q40_gemv_nq", &q40_gemv_nq, py::arg("qs"), py::arg("scales"),
           py::arg("x"), py::arg("wnorm"), py::arg("ssq"), py::arg("eps"),

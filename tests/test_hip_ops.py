@@ -422,3 +422,27 @@ def test_sync_quant_pack_matches_two_kernel_path(k):
     got = torch.zeros_like(want)
     k.sync_quant_pack(x, got)
     assert torch.equal(got, want)
+
+
+@pytest.mark.skipif(not __import__("os").environ.get("DLLAMA_EXPERIMENTAL"),
+                    reason="round-2 experimental kernel (DLLAMA_EXPERIMENTAL=1)")
+def test_q40_gemm_v2_matches_v1(k):
+    """LDS-staged GEMM v2 (use_v2=True / DLLAMA_GEMM_V2) vs the production
+    kernel, with and without K-split partials, including row masking."""
+    for d, n, B in ((160, 1024, 20), (512, 2048, 32), (256, 4096, 9)):
+        qs, sc, _ = _mk_linear(d, n, 60 + d % 7)
+        x = rand(32, n, seed=61, scale=0.5)
+        q = torch.zeros(32, n, dtype=torch.int8, device=DEV)
+        s = torch.zeros(32, n // 32, device=DEV)
+        bs = torch.zeros(32, n // 32, device=DEV)
+        k.q80_quantize(x, q, s, bs)
+        part = torch.zeros(16 * 32 * d, device=DEV)
+        for use_part in (False, True):
+            y1 = torch.zeros(32, d, device=DEV)
+            y2 = torch.zeros(32, d, device=DEV)
+            pa = part if use_part else None
+            k.q40_gemm(qs, sc, q, s, y1, B, pa)
+            k.q40_gemm(qs, sc, q, s, y2, B, pa, use_v2=True)
+            assert torch.allclose(y2[:B], y1[:B], atol=1e-3, rtol=1e-4), \
+                (d, n, B, use_part,
+                 (y2[:B] - y1[:B]).abs().max().item())
