@@ -834,6 +834,12 @@ __global__ void k_attn_combine(const float *__restrict__ ml_scratch,
 // Grouped (MoE) variant: weights [n_experts, d, ...]; slot s uses expert
 // expert_idx[s] and input row slot_batch[s]; y [S, d]
 // (reference 3-D expert matmul with index indirection, nn-core.hpp:209-213).
+// LPP = lanes per 2-row pair. 64 is the production default; smaller LPP
+// (round-2 experimental, DLLAMA_MOE_V2=1) tiles a wave into 64/LPP
+// row-pair groups so every lane stays busy at MoE shapes where nbp < 64
+// (Qwen3-30B w2: nbp=12 left 52 of 64 lanes idle -> the ~5x-off-stream
+// grouped GEMV; see tools/moe_gemv_probe.hip).
+template <int LPP>
 __global__ void k_q40_gemv_grouped(const uint8_t *__restrict__ qs,
                                    const __half *__restrict__ scales,
                                    const int8_t *__restrict__ xq,
@@ -842,11 +848,14 @@ __global__ void k_q40_gemv_grouped(const uint8_t *__restrict__ qs,
                                    const int *__restrict__ expert_idx,
                                    float *__restrict__ y,
                                    int d, int n, int k_slots) {
+    constexpr int NGRP = WAVE / LPP;
     const int wpb = blockDim.x / WAVE;
-    const int row0 = (blockIdx.x * wpb + threadIdx.x / WAVE) * 2;  // 2 rows/wave
+    const int full_lane = threadIdx.x % WAVE;
+    const int grp = full_lane / LPP;
+    const int row0 = ((blockIdx.x * wpb + threadIdx.x / WAVE) * NGRP + grp) * 2;
     const int slot = blockIdx.y;
     if (row0 >= d) return;
-    const int lane = threadIdx.x % WAVE;
+    const int lane = full_lane % LPP;
     const int nb = n / QB;
     const int nbp = nb >> 1;
     const int e = expert_idx[slot];
@@ -859,7 +868,7 @@ __global__ void k_q40_gemv_grouped(const uint8_t *__restrict__ qs,
     const __half *srow0 = scales + ((int64_t)e * d + row0) * nb;
     const __half *srow1 = scales + ((int64_t)e * d + row1) * nb;
     float acc0 = 0.0f, acc1 = 0.0f;
-    for (int jp = lane; jp < nbp; jp += WAVE) {
+    for (int jp = lane; jp < nbp; jp += LPP) {
         const int j = jp << 1;
         const uint4 a0 = wrow0[j], a1 = wrow0[j + 1];
         const uint4 b0 = wrow1[j], b1 = wrow1[j + 1];
@@ -884,11 +893,14 @@ __global__ void k_q40_gemv_grouped(const uint8_t *__restrict__ qs,
                     (float)q40_block_dot(wrow1[j], xb[0], xb[1])
                     - 8.0f * xbs[(int64_t)b * nb + j], acc1);
     }
-    const float r0 = wave_reduce_sum(acc0);
-    const float r1 = wave_reduce_sum(acc1);
+    #pragma unroll
+    for (int o = LPP / 2; o > 0; o >>= 1) {
+        acc0 += __shfl_down(acc0, o, WAVE);
+        acc1 += __shfl_down(acc1, o, WAVE);
+    }
     if (lane == 0) {
-        y[(int64_t)slot * d + row0] = r0;
-        if (row0 + 1 < d) y[(int64_t)slot * d + row0 + 1] = r1;
+        y[(int64_t)slot * d + row0] = acc0;
+        if (row0 + 1 < d) y[(int64_t)slot * d + row0 + 1] = acc1;
     }
 }
 
@@ -1754,19 +1766,39 @@ int64_t q40_gemv_argmax_blocks(int64_t d) {
 
 void q40_gemv_grouped(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
                       torch::Tensor xs, torch::Tensor xbs, torch::Tensor expert_idx,
-                      torch::Tensor y, int64_t k_slots) {
+                      torch::Tensor y, int64_t k_slots, bool use_v2 = false) {
     CHECK_CUDA(qs); CHECK_CONT(qs);
     const int d = qs.size(1);
     const int n = qs.size(2) * 2;
     const int n_slots = expert_idx.numel();
     const int waves_per_block = 4;
-    const dim3 grid(ceil_div(d, waves_per_block), n_slots);
-    hipLaunchKernelGGL(k_q40_gemv_grouped, grid, dim3(waves_per_block * WAVE), 0,
-                       cur_stream(), qs.data_ptr<uint8_t>(),
-                       reinterpret_cast<const __half *>(scales.data_ptr<at::Half>()),
-                       xq.data_ptr<int8_t>(), xs.data_ptr<float>(),
-                       xbs.data_ptr<float>(), expert_idx.data_ptr<int>(),
-                       y.data_ptr<float>(), d, n, (int)k_slots);
+    static const bool env_v2 =
+        std::getenv("DLLAMA_MOE_V2") && atoi(std::getenv("DLLAMA_MOE_V2")) == 1;
+    // v2: LPP = smallest power of two >= nbp (clamped [8, 64]) so every
+    // lane has work; rows per wave scale up by 64/LPP
+    int lpp = 64;
+    if (use_v2 || env_v2) {
+        const int nbp = (n / QB) >> 1;
+        lpp = 8;
+        while (lpp < nbp && lpp < 64) lpp <<= 1;
+    }
+    const int rows_per_wg = waves_per_block * 2 * (WAVE / lpp);
+    const dim3 grid(ceil_div(d, rows_per_wg), n_slots);
+    const dim3 block(waves_per_block * WAVE);
+    auto launch = [&](auto k) {
+        hipLaunchKernelGGL(k, grid, block, 0, cur_stream(),
+                           qs.data_ptr<uint8_t>(),
+                           reinterpret_cast<const __half *>(scales.data_ptr<at::Half>()),
+                           xq.data_ptr<int8_t>(), xs.data_ptr<float>(),
+                           xbs.data_ptr<float>(), expert_idx.data_ptr<int>(),
+                           y.data_ptr<float>(), d, n, (int)k_slots);
+    };
+    switch (lpp) {
+        case 8: launch(k_q40_gemv_grouped<8>); break;
+        case 16: launch(k_q40_gemv_grouped<16>); break;
+        case 32: launch(k_q40_gemv_grouped<32>); break;
+        default: launch(k_q40_gemv_grouped<64>); break;
+    }
 }
 
 void rope(torch::Tensor x, torch::Tensor cache, torch::Tensor pos,
@@ -2185,7 +2217,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("router_gemv", &router_gemv);
     m.def("norm_f32", &norm_f32);
     m.def("add_ssq", &add_ssq);
-    m.def("q40_gemv_grouped", &q40_gemv_grouped);
+    m.def("q40_gemv_grouped", &q40_gemv_grouped, py::arg("qs"),
+          py::arg("scales"), py::arg("xq"), py::arg("xs"), py::arg("xbs"),
+          py::arg("expert_idx"), py::arg("y"), py::arg("k_slots"),
+          py::arg("use_v2") = false);
     m.def("rope", &rope);
     m.def("rope_kv", &rope_kv);
     m.def("kv_append", &kv_append);

@@ -446,3 +446,27 @@ def test_q40_gemm_v2_matches_v1(k):
             assert torch.allclose(y2[:B], y1[:B], atol=1e-3, rtol=1e-4), \
                 (d, n, B, use_part,
                  (y2[:B] - y1[:B]).abs().max().item())
+
+
+@pytest.mark.skipif(not __import__("os").environ.get("DLLAMA_EXPERIMENTAL"),
+                    reason="round-2 experimental kernel (DLLAMA_EXPERIMENTAL=1)")
+def test_q40_gemv_grouped_v2_matches_v1(k):
+    """Lane-tiled grouped GEMV (use_v2 / DLLAMA_MOE_V2) vs production at
+    MoE-like shapes, including tiny nbp (w2-shape n=768 -> nbp=12)."""
+    for d, n in ((128, 512), (96, 768), (160, 2048)):
+        E, B, ka = 4, 2, 2
+        lins = [_mk_linear(d, n, 70 + e) for e in range(E)]
+        qs = torch.stack([l[0] for l in lins])
+        sc = torch.stack([l[1] for l in lins])
+        x = rand(B, n, seed=71, scale=0.5)
+        q = torch.zeros(B, n, dtype=torch.int8, device=DEV)
+        s = torch.zeros(B, n // 32, device=DEV)
+        bs = torch.zeros(B, n // 32, device=DEV)
+        k.q80_quantize(x, q, s, bs)
+        idx = torch.tensor([1, 3, 0, 2], dtype=torch.int32, device=DEV)
+        y1 = torch.zeros(B * ka, d, device=DEV)
+        y2 = torch.zeros(B * ka, d, device=DEV)
+        k.q40_gemv_grouped(qs, sc, q, s, bs, idx, y1, ka)
+        k.q40_gemv_grouped(qs, sc, q, s, bs, idx, y2, ka, use_v2=True)
+        assert torch.allclose(y2, y1, atol=1e-4, rtol=1e-5), \
+            (d, n, (y2 - y1).abs().max().item())
