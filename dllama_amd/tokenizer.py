@@ -412,6 +412,9 @@ class ChatTemplateGenerator:
             for it in items:
                 if it.role in ("system", "user", "assistant"):
                     out.append("<|im_start|>" + it.role + "\n" + it.message + "<|im_end|>\n")
+                # quirk kept for parity: the reference appends the generation
+                # prompt INSIDE the item loop (tokenizer.cpp:615-627), i.e.
+                # once per message, not once at the end
                 if append_generation_prompt:
                     out.append("<|im_start|>assistant\n")
         return GeneratedChat("".join(out), public_prompt)

@@ -76,6 +76,41 @@ def test_template_llama3_output():
                            "<|start_header_id|>assistant<|end_header_id|>\n\n")
 
 
+def test_template_llama2_output():
+    """Golden vs reference tokenizer.cpp:578-591 ([INST]/<<SYS>> fusion of
+    a leading system+user pair, eos after every turn)."""
+    g = tok.ChatTemplateGenerator(tok.TEMPLATE_LLAMA2, None, "</s>")
+    out = g.generate([tok.ChatItem("system", "S"),
+                      tok.ChatItem("user", "U1"),
+                      tok.ChatItem("assistant", "A1"),
+                      tok.ChatItem("user", "U2")], True)
+    assert out.content == ("[INST] <<SYS>>\nS\n<</SYS>>\n\nU1 [/INST]</s>"
+                           "A1</s>[INST] U2 [/INST]</s>")
+
+
+def test_template_deepseek3_output():
+    """Golden vs reference tokenizer.cpp:597-614: bare system prefix,
+    <think> generation prompt whose public part is the trailing 8 bytes."""
+    g = tok.ChatTemplateGenerator(tok.TEMPLATE_DEEP_SEEK3, None, "<eos>")
+    out = g.generate([tok.ChatItem("system", "S"),
+                      tok.ChatItem("user", "U")], True)
+    assert out.content == "S<\uff5cUser\uff5c>U<\uff5cAssistant\uff5c><think>\n"
+    assert out.public_prompt == "<think>\n"
+
+
+def test_template_chatml_output():
+    """Golden vs reference tokenizer.cpp:615-627 INCLUDING its quirk: the
+    generation prompt is appended inside the per-item loop (once per
+    message), kept for exact parity."""
+    g = tok.ChatTemplateGenerator(tok.TEMPLATE_CHATML, None, "<eos>")
+    out = g.generate([tok.ChatItem("user", "U")], True)
+    assert out.content == "<|im_start|>user\nU<|im_end|>\n<|im_start|>assistant\n"
+    out2 = g.generate([tok.ChatItem("user", "U"),
+                       tok.ChatItem("assistant", "A")], True)
+    assert out2.content == ("<|im_start|>user\nU<|im_end|>\n<|im_start|>assistant\n"
+                            "<|im_start|>assistant\nA<|im_end|>\n<|im_start|>assistant\n")
+
+
 def test_eos_detector_exact():
     d = tok.EosDetector([99], ["<stop>"], 0, 0)
     assert d.append(1, "<sto") == tok.MAYBE_EOS
