@@ -73,3 +73,27 @@ def test_moe_walk():
     assert ("block_matmul_w2", 3) in names
     assert ("block_norm_q", -1) in names  # qwen3 qk-norm present
     assert h.rope_type == mf.ROPE_FALCON  # qwen3 forces falcon rope
+
+
+def test_bad_magic_and_truncation(tmp_path):
+    """Corrupt .m files produce clear errors, not garbage loads (reference
+    llm.cpp:36-116 asserts the magic and walks to EOF)."""
+    import pytest
+    from dllama_amd.utils.testing import make_tiny_llama
+    bad = tmp_path / "bad.m"
+    bad.write_bytes(b"\x00" * 64)
+    with pytest.raises(ValueError, match="magic"):
+        mf.read_header(str(bad))
+
+    p = tmp_path / "trunc.m"
+    make_tiny_llama(str(p))
+    data = p.read_bytes()
+    p.write_bytes(data[: len(data) - 100])
+    with pytest.raises(ValueError, match="missing"):
+        mf.ModelFile(str(p))
+
+
+def test_unknown_preset():
+    import pytest
+    with pytest.raises(KeyError, match="unknown preset"):
+        mf.preset_header("llama-9000b")
