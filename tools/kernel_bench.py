@@ -106,3 +106,32 @@ for d, n, tag in ((6144, 4096, "qkv"), (28672, 4096, "w13"), (128256, 4096, "cls
     t_sep = bench(sep, n=64)
     t_pro = bench(lambda: k.q40_gemv_nq(qs, sc, xf, wn, sq, 1e-5, y, 1), n=64)
     print(f"PRO {tag:4s}: norm_quant+gemv={t_sep:6.2f}us  fused-nq={t_pro:6.2f}us")
+
+# prefill GEMM v1 vs v2 (round-2 DLLAMA_GEMM_V2 candidate), batch 32
+print("\n-- GEMM v1 vs v2 (batch 32) --")
+part = torch.zeros(16 * 32 * 28672, device=dev)
+for d, n, tag in ((6144, 4096, "qkv"), (4096, 14336, "w2"), (28672, 4096, "w13")):
+    qs, sc = mklin(d, n)
+    xq, xs, _ = mkx(32, n)
+    y = torch.zeros(32, d, device=dev)
+    t1 = bench(lambda: k.q40_gemm(qs, sc, xq, xs, y, 32, part), n=32)
+    t2 = bench(lambda: k.q40_gemm(qs, sc, xq, xs, y, 32, part, use_v2=True), n=32)
+    ideal = (d * n * 0.5625) / 6.3e3 / 1000
+    print(f"gemm {tag:4s} v1={t1:7.2f}us v2={t2:7.2f}us ideal={ideal:6.2f}us "
+          f"v2-eff={100*ideal/t2:4.0f}%")
+
+# grouped GEMV v1 vs v2 at Qwen3-30B-A3B decode shapes
+print("\n-- grouped GEMV v1 vs v2 (qwen3-30b shapes, 8 slots) --")
+for d, n, ks, tag in ((1536, 2048, 8, "w13"), (2048, 768, 1, "w2")):
+    E = 16
+    qs = torch.randint(0, 256, (E, d, n // 2), dtype=torch.uint8, device=dev, generator=g)
+    sc = (torch.rand((E, d, n // 32), device=dev, generator=g) * 0.01).to(torch.float16)
+    nx = 8 // ks
+    xq, xs, xbs = mkx(max(1, nx), n)
+    idx = torch.arange(8, dtype=torch.int32, device=dev) % E
+    y = torch.zeros(8, d, device=dev)
+    t1 = bench(lambda: k.q40_gemv_grouped(qs, sc, xq, xs, xbs, idx, y, ks), n=64)
+    t2 = bench(lambda: k.q40_gemv_grouped(qs, sc, xq, xs, xbs, idx, y, ks, use_v2=True), n=64)
+    ideal = (8 * d * n * 0.5625) / 6.3e3 / 1000
+    print(f"grouped {tag:4s} v1={t1:7.2f}us v2={t2:7.2f}us ideal={ideal:6.2f}us "
+          f"v2-eff={100*ideal/t2:4.0f}%")
