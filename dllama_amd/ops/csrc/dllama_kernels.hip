@@ -1032,7 +1032,6 @@ k_q40_gemm_v2(const uint8_t *__restrict__ qs,
               float *__restrict__ y,
               float *__restrict__ part,
               int d, int n, int batch) {
-    constexpr int RING = 4;
     __shared__ int8_t lds_a[2][GEMM_V2_CHUNK][32][QB];
     __shared__ float lds_s[2][GEMM_V2_CHUNK][32];
     const int wave = threadIdx.x / WAVE;
@@ -1077,12 +1076,16 @@ k_q40_gemm_v2(const uint8_t *__restrict__ qs,
     for (int r = 0; r < 8; r++) facc[r] = make_float2(0.0f, 0.0f);
 
     stage(0, j0);
-    uint4 ring[RING];
-    __half ring_s[RING];
-    #pragma unroll
-    for (int p = 0; p < RING; p++) {
-        const int j = j0 + p;
-        if (live && j < j1) { ring[p] = wrow[j]; ring_s[p] = srow[j]; }
+    // weight pipeline as NAMED scalars (cur/next): an indexed prefetch ring
+    // array gets allocated to scratch (80 B/lane round-trip in the hot
+    // loop — final code-object metadata shows it even when -Rpass remarks
+    // claim zero spill); two named uint4s cannot spill, and the compiler's
+    // vmcnt counting still overlaps the next load with the current MFMA
+    uint4 wq_cur = {};
+    float sw_cur = 0.0f;
+    if (live && j0 < j1) {
+        wq_cur = wrow[j0];
+        sw_cur = __half2float(srow[j0]);
     }
     __syncthreads();
 
@@ -1095,11 +1098,12 @@ k_q40_gemm_v2(const uint8_t *__restrict__ qs,
             for (int jj = 0; jj < GEMM_V2_CHUNK; jj++) {
                 if (jj >= nblk) break;
                 const int j = jc + jj;
-                const int slot = (j - j0) % RING;  // CHUNK % RING == 0: static
-                const uint4 wq = ring[slot];
-                const float sw = __half2float(ring_s[slot]);
-                const int jn = j + RING;
-                if (jn < j1) { ring[slot] = wrow[jn]; ring_s[slot] = srow[jn]; }
+                const uint4 wq = wq_cur;
+                const float sw = sw_cur;
+                if (j + 1 < j1) {
+                    wq_cur = wrow[j + 1];
+                    sw_cur = __half2float(srow[j + 1]);
+                }
                 v4i32_t a = *reinterpret_cast<const v4i32_t *>(
                     &lds_a[buf][jj][lane & 31][khi * 16]);
                 v4i32_t b;
