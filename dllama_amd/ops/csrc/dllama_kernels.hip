@@ -1615,37 +1615,58 @@ static void gemv_launch(torch::Tensor &qs, torch::Tensor &scales, torch::Tensor 
     };
     std::integral_constant<int, 1> r1;
     std::integral_constant<int, 2> r2;
-    if (EPI == EPI_ROPE) {
-        // rotation pairs require RPW=2 at every batch size
+    // constexpr: a plain `if` would still instantiate the other branch's
+    // kernels (dead RPW combinations that additionally spill)
+    if constexpr (EPI == EPI_ROPE) {
+        // rotation pairs require RPW=2 at every batch size.
+        // (`else` below matters: code after a returning `if constexpr` is
+        // still instantiated, which kept dead spilled variants alive)
         switch (batch) {
             case 1: launch(std::integral_constant<int, 1>{}, r2); break;
             case 2: launch(std::integral_constant<int, 2>{}, r2); break;
             case 4: launch(std::integral_constant<int, 4>{}, r2); break;
-            case 8: launch(std::integral_constant<int, 8>{}, r2); break;
-            case 16: launch(std::integral_constant<int, 16>{}, r2); break;
-            case 32: launch(std::integral_constant<int, 32>{}, r2); break;
+            case 8:
+            case 16:
+            case 32:
+                // PRO (fused norm prologue) is a batch<=4 decode path
+                if constexpr (PRO) {
+                    TORCH_CHECK(false, "fused-norm GEMV supports batch <= 4");
+                } else {
+                    if (batch == 8) launch(std::integral_constant<int, 8>{}, r2);
+                    else if (batch == 16) launch(std::integral_constant<int, 16>{}, r2);
+                    else launch(std::integral_constant<int, 32>{}, r2);
+                }
+                break;
             default: TORCH_CHECK(false, "unsupported batch ", batch);
         }
-        return;
-    }
-    switch (batch) {
-        case 1: {
-            // RPW=2 halves the wave count for 2x per-lane loads in flight;
-            // tunable crossover (DLLAMA_RPW1_MAX: use RPW1 up to that d)
-            static const int rpw1_max =
-                std::getenv("DLLAMA_RPW1_MAX") ? atoi(std::getenv("DLLAMA_RPW1_MAX")) : 0;
-            if (e.force_rpw2 || (d >= 2048 && d > rpw1_max))
-                launch(std::integral_constant<int, 1>{}, r2);
-            else
-                launch(std::integral_constant<int, 1>{}, r1);
-            break;
+    } else {
+        switch (batch) {
+            case 1: {
+                // RPW=2 halves the wave count for 2x per-lane loads in
+                // flight; tunable crossover (DLLAMA_RPW1_MAX)
+                static const int rpw1_max =
+                    std::getenv("DLLAMA_RPW1_MAX") ? atoi(std::getenv("DLLAMA_RPW1_MAX")) : 0;
+                if (e.force_rpw2 || (d >= 2048 && d > rpw1_max))
+                    launch(std::integral_constant<int, 1>{}, r2);
+                else
+                    launch(std::integral_constant<int, 1>{}, r1);
+                break;
+            }
+            case 2: launch(std::integral_constant<int, 2>{}, r2); break;
+            case 4: launch(std::integral_constant<int, 4>{}, r2); break;
+            case 8:
+            case 16:
+            case 32:
+                if constexpr (PRO) {
+                    TORCH_CHECK(false, "fused-norm GEMV supports batch <= 4");
+                } else {
+                    if (batch == 8) launch(std::integral_constant<int, 8>{}, r1);
+                    else if (batch == 16) launch(std::integral_constant<int, 16>{}, r1);
+                    else launch(std::integral_constant<int, 32>{}, r1);
+                }
+                break;
+            default: TORCH_CHECK(false, "unsupported batch ", batch);
         }
-        case 2: launch(std::integral_constant<int, 2>{}, r2); break;
-        case 4: launch(std::integral_constant<int, 4>{}, r2); break;
-        case 8: launch(std::integral_constant<int, 8>{}, r1); break;
-        case 16: launch(std::integral_constant<int, 16>{}, r1); break;
-        case 32: launch(std::integral_constant<int, 32>{}, r1); break;
-        default: TORCH_CHECK(false, "unsupported batch ", batch);
     }
 }
 
