@@ -160,3 +160,29 @@ def test_sampler_deterministic_seed():
     b = tok.Sampler(100, 0.8, 0.9, 7)
     logits = np.random.default_rng(0).standard_normal(100).astype(np.float32)
     assert [a.sample(logits) for _ in range(10)] == [b.sample(logits) for _ in range(10)]
+
+
+def test_sampler_torch_path_matches_numpy():
+    """sample_torch (the GPU serving path — works on CPU tensors too) must
+    track the numpy reference path: identical greedy picks, and identical
+    sampled tokens for the same xorshift coin on a well-separated
+    distribution (boundary-exact ties are measure-zero and excluded by
+    construction)."""
+    import torch
+    import numpy as np
+    logits = np.zeros(64, dtype=np.float32)
+    logits[7], logits[21], logits[42] = 8.0, 6.0, 4.0  # dominant trio
+    t = torch.tensor(logits)
+
+    g1 = tok.Sampler(64, 0.0, 0.9, 3)
+    g2 = tok.Sampler(64, 0.0, 0.9, 3)
+    assert g1.sample(logits) == g2.sample_torch(t) == 7
+
+    for topp in (0.95, 1.0):
+        s1 = tok.Sampler(64, 1.0, topp, 1234)
+        s2 = tok.Sampler(64, 1.0, topp, 1234)
+        picks1 = [s1.sample(logits) for _ in range(100)]
+        picks2 = [s2.sample_torch(t) for _ in range(100)]
+        assert picks1 == picks2, (topp, picks1[:5], picks2[:5])
+        # the dominant trio carries ~98% of the mass
+        assert sum(p in (7, 21, 42) for p in picks1) >= 90
