@@ -178,21 +178,41 @@ def run_chat(args) -> int:
     stops = chat_stops(tok)
     quiet = comm.rank != 0
     is_start = True
+    # optional system prompt, included in the first turn's delta items
+    # (reference dllama.cpp:182-185)
+    try:
+        sys_prompt = input("💻 System prompt (optional): ") if not quiet else input()
+    except EOFError:
+        return 0
     while True:
         try:
-            user = input("\n💻 > ") if not quiet else input()
+            user = input("\n👱 User\n> ") if not quiet else input()
         except EOFError:
             return 0
         if not user.strip():
             continue
-        items = [ChatItem("user", user)]
-        text = gen.generate(items, True).content
-        tokens = tok.encode(text, is_start=is_start)
+        items = []
+        if is_start and sys_prompt.strip():
+            items.append(ChatItem("system", sys_prompt))
+        items.append(ChatItem("user", user))
+        chat_out = gen.generate(items, True)
+        text = chat_out.content
+        try:
+            tokens = tok.encode(text, is_start=is_start)
+        except ValueError:
+            print("(encode error)", file=sys.stderr)
+            continue
         is_start = False
+        if engine.pos + len(tokens) + 1 >= m.header.seq_len:
+            print("(end of context)")  # reference dllama.cpp:257
+            return 0
         detector = EosDetector(tok.eos_token_ids, stops)
         tok.reset_decoder()
         if not quiet:
-            print("🤖 ", end="", flush=True)
+            # deepseek's "<think>\n" tail is part of the prompt but shown as
+            # assistant output (reference publicPrompt, dllama.cpp:233-235)
+            print("\n🤖 Assistant\n" + (chat_out.public_prompt or ""),
+                  end="", flush=True)
 
         def on_token(t):
             # route through the streaming decoder first (UTF-8-safe pieces;

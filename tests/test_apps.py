@@ -185,7 +185,7 @@ def test_cli_chat(assets, capsys, monkeypatch):
     """Chat REPL smoke: one user turn then EOF exits cleanly."""
     from dllama_amd.apps import main as main_mod
     mp, tp = assets
-    answers = iter(["hello there"])
+    answers = iter(["", "hello there"])  # empty system prompt, one user turn
 
     def fake_input(prompt=""):
         try:
@@ -355,3 +355,26 @@ def test_api_error_isolation(assets):
     finally:
         server.shutdown()
         server.server_close()
+
+
+def test_cli_chat_system_prompt_and_context_end(assets, capsys, monkeypatch):
+    """System prompt is included in the first turn (reference
+    dllama.cpp:182-185) and a full context window ends the REPL with
+    '(end of context)' instead of crashing (dllama.cpp:257)."""
+    from dllama_amd.apps import main as main_mod
+    mp, tp = assets
+    answers = iter(["be brief"] + ["tell me more " * 40] * 20)
+
+    def fake_input(prompt=""):
+        try:
+            return next(answers)
+        except StopIteration:
+            raise EOFError
+
+    monkeypatch.setattr("builtins.input", fake_input)
+    rc = main_mod.main(["chat", "--model", mp, "--tokenizer", tp,
+                        "--steps", "8", "--temperature", "0",
+                        "--gpu-index", "-1"])
+    assert rc == 0
+    out = capsys.readouterr().out
+    assert "(end of context)" in out
