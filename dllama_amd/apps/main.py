@@ -257,7 +257,8 @@ def run_perplexity(args) -> int:
         pos += len(chunk)
     ppl = float(np.exp(nll / count))
     if comm.rank == 0:
-        print(f"Perplexity: {ppl:.4f}  (nll/token {nll / count:.4f}, {count} tokens)")
+        print(f"Perplexity: {ppl:.4f}  (nll/token {nll / count:.4f}, "
+              f"bitPerToken {nll / count / np.log(2.0):.4f}, {count} tokens)")
     return 0
 
 
