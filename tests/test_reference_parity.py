@@ -25,7 +25,18 @@ def test_greedy_parity_with_reference(tmp_path_factory):
     except Exception as e:  # noqa: BLE001  (environment-dependent toolchain)
         pytest.skip(f"could not build the reference binary: {e}")
     for arch in ("llama", "qwen3", "qwen3_moe"):
-        model, tok = rp.make_ascii_assets(workdir, arch=arch)
+        # per-arch vocab: see the comment in reference_parity.main()
+        model, tok = rp.make_ascii_assets(
+            workdir, vocab_size=101 if arch == "llama" else 128, arch=arch)
         ref = rp.run_reference(binary, model, tok, "hello world, this is", 48)
         ours = rp.run_ours(model, tok, "hello world, this is", 48)
         assert ref == ours, f"{arch}: {ref!r} != {ours!r}"
+
+    # multi-turn chat parity (template generation, KV continuity across
+    # turns, EOS detection) against the reference's interactive loop
+    model, tok = rp.make_ascii_assets(workdir, arch="llama")
+    sys_prompt, users = "keep it short", ["hello ab", "more cd"]
+    ref_turns = rp.run_reference_chat(binary, model, tok, sys_prompt, users)
+    our_turns = rp.run_ours_chat(model, tok, sys_prompt, users)
+    assert len(ref_turns) == 2, ref_turns
+    assert ref_turns == our_turns, (ref_turns, our_turns)

@@ -35,18 +35,25 @@ def build_reference(workdir: str) -> str:
     return binary
 
 
-def make_ascii_assets(workdir: str, vocab_size: int = 128, arch: str = "llama"):
+def make_ascii_assets(workdir: str, vocab_size: int = 101, arch: str = "llama"):
     """Tokenizer whose every regular token is printable ASCII -> decode is
-    lossless text on both sides."""
-    printable = [chr(c).encode() for c in range(0x20, 0x7F)]  # 95 tokens
-    filler = [b"~"] * (vocab_size - 5 - len(printable))       # dup ids unused
-    vocab = printable + filler
+    lossless text on both sides. No duplicate filler tokens: the structured
+    model's greedy walk visits consecutive ids, and duplicate-embedding
+    runs have near-tie margins that flip between the reference's int8 path
+    and the f32 oracle on long chat decodes."""
+    # newline included: the llama3 chat template emits "\n\n"
+    printable = [b"\n"] + [chr(c).encode() for c in range(0x20, 0x7F)]
+    vocab = list(printable) + [b"~"] * (vocab_size - 5 - len(printable))
     scores = [0.0] * len(vocab)
     bos_id = len(vocab)
-    vocab += [b"<|begin|>", b"<|end|>", b"<|h|>", b"<|e|>", b"<|eot|>"]
+    # llama3-named specials so the chat template's literal special-token
+    # strings are encodable (chat-mode parity needs this)
+    vocab += [b"<|begin_of_text|>", b"<|end_of_text|>", b"<|start_header_id|>",
+              b"<|end_header_id|>", b"<|eot_id|>"]
     scores += [0.0] * 5
     tok_path = os.path.join(workdir, "parity.t")
-    tk.write_tokenizer(tok_path, vocab, scores, bos_id, True, [bos_id + 1],
+    tk.write_tokenizer(tok_path, vocab, scores, bos_id, True,
+                       [bos_id + 1, bos_id + 4],  # end_of_text, eot_id
                        "{{<|start_header_id|>}}")
 
     if arch == "qwen3_moe":
@@ -133,13 +140,117 @@ def run_ours(model, tok_path, prompt, steps) -> str:
     return "".join(p for p in (tok.decode(t) for t in out) if p)
 
 
+def run_reference_chat(binary, model, tok, sys_prompt, users):
+    """Scripted multi-turn chat against the reference binary; returns the
+    assistant text of each turn. The reference spins on stdin EOF, so a
+    timeout + kill is part of the protocol (partial output is kept)."""
+    stdin = sys_prompt + "\n" + "\n".join(users) + "\n"
+    # No stdbuf here: its LD_PRELOAD shifts the heap layout and the
+    # reference chat's handoff token (`inputTokens[i + 1]` one past the
+    # prompt, dllama.cpp:223) is UNINITIALIZED memory — zero in the plain
+    # environment (what this harness locks in), garbage -> segfault under
+    # stdbuf. Pipe stdout is fully buffered, so completion cannot be
+    # detected incrementally; keep stdin OPEN after the scripted turns (on
+    # EOF readStdin re-prompts in a tight loop, GBs of spam; blocked in
+    # fgets it is quiet), give the turns a fixed window, then kill and
+    # collect the buffered output.
+    p = subprocess.Popen(
+        [binary, "chat", "--model", model, "--tokenizer", tok,
+         "--temperature", "0", "--nthreads", "2", "--buffer-float-type",
+         "q80", "--max-seq-len", "256"],
+        stdin=subprocess.PIPE, stdout=subprocess.PIPE,
+        stderr=subprocess.DEVNULL)
+    p.stdin.write(stdin.encode())
+    p.stdin.flush()
+    try:
+        p.wait(timeout=30)
+    except subprocess.TimeoutExpired:
+        p.kill()
+    p.wait()
+    out = p.stdout.read().decode("utf-8", "replace")
+    p.stdout.close()
+    p.stdin.close()
+    turns = []
+    for part in out.split("🤖 Assistant\n")[1:]:
+        for stop in ("\n👱 User", "👱 User", "(end of context)"):
+            i = part.find(stop)
+            if i >= 0:
+                part = part[:i]
+        turns.append(part)
+    return turns
+
+
+def run_ours_chat(model, tok_path, sys_prompt, users):
+    """Multi-turn chat with THIS framework's template generator, tokenizer,
+    EOS detector and CPU model — reproducing the reference chat loop's
+    prompt handoff (dllama.cpp:223: after the prefill loop the next decode
+    token is read one past the encoded prompt, i.e. a zero) so trajectories
+    are comparable. Returns per-turn assistant text."""
+    import torch
+    from dllama_amd.models.config import ModelConfig
+    from dllama_amd.models.cpu_model import CpuTransformer
+    from dllama_amd.tokenizer import (ChatItem, ChatTemplateGenerator,
+                                      EosDetector, Tokenizer, chat_stops,
+                                      EOS, MAYBE_EOS, TEMPLATE_UNKNOWN)
+    m = mf.ModelFile(model, max_seq_len=256)
+    tok = Tokenizer(tok_path)
+    net = CpuTransformer(m, ModelConfig.from_header(m.header))
+    seq_len = m.header.seq_len
+    stops = chat_stops(tok)
+    max_stop = max(len(s) for s in stops)
+    gen = ChatTemplateGenerator(TEMPLATE_UNKNOWN, tok.chat_template,
+                                stops[0])
+    pos = 0
+    turns = []
+    for i, user in enumerate(users):
+        items = []
+        if i == 0 and sys_prompt:
+            items.append(ChatItem("system", sys_prompt))
+        items.append(ChatItem("user", user))
+        text = gen.generate(items, True).content
+        tokens = tok.encode(text, is_start=(pos == 0))
+        if pos + len(tokens) - 1 >= seq_len:
+            break
+        # prefill tokens[:-1]; the reference's decode handoff token is the
+        # uninitialized slot one past the prompt (zero in practice)
+        n_feed = len(tokens) - 1
+        for j in range(0, n_feed, 32):
+            chunk = tokens[j: min(j + 32, n_feed)]
+            net.forward(torch.tensor(chunk),
+                        torch.arange(pos, pos + len(chunk)))
+            pos += len(chunk)
+        token = 0
+        tok.reset_decoder()
+        det = EosDetector(tok.eos_token_ids, stops, max_stop, max_stop)
+        out = []
+        while pos < seq_len:
+            logits = net.forward(torch.tensor([token]), torch.tensor([pos]))
+            token = int(torch.argmax(logits[0]))
+            pos += 1
+            kind = det.append(token, tok.decode(token))
+            if kind != MAYBE_EOS:
+                delta = det.get_delta()
+                if delta:
+                    out.append(delta)
+                det.reset()
+            if kind == EOS:
+                break
+        turns.append("".join(out))
+    return turns
+
+
 def main():
     workdir = sys.argv[1] if len(sys.argv) > 1 else tempfile.mkdtemp(prefix="parity")
     os.makedirs(workdir, exist_ok=True)
     binary = build_reference(workdir)
     rc = 0
     for arch in ("llama", "qwen3", "qwen3_moe"):
-        model, tok = make_ascii_assets(workdir, arch=arch)
+        # per-arch vocab picks a trajectory with decisive greedy margins
+        # (a near-tie anywhere flips between the reference's int8 path and
+        # the f32 oracle); chat needs the filler-free 101 vocab (duplicate
+        # '~' embeddings = near-tie runs on long decodes)
+        model, tok = make_ascii_assets(
+            workdir, vocab_size=101 if arch == "llama" else 128, arch=arch)
         prompt = "hello world, this is"
         steps = 48
         ref_text = run_reference(binary, model, tok, prompt, steps)
@@ -153,6 +264,21 @@ def main():
             n = sum(1 for a, b in zip(ref_text, our_text) if a == b)
             print(f"❌ [{arch}] MISMATCH after {n} matching chars")
             rc = 1
+
+    # chat-mode parity (llama): template generation, multi-turn KV
+    # continuity, EOS detection — vs the reference's interactive loop
+    model, tok = make_ascii_assets(workdir, arch="llama")
+    sys_prompt, users = "keep it short", ["hello ab", "more cd"]
+    ref_turns = run_reference_chat(binary, model, tok, sys_prompt, users)
+    our_turns = run_ours_chat(model, tok, sys_prompt, users)
+    print(f"[chat] reference turns: {[t[:40] for t in ref_turns]!r}")
+    print(f"[chat] dllama_amd turns:{[t[:40] for t in our_turns]!r}")
+    if ref_turns and ref_turns == our_turns:
+        print(f"✅ [chat] PARITY: {len(ref_turns)} multi-turn assistant "
+              "replies identical (template + KV continuity + EOS)")
+    else:
+        print("❌ [chat] MISMATCH")
+        rc = 1
     return rc
 
 
