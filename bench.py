@@ -42,12 +42,17 @@ def _build_cpu_smoke(args, world, comm):
     from dllama_amd.models.config import ModelConfig
     from dllama_amd.models.cpu_model import CpuTransformer
     from dllama_amd.quants import F32, Q80
-    from dllama_amd.utils.testing import make_tiny_llama
 
     path = os.path.join(tempfile.gettempdir(),
                         f"dllama_bench_smoke_{os.getppid()}.m")
     if comm.rank == 0 and not os.path.exists(path):
-        make_tiny_llama(path, vocab_size=256)
+        # 8 kv heads so the smoke works at any driver TP degree (1..8)
+        h = mf.LlmHeader(arch_type=mf.ARCH_LLAMA, dim=64, hidden_dim=128,
+                         n_layers=2, n_heads=8, n_kv_heads=8, head_dim=64,
+                         vocab_size=256, seq_len=128, rope_theta=10000,
+                         rope_type=mf.ROPE_LLAMA)
+        h.finalize()
+        mf.write_synthetic_model(path, h, seed=5)
     comm.barrier()
     sync = Q80 if args.sync == "q80" else F32
     m = mf.ModelFile(path, sync_type=sync)
