@@ -186,3 +186,29 @@ def test_sampler_torch_path_matches_numpy():
         assert picks1 == picks2, (topp, picks1[:5], picks2[:5])
         # the dominant trio carries ~98% of the mass
         assert sum(p in (7, 21, 42) for p in picks1) >= 90
+
+
+def test_tokenizer_writer_byte_exact_vs_reference_converter(tmp_path):
+    """Our .t writer must produce byte-identical files to the reference's
+    own converter (converter/tokenizer-writer.py) for the same vocab."""
+    import importlib.util
+    import io
+    import os
+
+    path = "/root/reference/converter/tokenizer-writer.py"
+    if not os.path.exists(path):
+        pytest.skip("reference converter not available")
+    spec = importlib.util.spec_from_file_location("ref_tok_writer", path)
+    ref = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(ref)
+
+    vocab = [b"a", b"b", b"ab", b"<|eot|>"]
+    scores = [0.0, 0.5, 1.0, 0.0]
+    template = b"{{x [INST] y}}"
+    buf = io.BytesIO()
+    ref.writeTokenizer(buf, vocab, scores, template, 3, True, [3])
+
+    ours = str(tmp_path / "t.t")
+    tok.write_tokenizer(ours, vocab, scores, 3, True, [3],
+                        template.decode())
+    assert open(ours, "rb").read() == buf.getvalue()
