@@ -248,6 +248,7 @@ def test_convert_llama_pth(tmp_path):
 def test_convert_tokenizer_hf(tmp_path):
     """Fabricated HF byte-level BPE tokenizer.json -> .t -> encode parity
     with the HF fast tokenizer."""
+    pytest.importorskip("transformers")
     import sys, os
     sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(
         os.path.abspath(__file__))), "converter"))
