@@ -17,8 +17,12 @@ def test_q80_roundtrip_bounded_error(vals):
     y = quants.q80_roundtrip(x)
     blocks = x.reshape(-1, 32)
     amax = np.abs(blocks).max(axis=1, keepdims=True)
-    # per-element error bounded by half a quantization step per block
-    assert np.all(np.abs(y.reshape(-1, 32) - blocks) <= amax * (0.51 / 127.0 + 2**-10) + 1e-6)
+    # per-element error: half a quantization step + f16 scale rounding.
+    # The absolute floor covers f16-SUBNORMAL scales (amax < ~7.8e-3 gives
+    # d = amax/127 < 6.1e-5): subnormal rounding is up to 2^-25 absolute,
+    # amplified by |q| <= 127 (found by hypothesis with a 1e-5 block).
+    bound = amax * (0.51 / 127.0 + 2**-10) + 127 * 2**-25 + 1e-6
+    assert np.all(np.abs(y.reshape(-1, 32) - blocks) <= bound)
 
 
 @settings(max_examples=50, deadline=None)
