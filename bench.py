@@ -64,12 +64,13 @@ def _build_cpu_smoke(args, world, comm):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=64)
-    ap.add_argument("--warmup", type=int, default=16)
+    ap.add_argument("--steps", type=int, default=200)
+    ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--model", default="llama-3.1-8b")
     ap.add_argument("--seq-len", type=int, default=4096)
-    ap.add_argument("--prefill", type=int, default=32,
-                    help="prompt tokens evaluated before the timed decode")
+    ap.add_argument("--prefill", "--ctx", type=int, default=32, dest="prefill",
+                    help="prompt tokens evaluated before the timed decode "
+                         "(--ctx 1024 measures long-context decode)")
     ap.add_argument("--no-graph", action="store_true")
     ap.add_argument("--sync", choices=["q80", "f32"], default="q80")
     # hidden: --device cpu runs a tiny model through the SAME distributed

@@ -56,6 +56,11 @@ class ApiState:
                                               self.tok.chat_template, eos_piece)
         self.cache = NaiveCache()
         self.model_name = "dllama"
+        # per-request sampler defaults (reference re-parses params with CLI
+        # defaults each request, dllama-api.cpp:491-520)
+        self.default_temp = args.temperature
+        self.default_topp = args.topp
+        self.default_seed = args.seed if args.seed is not None else int(time.time())
 
     def complete(self, body: dict, emit):
         """Run one chat completion; emit(delta_text) streams chunks."""
@@ -66,8 +71,15 @@ class ApiState:
         start = self.cache.resolve(tokens)
         self.engine.reset(start)
         max_tokens = int(body.get("max_tokens") or 256)
+        # reset to CLI defaults so one request's overrides don't leak into the
+        # next (reference dllama-api.cpp:491-520)
+        self.engine.sampler.set_temp(self.default_temp)
+        self.engine.sampler.topp = self.default_topp
+        self.engine.sampler.set_seed(self.default_seed)
         if body.get("temperature") is not None:
             self.engine.sampler.set_temp(float(body["temperature"]))
+        if body.get("top_p") is not None:
+            self.engine.sampler.topp = float(body["top_p"])
         if body.get("seed") is not None:
             self.engine.sampler.set_seed(int(body["seed"]))
         user_stop = body.get("stop") or []
@@ -92,7 +104,10 @@ class ApiState:
         gen_tokens, _ = self.engine.generate(
             tokens[start:], max_tokens, on_token=on_token,
             stop_check=lambda t: detector.is_eos(t) or detector.eos_pos >= 0)
-        self.cache.update(tokens + gen_tokens)
+        # cache only EVALUATED tokens: the last sampled token was never fed
+        # through the model, so its KV row does not exist (reference caches
+        # the evaluated endPos, dllama-api.cpp:470-473)
+        self.cache.update(tokens + gen_tokens[:-1])
         return "".join(out_text), len(tokens), len(gen_tokens)
 
 

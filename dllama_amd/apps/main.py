@@ -104,7 +104,8 @@ def load_engine(args):
         comm.broadcast_(st, src=0)
         seed = int(st.item())
     sampler = Sampler(m.header.vocab_size, args.temperature, args.topp, seed)
-    return InferenceEngine(model, tok, sampler, n_batches=args.n_batches), m, comm
+    nb = getattr(model, "n_batches", args.n_batches)  # HIP backend may round up
+    return InferenceEngine(model, tok, sampler, n_batches=nb), m, comm
 
 
 def _print_header(h, cfg, use_gpu):

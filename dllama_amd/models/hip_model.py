@@ -78,7 +78,16 @@ class HipTransformer:
         self.comm = comm or SingleComm()
         self.device = torch.device(device or "cuda")
         self.k = hip_ops()
-        self.n_batches = n_batches
+        # buffers are indexed with NB=_pow2_batch(B) rows and the prefill
+        # GEMM pads to 32 rows: a non-pow2 or <32 n_batches would launch
+        # grids over short buffers (OOB). Round up instead of trusting it.
+        nb = max(32, _pow2_batch(n_batches))
+        if nb != n_batches:
+            import warnings
+            warnings.warn(
+                f"HIP backend rounds --n-batches {n_batches} up to {nb} "
+                "(power of two >= 32 required by buffer/GEMM layout)")
+        self.n_batches = nb
         self.layers: list[dict] = []
         self.embedding = None
         self.final_norm = None

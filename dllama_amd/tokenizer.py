@@ -286,17 +286,24 @@ class Sampler:
         coin = self._random_f32()
         if self.topp <= 0 or self.topp >= 1:
             cdf = torch.cumsum(p, dim=-1)
-            return int(torch.searchsorted(cdf, torch.tensor(coin, device=p.device))
+            return int(torch.searchsorted(cdf, torch.tensor(coin, device=p.device),
+                                          right=True)
                        .clamp(0, self.vocab_size - 1).item())
-        probs, order = torch.sort(p, descending=True)
+        # mirror the numpy path exactly (cutoff pre-filter, stable sort,
+        # right-bisect) so GPU and CPU ranks stay in sampling lockstep
+        cutoff = (1.0 - self.topp) / (self.vocab_size - 1)
+        idx = torch.nonzero(p >= cutoff).reshape(-1)
+        order = idx[torch.argsort(-p[idx], stable=True)]
+        probs = p[order]
         c = torch.cumsum(probs, dim=-1)
-        last = int(torch.searchsorted(c, torch.tensor(self.topp, device=p.device))
-                   .item())
-        last = min(last, probs.numel() - 1)
+        last = int(torch.searchsorted(c, torch.tensor(self.topp, device=p.device),
+                                      right=True).item())
+        last = min(last, order.numel() - 1)
         r = coin * float(c[last].item())
         pick = int(torch.searchsorted(c[: last + 1],
-                                      torch.tensor(r, device=p.device)).item())
-        return int(order[min(pick, last)].item())
+                                      torch.tensor(r, device=p.device),
+                                      right=True).clamp(0, last).item())
+        return int(order[pick].item())
 
     def sample(self, logits) -> int:
         import torch
