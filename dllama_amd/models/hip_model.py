@@ -230,11 +230,13 @@ class HipTransformer:
         self.amax_scratch = torch.zeros(self.amax_blocks, dtype=torch.int64, device=dev)
         import os as _os
         self.attn_splits = int(_os.environ.get("DLLAMA_ATTN_SPLITS", "8"))
-        self.fused_sync = _os.environ.get("DLLAMA_FUSED_SYNC", "0") == "1"
-        # round-2 experimental: past this decode position, recapture the
-        # graph with S=16 splits (S=16 measured better at ~1.1k ctx);
-        # 0 = disabled
-        self.adaptive_thresh = int(_os.environ.get("DLLAMA_ADAPTIVE_SPLITS", "0"))
+        # fused quantize-into-wire sync: validated bit-identical on hardware
+        # (test_sync_quant_pack_matches_two_kernel_path); default on
+        self.fused_sync = _os.environ.get("DLLAMA_FUSED_SYNC", "1") == "1"
+        # past this decode position, recapture the graph with S=16 K-splits
+        # (measured: S=16 wins from ~pos 512 up, tools/attn_kv16_probe);
+        # 0 disables
+        self.adaptive_thresh = int(_os.environ.get("DLLAMA_ADAPTIVE_SPLITS", "512"))
         # S=8 measured best at decode (16/32: combine reads S partials)
         self.attn_ml = torch.zeros(NB * c.n_heads0 * self.attn_splits * 2, device=dev)
         self.attn_o = torch.zeros(NB * c.n_heads0 * self.attn_splits * c.head_dim,

@@ -1752,7 +1752,7 @@ void q40_gemv_nq_rope(torch::Tensor qs, torch::Tensor scales, torch::Tensor x,
 void q40_gemm(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
               torch::Tensor xs, torch::Tensor y, int64_t batch,
               c10::optional<torch::Tensor> part = c10::nullopt,
-              bool use_v2 = false) {
+              int64_t variant = -1) {
     // int8-MFMA batched matmul (prefill path); xq/xs must have >=32 rows
     CHECK_CUDA(qs); CHECK_CONT(qs); CHECK_CONT(xq);
     const int d = qs.size(0);
@@ -1768,9 +1768,12 @@ void q40_gemm(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
         while (ksplit > 1 && (int64_t)ksplit * 32 * d > part->numel()) ksplit--;
     }
     float *pp = ksplit > 1 ? part->data_ptr<float>() : nullptr;
-    static const bool env_v2 =
-        std::getenv("DLLAMA_GEMM_V2") && atoi(std::getenv("DLLAMA_GEMM_V2")) == 1;
-    auto *kern = (use_v2 || env_v2) ? k_q40_gemm_v2 : k_q40_gemm;
+    // v2 (LDS-staged) is the default since round 2 (validated on hardware:
+    // ~25% faster at all three prefill shapes); DLLAMA_GEMM_V2=0 reverts
+    static const bool env_v2 = !std::getenv("DLLAMA_GEMM_V2")
+        || atoi(std::getenv("DLLAMA_GEMM_V2")) != 0;
+    const bool v2 = variant < 0 ? env_v2 : variant == 1;
+    auto *kern = v2 ? k_q40_gemm_v2 : k_q40_gemm;
     hipLaunchKernelGGL(kern, dim3(mtiles, ksplit), dim3(256), 0,
                        cur_stream(), qs.data_ptr<uint8_t>(),
                        reinterpret_cast<const __half *>(scales.data_ptr<at::Half>()),
@@ -1791,18 +1794,20 @@ int64_t q40_gemv_argmax_blocks(int64_t d) {
 
 void q40_gemv_grouped(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
                       torch::Tensor xs, torch::Tensor xbs, torch::Tensor expert_idx,
-                      torch::Tensor y, int64_t k_slots, bool use_v2 = false) {
+                      torch::Tensor y, int64_t k_slots, int64_t variant = -1) {
     CHECK_CUDA(qs); CHECK_CONT(qs);
     const int d = qs.size(1);
     const int n = qs.size(2) * 2;
     const int n_slots = expert_idx.numel();
     const int waves_per_block = 4;
-    static const bool env_v2 =
-        std::getenv("DLLAMA_MOE_V2") && atoi(std::getenv("DLLAMA_MOE_V2")) == 1;
+    // lane-tiled (v2) default since round 2 (validated: 1.4-1.8x at
+    // Qwen3-30B shapes); DLLAMA_MOE_V2=0 reverts to the 64-lane layout
+    static const bool env_v2 = !std::getenv("DLLAMA_MOE_V2")
+        || atoi(std::getenv("DLLAMA_MOE_V2")) != 0;
     // v2: LPP = smallest power of two >= nbp (clamped [8, 64]) so every
     // lane has work; rows per wave scale up by 64/LPP
     int lpp = 64;
-    if (use_v2 || env_v2) {
+    if (variant < 0 ? env_v2 : variant == 1) {
         const int nbp = (n / QB) >> 1;
         lpp = 8;
         while (lpp < nbp && lpp < 64) lpp <<= 1;
@@ -2228,7 +2233,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("q40_gemv_resid", &q40_gemv_resid);
     m.def("q40_gemm", &q40_gemm, py::arg("qs"), py::arg("scales"),
           py::arg("xq"), py::arg("xs"), py::arg("y"), py::arg("batch"),
-          py::arg("part") = py::none(), py::arg("use_v2") = false);
+          py::arg("part") = py::none(), py::arg("variant") = -1);
     m.def("q40_gemv_rope", &q40_gemv_rope);
     m.def("q40_gemv_nq", &q40_gemv_nq, py::arg("qs"), py::arg("scales"),
           py::arg("x"), py::arg("wnorm"), py::arg("ssq"), py::arg("eps"),
@@ -2245,7 +2250,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("q40_gemv_grouped", &q40_gemv_grouped, py::arg("qs"),
           py::arg("scales"), py::arg("xq"), py::arg("xs"), py::arg("xbs"),
           py::arg("expert_idx"), py::arg("y"), py::arg("k_slots"),
-          py::arg("use_v2") = false);
+          py::arg("variant") = -1);
     m.def("rope", &rope);
     m.def("rope_kv", &rope_kv);
     m.def("kv_append", &kv_append);

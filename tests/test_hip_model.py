@@ -114,11 +114,9 @@ def test_hip_prefill_gemm_path(tiny_path):
     assert _rel_err(batch, want) < 0.02
 
 
-@pytest.mark.skipif(not __import__("os").environ.get("DLLAMA_EXPERIMENTAL"),
-                    reason="round-2 experimental path (DLLAMA_EXPERIMENTAL=1)")
 def test_adaptive_splits_recapture(tiny_path):
-    """DLLAMA_ADAPTIVE_SPLITS: crossing the position threshold re-captures
-    the decode graph with S=16; logits must keep matching eager decode."""
+    """DLLAMA_ADAPTIVE_SPLITS (default 512): crossing the position threshold
+    re-captures the decode graph with S=16; logits must keep matching eager."""
     from dllama_amd.models.hip_model import HipTransformer
     m = mf.ModelFile(tiny_path)
     cfg = ModelConfig.from_header(m.header)

@@ -424,11 +424,9 @@ def test_sync_quant_pack_matches_two_kernel_path(k):
     assert torch.equal(got, want)
 
 
-@pytest.mark.skipif(not __import__("os").environ.get("DLLAMA_EXPERIMENTAL"),
-                    reason="round-2 experimental kernel (DLLAMA_EXPERIMENTAL=1)")
 def test_q40_gemm_v2_matches_v1(k):
-    """LDS-staged GEMM v2 (use_v2=True / DLLAMA_GEMM_V2) vs the production
-    kernel, with and without K-split partials, including row masking."""
+    """LDS-staged GEMM v2 (the default since round 2) vs the v1 kernel,
+    with and without K-split partials, including row masking."""
     for d, n, B in ((160, 1024, 20), (512, 2048, 32), (256, 4096, 9)):
         qs, sc, _ = _mk_linear(d, n, 60 + d % 7)
         x = rand(32, n, seed=61, scale=0.5)
@@ -441,18 +439,16 @@ def test_q40_gemm_v2_matches_v1(k):
             y1 = torch.zeros(32, d, device=DEV)
             y2 = torch.zeros(32, d, device=DEV)
             pa = part if use_part else None
-            k.q40_gemm(qs, sc, q, s, y1, B, pa)
-            k.q40_gemm(qs, sc, q, s, y2, B, pa, use_v2=True)
+            k.q40_gemm(qs, sc, q, s, y1, B, pa, variant=0)
+            k.q40_gemm(qs, sc, q, s, y2, B, pa, variant=1)
             assert torch.allclose(y2[:B], y1[:B], atol=1e-3, rtol=1e-4), \
                 (d, n, B, use_part,
                  (y2[:B] - y1[:B]).abs().max().item())
 
 
-@pytest.mark.skipif(not __import__("os").environ.get("DLLAMA_EXPERIMENTAL"),
-                    reason="round-2 experimental kernel (DLLAMA_EXPERIMENTAL=1)")
 def test_q40_gemv_grouped_v2_matches_v1(k):
-    """Lane-tiled grouped GEMV (use_v2 / DLLAMA_MOE_V2) vs production at
-    MoE-like shapes, including tiny nbp (w2-shape n=768 -> nbp=12)."""
+    """Lane-tiled grouped GEMV (the default since round 2) vs the 64-lane
+    v1 at MoE-like shapes, including tiny nbp (w2-shape n=768 -> nbp=12)."""
     for d, n in ((128, 512), (96, 768), (160, 2048)):
         E, B, ka = 4, 2, 2
         lins = [_mk_linear(d, n, 70 + e) for e in range(E)]
@@ -466,7 +462,7 @@ def test_q40_gemv_grouped_v2_matches_v1(k):
         idx = torch.tensor([1, 3, 0, 2], dtype=torch.int32, device=DEV)
         y1 = torch.zeros(B * ka, d, device=DEV)
         y2 = torch.zeros(B * ka, d, device=DEV)
-        k.q40_gemv_grouped(qs, sc, q, s, bs, idx, y1, ka)
-        k.q40_gemv_grouped(qs, sc, q, s, bs, idx, y2, ka, use_v2=True)
+        k.q40_gemv_grouped(qs, sc, q, s, bs, idx, y1, ka, variant=0)
+        k.q40_gemv_grouped(qs, sc, q, s, bs, idx, y2, ka, variant=1)
         assert torch.allclose(y2, y1, atol=1e-4, rtol=1e-5), \
             (d, n, (y2 - y1).abs().max().item())
