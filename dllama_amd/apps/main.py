@@ -87,8 +87,16 @@ def load_engine(args):
             else int(os.environ.get("LOCAL_RANK", "0"))
         model = HipTransformer.from_file(m, cfg, device=f"cuda:{dev}", comm=comm,
                                          n_batches=args.n_batches)
-        if not args.no_graph and comm.world == 1:
-            model.capture_decode_graph()
+        if not args.no_graph:
+            # RCCL collectives are hipGraph-capturable, so TP decode is one
+            # graph replay per token too (every rank captures in lockstep —
+            # the warmup forwards inside contain collectives)
+            try:
+                model.capture_decode_graph()
+            except Exception as e:  # noqa: BLE001
+                if comm.rank == 0:
+                    print(f"⚠️  decode graph capture failed ({e}); "
+                          "running eager (slower)", file=sys.stderr)
     else:
         from ..models.cpu_model import CpuTransformer
         torch.set_num_threads(max(1, args.nthreads))

@@ -73,9 +73,13 @@ def _pow2_batch(b: int) -> int:
 
 class HipTransformer:
     def __init__(self, config: ModelConfig, device=None, comm: Comm | None = None,
-                 n_batches: int = 32):
+                 n_batches: int = 32, force_sync: bool = False):
         self.cfg = config
         self.comm = comm or SingleComm()
+        # force_sync: run the full TP sync/gather/concat path at world=1
+        # (SingleComm collectives are identity) — lets a 1-GPU box validate
+        # every TP kernel and the graph-captured sync step end to end
+        self.tp_path = config.world > 1 or force_sync
         self.device = torch.device(device or "cuda")
         self.k = hip_ops()
         # buffers are indexed with NB=_pow2_batch(B) rows and the prefill
@@ -104,14 +108,15 @@ class HipTransformer:
 
     @classmethod
     def from_file(cls, m: ModelFile, config: ModelConfig, device=None,
-                  comm: Comm | None = None, n_batches: int = 32) -> "HipTransformer":
+                  comm: Comm | None = None, n_batches: int = 32,
+                  force_sync: bool = False) -> "HipTransformer":
         from ..quants import Q40
         if m.header.weight_type != Q40:
             raise ValueError(
                 "the MI355X HIP backend runs Q40 weights (the reference's "
                 "shipped format); f32/q80 .m files run on the CPU backend "
                 "(--gpu-index -1) or can be re-quantized with convert_hf.py")
-        self = cls(config, device, comm, n_batches)
+        self = cls(config, device, comm, n_batches, force_sync)
         c, dev = config, self.device
         r, w = c.rank, c.world
 
@@ -168,10 +173,11 @@ class HipTransformer:
 
     @classmethod
     def synthetic(cls, config: ModelConfig, device=None, comm: Comm | None = None,
-                  n_batches: int = 32, seed: int = 1234) -> "HipTransformer":
+                  n_batches: int = 32, seed: int = 1234,
+                  force_sync: bool = False) -> "HipTransformer":
         """Random-init weights built directly on device (benches: no network
         for checkpoints, and an 8B .m file round-trip is pointless there)."""
-        self = cls(config, device, comm, n_batches)
+        self = cls(config, device, comm, n_batches, force_sync)
         c, dev = config, self.device
         gen = torch.Generator(device=dev)
         gen.manual_seed(seed + c.rank)
@@ -242,12 +248,15 @@ class HipTransformer:
         self.attn_o = torch.zeros(NB * c.n_heads0 * self.attn_splits * c.head_dim,
                                   device=dev)
         self.attn_counter = torch.zeros(NB * c.n_heads0, dtype=torch.int32, device=dev)
-        if c.world > 1:
+        if self.tp_path:
             # per-batch-size contiguous gather buffers: collectives need a
             # flat contiguous output (world, nb*...) — a [:, :nb] slice is not
             nbs = [n for n in (1, 2, 4, 8, 16, 32) if n <= NB]
             self.logits_gather = {n: torch.zeros(c.world, n, c.vocab0, device=dev)
                                   for n in nbs}
+            self.logits_full = torch.zeros(NB, c.world * c.vocab0, device=dev)
+            self.argmax_scratch_full = torch.zeros(
+                -(-c.world * c.vocab0 // 4096), dtype=torch.int64, device=dev)
             if c.sync_type == Q80:
                 row_bytes = c.dim + 2 * (c.dim // QB)
                 self.sync_out = torch.zeros(NB * row_bytes, dtype=torch.uint8, device=dev)
@@ -313,12 +322,12 @@ class HipTransformer:
         into the GEMV epilogue (+1us with cacheline-strided ssq slots — the
         earlier +6.5us was atomics serializing on one cacheline)."""
         k = self.k
-        if self.cfg.world == 1 and NB < 8:
+        if not self.tp_path and NB < 8:
             k.q40_gemv_resid(lin.qs, lin.scales, qb.q, qb.s, qb.bs,
                              self.x, self.ssq[slot], NB)
             return
         self._mm(lin, qb, self.partial, NB)
-        if self.cfg.world == 1:
+        if not self.tp_path:
             k.add_ssq(self.x[:NB], self.partial[:NB], self.ssq[slot], NB)
         else:
             self._sync_partial(NB, slot)
@@ -403,10 +412,10 @@ class HipTransformer:
 
         if self.skip_logits and B > 1:
             return
-        use_amax = (self.greedy_feedback and B == 1 and c.world == 1)
+        use_amax = (self.greedy_feedback and B == 1 and not self.tp_path)
         norm_gemv(self.wcls, self.final_norm, slot, self.logits0,
                   self.amax_scratch if use_amax else None)
-        if c.world > 1:
+        if self.tp_path:
             self.comm.all_gather(self.logits_gather[NB], self.logits0[:NB])
         if self.greedy_feedback and B == 1:
             # on-device greedy sampling feeding the next decode step (used by
@@ -414,8 +423,11 @@ class HipTransformer:
             if use_amax:
                 k.token_from_argmax(self.tokens, self.amax_scratch, self.amax_blocks)
             else:
-                full = self.logits_gather[1][:, 0].reshape(-1)
-                self.tokens[0].copy_(torch.argmax(full))
+                # TP: gathered slices are contiguous in global vocab order
+                # (row-split wcls), so the flat gather buffer IS the full
+                # logits row; two-kernel argmax, no ATen in the graph
+                k.argmax_token(self.tokens, self.logits_gather[1].view(-1),
+                               self.argmax_scratch_full)
 
     def _moe_ffn(self, B: int, NB: int, lw: dict, slot: int):
         """Router + grouped expert GEMVs (reference llm.cpp:450-487);
@@ -434,13 +446,11 @@ class HipTransformer:
         k.q40_gemv_grouped(lw["w2"].qs, lw["w2"].scales, self.moe_dq.q,
                            self.moe_dq.s, self.moe_dq.bs, self.moe_idx[:S],
                            self.moe_y, 1)
-        if c.world == 1:
+        if not self.tp_path:
             k.scale_merge_add(self.x[:NB], self.moe_y, self.moe_wts,
                               self.ssq[slot], NB, ka)
         else:
-            y = self.moe_y[:S].reshape(NB, ka, c.dim)
-            torch.sum(y * self.moe_wts[:NB].unsqueeze(-1), dim=1,
-                      out=self.partial[:NB])
+            k.scale_merge(self.partial[:NB], self.moe_y, self.moe_wts, NB, ka)
             self._sync_partial(NB, slot)
 
     # ------------------------------------------------------------ engine API
@@ -483,10 +493,10 @@ class HipTransformer:
             self._graph_pos = None
             self.forward_buffers(B)
         c = self.cfg
-        if c.world > 1:
+        if self.tp_path:
             NBp = _pow2_batch(B)
-            return (self.logits_gather[NBp][:, :B].permute(1, 0, 2)
-                    .reshape(B, c.vocab_size))
+            self.k.logits_concat(self.logits_full, self.logits_gather[NBp], B)
+            return self.logits_full[:B]
         return self.logits0[:B]
 
     # ------------------------------------------------------------ graphs

@@ -1276,6 +1276,65 @@ __global__ void k_scale_merge_add(float *__restrict__ x,
     }
 }
 
+// TP variant of the above: weighted expert sum into the partial buffer
+// (no residual fold, no ssq — those happen after the Q80 sync merge-add;
+// replaces the eager torch.sum fallback so TP MoE decode stays
+// graph-capturable with zero ATen in the step)
+__global__ void k_scale_merge(float *__restrict__ partial,
+                              const float *__restrict__ y,
+                              const float *__restrict__ wts,
+                              int n, int topk) {
+    const int b = blockIdx.y;
+    for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * blockDim.x) {
+        float acc = 0.0f;
+        for (int s = 0; s < topk; s++)
+            acc = fmaf(wts[(int64_t)b * topk + s],
+                       y[((int64_t)b * topk + s) * n + i], acc);
+        partial[(int64_t)b * n + i] = acc;
+    }
+}
+
+// TP logits assembly: the row-split wcls puts rank w's slice at global vocab
+// offset w*vocab0, so full logits for batch row b are the concatenation of
+// gather[w, b, :] over w (reference gathers the same slices to root,
+// nn-network.cpp:568-600 with onlyFromWorkerToRoot). One strided copy kernel
+// instead of a per-token ATen permute+reshape.
+__global__ void k_logits_concat(float *__restrict__ dst,
+                                const float *__restrict__ src,
+                                int nb, int vocab0, int world) {
+    const int b = blockIdx.y;
+    const int64_t total = (int64_t)world * vocab0;
+    for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        const int w = i / vocab0, j = i - (int64_t)w * vocab0;
+        dst[(int64_t)b * total + i] =
+            src[((int64_t)w * nb + b) * vocab0 + j];
+    }
+}
+
+// generic argmax stage 1 over a flat f32 array (the TP greedy path: argmax
+// of the gathered full logits). Each block writes its packed best to
+// scratch[blockIdx]; k_token_from_argmax reduces scratch.
+__global__ void k_argmax_stage1(const float *__restrict__ x, int64_t n,
+                                unsigned long long *__restrict__ scratch) {
+    unsigned long long best = 0ull;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x)
+        best = max(best, argmax_pack(x[i], (int)i));
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        best = max(best, (unsigned long long)__shfl_xor((long long)best, off, WAVE));
+    __shared__ unsigned long long red[16];
+    const int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    if (lane == 0) red[wid] = best;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        for (int i = 1; i < blockDim.x / WAVE; i++) best = max(best, red[i]);
+        scratch[blockIdx.x] = best;
+    }
+}
+
 // ------------------------------------------------------------------ rope
 // style 0 = llama interleaved pairs (reference ropeLlama_F32,
 // nn-cpu-ops.cpp:843-863), style 1 = falcon/neox half-rotated
@@ -1941,6 +2000,41 @@ void scale_merge_add(torch::Tensor x, torch::Tensor y, torch::Tensor wts,
                        wts.data_ptr<float>(), ssq.data_ptr<float>(), n, (int)topk);
 }
 
+void scale_merge(torch::Tensor partial, torch::Tensor y, torch::Tensor wts,
+                 int64_t batch, int64_t topk) {
+    CHECK_CUDA(partial);
+    const int n = partial.size(-1);
+    hipLaunchKernelGGL(k_scale_merge, dim3(ceil_div(n, 256), batch), dim3(256),
+                       0, cur_stream(), partial.data_ptr<float>(),
+                       y.data_ptr<float>(), wts.data_ptr<float>(), n, (int)topk);
+}
+
+void logits_concat(torch::Tensor dst, torch::Tensor src, int64_t batch) {
+    // src [world, nb, vocab0] (contiguous all-gather output) -> dst rows
+    // 0..batch-1 of [.., world*vocab0]
+    CHECK_CUDA(dst); CHECK_CONT(src);
+    const int world = src.size(0);
+    const int nb = src.size(1);
+    const int vocab0 = src.size(2);
+    const int64_t total = (int64_t)world * vocab0;
+    hipLaunchKernelGGL(k_logits_concat,
+                       dim3(std::min<int64_t>(ceil_div(total, 256), 2048), batch),
+                       dim3(256), 0, cur_stream(), dst.data_ptr<float>(),
+                       src.data_ptr<float>(), nb, vocab0, world);
+}
+
+void argmax_token(torch::Tensor token, torch::Tensor x, torch::Tensor scratch) {
+    // two-stage argmax of flat f32 x into token[0] (TP greedy decode)
+    CHECK_CUDA(x); CHECK_CONT(x);
+    const int64_t n = x.numel();
+    const int blocks = (int)std::min<int64_t>(scratch.numel(), ceil_div(n, 4096));
+    auto *sp = reinterpret_cast<unsigned long long *>(scratch.data_ptr<int64_t>());
+    hipLaunchKernelGGL(k_argmax_stage1, dim3(blocks), dim3(256), 0, cur_stream(),
+                       x.data_ptr<float>(), n, sp);
+    hipLaunchKernelGGL(k_token_from_argmax, dim3(1), dim3(1024), 0, cur_stream(),
+                       (long *)token.data_ptr<int64_t>(), sp, blocks);
+}
+
 void router_gemv(torch::Tensor gate, torch::Tensor t, torch::Tensor logits,
                  int64_t batch) {
     CHECK_CUDA(gate);
@@ -2244,6 +2338,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("eps"), py::arg("yout") = py::none());
     m.def("moe_gate", &moe_gate);
     m.def("scale_merge_add", &scale_merge_add);
+    m.def("scale_merge", &scale_merge);
+    m.def("logits_concat", &logits_concat);
+    m.def("argmax_token", &argmax_token);
     m.def("router_gemv", &router_gemv);
     m.def("norm_f32", &norm_f32);
     m.def("add_ssq", &add_ssq);
