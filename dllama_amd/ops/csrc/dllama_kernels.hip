@@ -383,6 +383,14 @@ __global__ void k_add_ssq(float *__restrict__ x,
     }
 }
 
+// KV-cache dtype helpers (f16 KV is the default; torch extensions build
+// with __HIP_NO_HALF_CONVERSIONS__, so conversions must be explicit)
+__device__ __forceinline__ float kv_f(float v) { return v; }
+__device__ __forceinline__ float kv_f(__half v) { return __half2float(v); }
+template <typename T> __device__ __forceinline__ T kv_c(float v);
+template <> __device__ __forceinline__ float kv_c<float>(float v) { return v; }
+template <> __device__ __forceinline__ __half kv_c<__half>(float v) { return __float2half(v); }
+
 // argmax pack: monotonic unsigned ordering of (float value, smallest index
 // wins ties) for a single global atomicMax — on-device greedy sampling.
 __device__ __forceinline__ unsigned long long argmax_pack(float v, int idx) {
@@ -515,7 +523,7 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
                            const float *__restrict__ xf,
                            const float *__restrict__ wnorm,
                            const float *__restrict__ ssq_in,
-                           float eps) {
+                           float eps, int kv_f16) {
     const int wpb = blockDim.x / WAVE;
     const int wid = threadIdx.x / WAVE;
     const int row0 = (blockIdx.x * wpb + wid) * RPW;
@@ -633,12 +641,29 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
                 const int j = (rk % hd) >> 1;
                 const float cr = rope_cache[(int64_t)pb * hd + 2 * j];
                 const float ci = rope_cache[(int64_t)pb * hd + 2 * j + 1];
-                kc[(int64_t)pb * kv_dim0 + rk] = v[0] * cr - v[1 % RPW] * ci;
-                kc[(int64_t)pb * kv_dim0 + rk + 1] = v[0] * ci + v[1 % RPW] * cr;
+                const float o0 = v[0] * cr - v[1 % RPW] * ci;
+                const float o1 = v[0] * ci + v[1 % RPW] * cr;
+                // runtime KV-dtype branch: lane 0 only, once per wave —
+                // free next to the weight stream (f16 KV is the default)
+                if (kv_f16) {
+                    __half *kh = reinterpret_cast<__half *>(kc);
+                    kh[(int64_t)pb * kv_dim0 + rk] = __float2half(o0);
+                    kh[(int64_t)pb * kv_dim0 + rk + 1] = __float2half(o1);
+                } else {
+                    kc[(int64_t)pb * kv_dim0 + rk] = o0;
+                    kc[(int64_t)pb * kv_dim0 + rk + 1] = o1;
+                }
             } else {
                 const int rv = rbase - q_dim0 - kv_dim0;
-                vc[(int64_t)pb * kv_dim0 + rv] = v[0];
-                if (RPW == 2) vc[(int64_t)pb * kv_dim0 + rv + 1] = v[1 % RPW];
+                if (kv_f16) {
+                    __half *vh = reinterpret_cast<__half *>(vc);
+                    vh[(int64_t)pb * kv_dim0 + rv] = __float2half(v[0]);
+                    if (RPW == 2)
+                        vh[(int64_t)pb * kv_dim0 + rv + 1] = __float2half(v[1 % RPW]);
+                } else {
+                    vc[(int64_t)pb * kv_dim0 + rv] = v[0];
+                    if (RPW == 2) vc[(int64_t)pb * kv_dim0 + rv + 1] = v[1 % RPW];
+                }
             }
         } else {
             #pragma unroll
@@ -685,10 +710,10 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
 // Stage 1: grid (H0, B, S); split sp covers t = (sp*4+wave) + 4*S*i — the
 // t-range is spread over S*4 waves so long contexts fill the chip.
 // Per-split (m, l, o) goes to scratch; stage 2 combines the S partials.
-template <int VEC, bool QUANT>
+template <int VEC, typename KVT>
 __global__ void k_attn_split(const float *__restrict__ q, int q_ld,
-                             const float *__restrict__ kc,
-                             const float *__restrict__ vc,
+                             const KVT *__restrict__ kc,
+                             const KVT *__restrict__ vc,
                              const int *__restrict__ pos,
                              int n_heads0, int kv_mul, int kv_dim0, float scale,
                              float *__restrict__ ml_scratch,
@@ -729,9 +754,10 @@ __global__ void k_attn_split(const float *__restrict__ q, int q_ld,
         const int tg = tb + group;  // this group's timestep
         float partial = 0.0f;
         if (tg < plen) {
-            const float *krow = kc + (int64_t)tg * kv_dim0 + kv_off + lane16 * VEC16;
+            const KVT *krow = kc + (int64_t)tg * kv_dim0 + kv_off + lane16 * VEC16;
             #pragma unroll
-            for (int v = 0; v < VEC16; v++) partial = fmaf(qreg[v], krow[v], partial);
+            for (int v = 0; v < VEC16; v++)
+                partial = fmaf(qreg[v], kv_f(krow[v]), partial);
         }
         const float sg = group16_reduce_sum(partial);
         float s4[4];
@@ -751,9 +777,9 @@ __global__ void k_attn_split(const float *__restrict__ q, int q_ld,
         #pragma unroll
         for (int gg = 0; gg < 4; gg++) {
             if (tb + gg >= plen) break;
-            const float *vrow = vc + (int64_t)(tb + gg) * kv_dim0 + kv_off + lane * VEC;
+            const KVT *vrow = vc + (int64_t)(tb + gg) * kv_dim0 + kv_off + lane * VEC;
             #pragma unroll
-            for (int v = 0; v < VEC; v++) o[v] = fmaf(w4[gg], vrow[v], o[v]);
+            for (int v = 0; v < VEC; v++) o[v] = fmaf(w4[gg], kv_f(vrow[v]), o[v]);
         }
         m = mn;
     }
@@ -1377,13 +1403,13 @@ __global__ void k_rope(float *__restrict__ x,
 // the cache row pos+b, and copies v to the cache — one launch replacing
 // rope(q), rope(k), kv_append (reference runs 4 separate ops here,
 // llm.cpp:300-330).
-template <int STYLE>
+template <int STYLE, typename KVT>
 __global__ void k_rope_kv(float *__restrict__ qkv, int ld,
                           int q_dim0, int kv_dim0,
                           const float *__restrict__ cache,
                           const int *__restrict__ pos,
-                          float *__restrict__ kc,
-                          float *__restrict__ vc,
+                          KVT *__restrict__ kc,
+                          KVT *__restrict__ vc,
                           int hd) {
     const int b = blockIdx.y;
     const int p = pos[0] + b;
@@ -1413,13 +1439,15 @@ __global__ void k_rope_kv(float *__restrict__ qkv, int ld,
             const float o1 = v0 * ci + v1 * cr;
             if (is_q) { src[i0] = o0; src[i1] = o1; }
             else {
-                float *dst = kc + (int64_t)p * kv_dim0;
-                dst[i0] = o0; dst[i1] = o1;
+                KVT *dst = kc + (int64_t)p * kv_dim0;
+                dst[i0] = kv_c<KVT>(o0); dst[i1] = kv_c<KVT>(o1);
             }
         } else {
             const int j = (i - qp - kp) * 4;
-            *reinterpret_cast<float4 *>(vc + (int64_t)p * kv_dim0 + j) =
-                *reinterpret_cast<const float4 *>(vrow + j);
+            const float4 vv = *reinterpret_cast<const float4 *>(vrow + j);
+            KVT *dst = vc + (int64_t)p * kv_dim0 + j;
+            dst[0] = kv_c<KVT>(vv.x); dst[1] = kv_c<KVT>(vv.y);
+            dst[2] = kv_c<KVT>(vv.z); dst[3] = kv_c<KVT>(vv.w);
         }
     }
 }
@@ -1427,18 +1455,19 @@ __global__ void k_rope_kv(float *__restrict__ qkv, int ld,
 // ------------------------------------------------------------------ kv append
 // copy k,v batch rows into the caches at row pos[0]+b (reference OP_SHIFT,
 // nn-cpu-ops.cpp:1419-1441).
+template <typename KVT>
 __global__ void k_kv_append(const float *__restrict__ k,
                             const float *__restrict__ v,
-                            float *__restrict__ kc,
-                            float *__restrict__ vc,
+                            KVT *__restrict__ kc,
+                            KVT *__restrict__ vc,
                             const int *__restrict__ pos,
                             int kv_dim0) {
     const int b = blockIdx.y;
     const int p = pos[0] + b;
     for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < kv_dim0;
          i += gridDim.x * blockDim.x) {
-        kc[(int64_t)p * kv_dim0 + i] = k[(int64_t)b * kv_dim0 + i];
-        vc[(int64_t)p * kv_dim0 + i] = v[(int64_t)b * kv_dim0 + i];
+        kc[(int64_t)p * kv_dim0 + i] = kv_c<KVT>(k[(int64_t)b * kv_dim0 + i]);
+        vc[(int64_t)p * kv_dim0 + i] = kv_c<KVT>(v[(int64_t)b * kv_dim0 + i]);
     }
 }
 
@@ -1637,9 +1666,10 @@ struct GemvEpi {
     float *ssq = nullptr;
     const float *cache = nullptr;
     const int *pos = nullptr;
-    float *kc = nullptr;
+    float *kc = nullptr;  // f32 or f16 rows; kv_f16 says which
     float *vc = nullptr;
     int q_dim0 = 0, kv_dim0 = 0, hd = 0;
+    int kv_f16 = 0;
     bool force_rpw2 = false;
     // PRO=1 (fused norm+quant prologue) inputs
     const float *xf = nullptr;
@@ -1670,7 +1700,7 @@ static void gemv_launch(torch::Tensor &qs, torch::Tensor &scales, torch::Tensor 
                            PRO ? nullptr : xbs.data_ptr<float>(), y, d, n,
                            e.slot, e.x_resid, e.ssq,
                            e.cache, e.pos, e.kc, e.vc, e.q_dim0, e.kv_dim0, e.hd,
-                           e.xf, e.wnorm, e.ssq_in, e.eps);
+                           e.xf, e.wnorm, e.ssq_in, e.eps, e.kv_f16);
     };
     std::integral_constant<int, 1> r1;
     std::integral_constant<int, 2> r2;
@@ -1759,8 +1789,9 @@ void q40_gemv_rope(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
     GemvEpi e;
     e.cache = cache.data_ptr<float>();
     e.pos = pos.data_ptr<int>();
-    e.kc = kc.data_ptr<float>();
-    e.vc = vc.data_ptr<float>();
+    e.kv_f16 = kc.scalar_type() == at::kHalf ? 1 : 0;
+    e.kc = static_cast<float *>(kc.data_ptr());
+    e.vc = static_cast<float *>(vc.data_ptr());
     e.q_dim0 = (int)q_dim0;
     e.kv_dim0 = (int)kv_dim0;
     e.hd = (int)head_dim;
@@ -1799,8 +1830,9 @@ void q40_gemv_nq_rope(torch::Tensor qs, torch::Tensor scales, torch::Tensor x,
     e.eps = (float)eps;
     e.cache = cache.data_ptr<float>();
     e.pos = pos.data_ptr<int>();
-    e.kc = kc.data_ptr<float>();
-    e.vc = vc.data_ptr<float>();
+    e.kv_f16 = kc.scalar_type() == at::kHalf ? 1 : 0;
+    e.kc = static_cast<float *>(kc.data_ptr());
+    e.vc = static_cast<float *>(vc.data_ptr());
     e.q_dim0 = (int)q_dim0;
     e.kv_dim0 = (int)kv_dim0;
     e.hd = (int)head_dim;
@@ -1912,10 +1944,17 @@ void kv_append(torch::Tensor k, torch::Tensor v, torch::Tensor kc,
     const int B = k.size(0);
     const int kv_dim0 = k.size(1);
     const dim3 grid(ceil_div(kv_dim0, 256), B);
-    hipLaunchKernelGGL(k_kv_append, grid, dim3(256), 0, cur_stream(),
-                       k.data_ptr<float>(), v.data_ptr<float>(),
-                       kc.data_ptr<float>(), vc.data_ptr<float>(),
-                       pos.data_ptr<int>(), kv_dim0);
+    if (kc.scalar_type() == at::kHalf)
+        hipLaunchKernelGGL(k_kv_append<__half>, grid, dim3(256), 0, cur_stream(),
+                           k.data_ptr<float>(), v.data_ptr<float>(),
+                           reinterpret_cast<__half *>(kc.data_ptr<at::Half>()),
+                           reinterpret_cast<__half *>(vc.data_ptr<at::Half>()),
+                           pos.data_ptr<int>(), kv_dim0);
+    else
+        hipLaunchKernelGGL(k_kv_append<float>, grid, dim3(256), 0, cur_stream(),
+                           k.data_ptr<float>(), v.data_ptr<float>(),
+                           kc.data_ptr<float>(), vc.data_ptr<float>(),
+                           pos.data_ptr<int>(), kv_dim0);
 }
 
 void attn(torch::Tensor q, int64_t q_ld, torch::Tensor kc, torch::Tensor vc,
@@ -1931,9 +1970,20 @@ void attn(torch::Tensor q, int64_t q_ld, torch::Tensor kc, torch::Tensor vc,
     const dim3 grid(n_heads0, batch, splits);
     const dim3 cgrid(n_heads0, batch);
     const bool quant = zq.has_value();
+    const bool kv16 = kc.scalar_type() == at::kHalf;
     auto run = [&](auto vec_const) {
         constexpr int V = decltype(vec_const)::value;
-        hipLaunchKernelGGL((k_attn_split<V, false>), grid, dim3(256), 0, cur_stream(),
+        if (kv16)
+            hipLaunchKernelGGL((k_attn_split<V, __half>), grid, dim3(256), 0,
+                               cur_stream(), q.data_ptr<float>(), (int)q_ld,
+                               reinterpret_cast<const __half *>(kc.data_ptr<at::Half>()),
+                               reinterpret_cast<const __half *>(vc.data_ptr<at::Half>()),
+                               pos.data_ptr<int>(),
+                               (int)n_heads0, (int)kv_mul, kv_dim0, scale,
+                               ml_scratch.data_ptr<float>(), o_scratch.data_ptr<float>(),
+                               counter.data_ptr<int>(), nullptr, nullptr, nullptr, nullptr);
+        else
+            hipLaunchKernelGGL((k_attn_split<V, float>), grid, dim3(256), 0, cur_stream(),
                            q.data_ptr<float>(), (int)q_ld, kc.data_ptr<float>(),
                            vc.data_ptr<float>(), pos.data_ptr<int>(),
                            (int)n_heads0, (int)kv_mul, kv_dim0, scale,
@@ -2135,16 +2185,24 @@ void rope_kv(torch::Tensor qkv, int64_t ld, int64_t q_dim0, int64_t kv_dim0,
     CHECK_CUDA(qkv);
     const int total = (int)(q_dim0 / 2 + kv_dim0 / 2 + kv_dim0 / 4);
     const dim3 grid(ceil_div(total, 256), batch);
-    if (style == 0)
-        hipLaunchKernelGGL(k_rope_kv<0>, grid, dim3(256), 0, cur_stream(),
-                           qkv.data_ptr<float>(), (int)ld, (int)q_dim0, (int)kv_dim0,
-                           cache.data_ptr<float>(), pos.data_ptr<int>(),
-                           kc.data_ptr<float>(), vc.data_ptr<float>(), (int)head_dim);
+    auto run = [&](auto kvt, auto kcp, auto vcp) {
+        using KVT = decltype(kvt);
+        if (style == 0)
+            hipLaunchKernelGGL((k_rope_kv<0, KVT>), grid, dim3(256), 0, cur_stream(),
+                               qkv.data_ptr<float>(), (int)ld, (int)q_dim0, (int)kv_dim0,
+                               cache.data_ptr<float>(), pos.data_ptr<int>(),
+                               kcp, vcp, (int)head_dim);
+        else
+            hipLaunchKernelGGL((k_rope_kv<1, KVT>), grid, dim3(256), 0, cur_stream(),
+                               qkv.data_ptr<float>(), (int)ld, (int)q_dim0, (int)kv_dim0,
+                               cache.data_ptr<float>(), pos.data_ptr<int>(),
+                               kcp, vcp, (int)head_dim);
+    };
+    if (kc.scalar_type() == at::kHalf)
+        run(__half{}, reinterpret_cast<__half *>(kc.data_ptr<at::Half>()),
+            reinterpret_cast<__half *>(vc.data_ptr<at::Half>()));
     else
-        hipLaunchKernelGGL(k_rope_kv<1>, grid, dim3(256), 0, cur_stream(),
-                           qkv.data_ptr<float>(), (int)ld, (int)q_dim0, (int)kv_dim0,
-                           cache.data_ptr<float>(), pos.data_ptr<int>(),
-                           kc.data_ptr<float>(), vc.data_ptr<float>(), (int)head_dim);
+        run(0.0f, kc.data_ptr<float>(), vc.data_ptr<float>());
 }
 
 void token_from_argmax(torch::Tensor token, torch::Tensor scratch, int64_t count) {

@@ -466,3 +466,53 @@ def test_q40_gemv_grouped_v2_matches_v1(k):
         k.q40_gemv_grouped(qs, sc, q, s, bs, idx, y2, ka, variant=1)
         assert torch.allclose(y2, y1, atol=1e-4, rtol=1e-5), \
             (d, n, (y2 - y1).abs().max().item())
+
+
+def test_attn_f16_kv_matches_f32(k):
+    """f16 KV cache (the default since round 2) vs f32 KV on the same data:
+    split attention + combine must agree to f16 rounding."""
+    B, H0, hd, n_kv0, seq = 2, 4, 128, 2, 1100
+    kv_dim0 = n_kv0 * hd
+    kc = rand(seq, kv_dim0, seed=270, scale=0.3)
+    vc = rand(seq, kv_dim0, seed=271)
+    q = rand(B, H0 * hd, seed=272)
+    pos = torch.tensor([1050], dtype=torch.int32, device=DEV)
+    S = 16
+    ml = torch.zeros(B * H0 * S * 2, device=DEV)
+    osc = torch.zeros(B * H0 * S * hd, device=DEV)
+    cnt = torch.zeros(B * H0, dtype=torch.int32, device=DEV)
+    y32 = torch.zeros(B, H0 * hd, device=DEV)
+    k.attn(q, H0 * hd, kc, vc, y32, pos, B, H0, H0 // n_kv0, hd, S, ml, osc, cnt)
+    y16 = torch.zeros(B, H0 * hd, device=DEV)
+    k.attn(q, H0 * hd, kc.half(), vc.half(), y16, pos, B, H0, H0 // n_kv0, hd,
+           S, ml, osc, cnt)
+    assert torch.allclose(y16, y32, atol=2e-3, rtol=1e-2), \
+        (y16 - y32).abs().max().item()
+
+
+def test_rope_kv_and_append_f16(k):
+    """rope_kv and kv_append writing f16 caches must match their f32 writes
+    to f16 rounding."""
+    B, hd, qh, kvh, seq = 2, 64, 4, 2, 32
+    q_dim0, kv_dim0 = qh * hd, kvh * hd
+    ld = q_dim0 + 2 * kv_dim0
+    buf = rand(B, ld, seed=286)
+    buf16 = buf.clone()
+    cache = R.rope_cache(seq, hd, 10000.0).to(DEV).reshape(seq, hd).contiguous()
+    kc = torch.zeros(seq, kv_dim0, device=DEV)
+    vc = torch.zeros(seq, kv_dim0, device=DEV)
+    kc16 = torch.zeros(seq, kv_dim0, dtype=torch.float16, device=DEV)
+    vc16 = torch.zeros(seq, kv_dim0, dtype=torch.float16, device=DEV)
+    pos = torch.tensor([3], dtype=torch.int32, device=DEV)
+    k.rope_kv(buf, ld, q_dim0, kv_dim0, cache, pos, kc, vc, hd, 0, B)
+    k.rope_kv(buf16, ld, q_dim0, kv_dim0, cache, pos, kc16, vc16, hd, 0, B)
+    assert torch.equal(buf, buf16)  # q rotation identical
+    assert torch.allclose(kc16.float(), kc, atol=2e-3)
+    assert torch.allclose(vc16.float(), vc, atol=2e-3)
+    kd = rand(2, kv_dim0, seed=287)
+    vd = rand(2, kv_dim0, seed=288)
+    p0 = torch.tensor([9], dtype=torch.int32, device=DEV)
+    k.kv_append(kd, vd, kc, vc, p0)
+    k.kv_append(kd, vd, kc16, vc16, p0)
+    assert torch.allclose(kc16[9:11].float(), kc[9:11], atol=1e-3)
+    assert torch.allclose(vc16[9:11].float(), vc[9:11], atol=1e-3)
