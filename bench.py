@@ -135,6 +135,11 @@ def main():
     use_graph = not args.no_graph and not use_cpu
     if use_graph:
         try:
+            # pick the K-split count for the timed region's context length
+            # (the graph replays below bypass forward()'s adaptive recapture)
+            sp = model._pick_splits(args.prefill)
+            if sp != model.attn_splits:
+                model._set_attn_splits(sp)
             model.capture_decode_graph()
             model.pos.fill_(args.prefill)
             model._graph_pos = args.prefill

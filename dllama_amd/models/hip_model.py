@@ -239,10 +239,15 @@ class HipTransformer:
         # fused quantize-into-wire sync: validated bit-identical on hardware
         # (test_sync_quant_pack_matches_two_kernel_path); default on
         self.fused_sync = _os.environ.get("DLLAMA_FUSED_SYNC", "1") == "1"
-        # past this decode position, recapture the graph with S=16 K-splits
-        # (measured: S=16 wins from ~pos 512 up, tools/attn_kv16_probe);
-        # 0 disables
+        # adaptive K-split schedule: S=1 fused single-kernel attention below
+        # pos 256 (no combine launch), S=8 to the threshold, S=16 beyond
+        # (tools/attn_kv16_probe: S=16 wins from ~pos 512). The decode graph
+        # is recaptured when pos crosses a boundary. DLLAMA_ADAPTIVE_SPLITS=0
+        # (or an explicit DLLAMA_ATTN_SPLITS) pins S.
         self.adaptive_thresh = int(_os.environ.get("DLLAMA_ADAPTIVE_SPLITS", "512"))
+        if ("DLLAMA_ATTN_SPLITS" in _os.environ
+                and "DLLAMA_ADAPTIVE_SPLITS" not in _os.environ):
+            self.adaptive_thresh = 0
         # S=8 measured best at decode (16/32: combine reads S partials)
         self.attn_ml = torch.zeros(NB * c.n_heads0 * self.attn_splits * 2, device=dev)
         self.attn_o = torch.zeros(NB * c.n_heads0 * self.attn_splits * c.head_dim,
@@ -475,9 +480,9 @@ class HipTransformer:
                 "(rebuild with a larger --max-seq-len / seq_len)")
         self.tokens[:B].copy_(tokens.to(self.device), non_blocking=True)
         if B == 1 and self._graph is not None:
-            if (self.adaptive_thresh and p0 >= self.adaptive_thresh
-                    and self.attn_splits < 16):
-                self._set_attn_splits(16)  # recapture with more K-splits
+            sp = self._pick_splits(p0)
+            if sp != self.attn_splits:
+                self._set_attn_splits(sp)  # recapture at the new K-split count
             if p0 != self._graph_pos:
                 self.pos.fill_(p0)
             self._graph.replay()
@@ -506,6 +511,15 @@ class HipTransformer:
         return self.logits0[:B]
 
     # ------------------------------------------------------------ graphs
+
+    def _pick_splits(self, pos: int) -> int:
+        if not self.adaptive_thresh:
+            return self.attn_splits
+        if pos < 256:
+            return 1
+        if pos < self.adaptive_thresh:
+            return 8
+        return 16
 
     def _set_attn_splits(self, s: int):
         """Switch the flash-decode K-split count mid-stream (long-context

@@ -67,6 +67,13 @@ __device__ __forceinline__ float group16_reduce_sum(float v) {
     return v;
 }
 
+__device__ __forceinline__ float group16_reduce_max(float v) {
+    #pragma unroll
+    for (int off = 8; off > 0; off >>= 1)
+        v = fmaxf(v, __shfl_xor(v, off, 16));
+    return v;
+}
+
 __device__ __forceinline__ float ssq_total(const float *ssq, int b) {
     float t = 0.0f;
     #pragma unroll
@@ -710,7 +717,7 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
 // Stage 1: grid (H0, B, S); split sp covers t = (sp*4+wave) + 4*S*i — the
 // t-range is spread over S*4 waves so long contexts fill the chip.
 // Per-split (m, l, o) goes to scratch; stage 2 combines the S partials.
-template <int VEC, typename KVT>
+template <int VEC, typename KVT, bool FUSE = false>
 __global__ void k_attn_split(const float *__restrict__ q, int q_ld,
                              const KVT *__restrict__ kc,
                              const KVT *__restrict__ vc,
@@ -797,14 +804,56 @@ __global__ void k_attn_split(const float *__restrict__ q, int q_ld,
     if (wave == 0) {
         const float L = sl[0] * __expf(sm[0] - M) + sl[1] * __expf(sm[1] - M)
                       + sl[2] * __expf(sm[2] - M) + sl[3] * __expf(sm[3] - M);
-        if (lane == 0) {
-            ml_scratch[slot * 2] = M;
-            ml_scratch[slot * 2 + 1] = L;
-        }
-        #pragma unroll
-        for (int v = 0; v < VEC; v++) {
-            const int i = lane * VEC + v;
-            o_scratch[slot * hd + i] = so[0][i] + so[1][i] + so[2][i] + so[3][i];
+        if constexpr (FUSE) {
+            // S==1 single-kernel path (short contexts): normalize and emit
+            // the Q80 triple of this head directly — no scratch round-trip,
+            // no combine launch (the 2-kernel pair costs ~11 us in-graph at
+            // pos<256 where the actual KV read is microseconds)
+            const float invL = 1.0f / L;
+            float vv[VEC];
+            #pragma unroll
+            for (int v = 0; v < VEC; v++) {
+                const int i = lane * VEC + v;
+                vv[v] = (so[0][i] + so[1][i] + so[2][i] + so[3][i]) * invL;
+            }
+            if constexpr (VEC == 2) {
+                // lane holds elems (2*lane, 2*lane+1): block = 16-lane group
+                const float amax = group16_reduce_max(fmaxf(fabsf(vv[0]), fabsf(vv[1])));
+                const float dd = amax / 127.0f;
+                const float qinv = dd > 0.0f ? 1.0f / dd : 0.0f;
+                const float q0 = rintf(vv[0] * qinv), q1 = rintf(vv[1] * qinv);
+                const int64_t obase = ((int64_t)b * n_heads0 + h0) * hd + lane * 2;
+                zq[obase] = (int8_t)q0;
+                zq[obase + 1] = (int8_t)q1;
+                const float bsum = group16_reduce_sum(q0 + q1);
+                if ((lane & 15) == 0) {
+                    const int blk = (h0 * hd + lane * 2) / QB;
+                    zs[(int64_t)b * (n_heads0 * hd / QB) + blk] = dd;
+                    zbs[(int64_t)b * (n_heads0 * hd / QB) + blk] = bsum;
+                }
+            } else {
+                const float amax = group32_reduce_max(fabsf(vv[0]));
+                const float dd = amax / 127.0f;
+                const float qinv = dd > 0.0f ? 1.0f / dd : 0.0f;
+                const float q0 = rintf(vv[0] * qinv);
+                zq[((int64_t)b * n_heads0 + h0) * hd + lane] = (int8_t)q0;
+                const float bsum = group32_reduce_sum(q0);
+                if ((lane & 31) == 0) {
+                    const int blk = (h0 * hd + lane) / QB;
+                    zs[(int64_t)b * (n_heads0 * hd / QB) + blk] = dd;
+                    zbs[(int64_t)b * (n_heads0 * hd / QB) + blk] = bsum;
+                }
+            }
+        } else {
+            if (lane == 0) {
+                ml_scratch[slot * 2] = M;
+                ml_scratch[slot * 2 + 1] = L;
+            }
+            #pragma unroll
+            for (int v = 0; v < VEC; v++) {
+                const int i = lane * VEC + v;
+                o_scratch[slot * hd + i] = so[0][i] + so[1][i] + so[2][i] + so[3][i];
+            }
         }
     }
 
@@ -1971,6 +2020,36 @@ void attn(torch::Tensor q, int64_t q_ld, torch::Tensor kc, torch::Tensor vc,
     const dim3 cgrid(n_heads0, batch);
     const bool quant = zq.has_value();
     const bool kv16 = kc.scalar_type() == at::kHalf;
+    // splits==1 + quant output: single fused kernel (normalize + Q80 emit
+    // in the split kernel), no combine launch
+    if (splits == 1 && quant) {
+        auto frun = [&](auto vec_const, auto kvt, auto kcp, auto vcp) {
+            constexpr int V = decltype(vec_const)::value;
+            using KVT = decltype(kvt);
+            hipLaunchKernelGGL((k_attn_split<V, KVT, true>), grid, dim3(256), 0,
+                               cur_stream(), q.data_ptr<float>(), (int)q_ld,
+                               kcp, vcp, pos.data_ptr<int>(),
+                               (int)n_heads0, (int)kv_mul, kv_dim0, scale,
+                               ml_scratch.data_ptr<float>(),
+                               o_scratch.data_ptr<float>(),
+                               counter.data_ptr<int>(), nullptr,
+                               zq->data_ptr<int8_t>(), zs->data_ptr<float>(),
+                               zbs->data_ptr<float>());
+        };
+        auto fdisp = [&](auto vec_const) {
+            if (kv16)
+                frun(vec_const, __half{},
+                     reinterpret_cast<const __half *>(kc.data_ptr<at::Half>()),
+                     reinterpret_cast<const __half *>(vc.data_ptr<at::Half>()));
+            else
+                frun(vec_const, 0.0f, (const float *)kc.data_ptr<float>(),
+                     (const float *)vc.data_ptr<float>());
+        };
+        if (head_dim == 128) fdisp(std::integral_constant<int, 2>{});
+        else if (head_dim == 64) fdisp(std::integral_constant<int, 1>{});
+        else TORCH_CHECK(false, "unsupported head_dim ", head_dim);
+        return;
+    }
     auto run = [&](auto vec_const) {
         constexpr int V = decltype(vec_const)::value;
         if (kv16)

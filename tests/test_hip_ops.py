@@ -516,3 +516,36 @@ def test_rope_kv_and_append_f16(k):
     k.kv_append(kd, vd, kc16, vc16, p0)
     assert torch.allclose(kc16[9:11].float(), kc[9:11], atol=1e-3)
     assert torch.allclose(vc16[9:11].float(), vc[9:11], atol=1e-3)
+
+
+def test_attn_fused_s1_matches_split(k):
+    """splits=1 + quant output runs the fused single-kernel attention (no
+    combine launch); its Q80 triple must match the S=8 split+combine pair."""
+    for hd, H0, n_kv0 in ((128, 4, 2), (64, 2, 1)):
+        B, seq = 2, 300
+        kv_dim0 = n_kv0 * hd
+        kc = rand(seq, kv_dim0, seed=290 + hd, scale=0.3).half()
+        vc = rand(seq, kv_dim0, seed=291 + hd).half()
+        q = rand(B, H0 * hd, seed=292)
+        pos = torch.tensor([200], dtype=torch.int32, device=DEV)
+        nb = H0 * hd // 32
+
+        def run(S):
+            ml = torch.zeros(B * H0 * S * 2, device=DEV)
+            osc = torch.zeros(B * H0 * S * hd, device=DEV)
+            cnt = torch.zeros(B * H0, dtype=torch.int32, device=DEV)
+            zq = torch.zeros(B, H0 * hd, dtype=torch.int8, device=DEV)
+            zs = torch.zeros(B, nb, device=DEV)
+            zbs = torch.zeros(B, nb, device=DEV)
+            k.attn(q, H0 * hd, kc, vc, torch.zeros(B, H0 * hd, device=DEV),
+                   pos, B, H0, H0 // n_kv0, hd, S, ml, osc, cnt, zq, zs, zbs)
+            return zq, zs, zbs
+
+        q1, s1, b1 = run(1)
+        q8, s8, b8 = run(8)
+        got = R.q80_dequantize(q1.cpu(), s1.cpu())
+        want = R.q80_dequantize(q8.cpu(), s8.cpu())
+        tol = want.abs().max().item() / 50 + 1e-5
+        assert torch.allclose(got, want, atol=tol), \
+            (hd, (got - want).abs().max().item())
+        assert torch.allclose(b1, b8, atol=2.0), (b1 - b8).abs().max().item()
