@@ -413,11 +413,21 @@ class HipTransformer:
                              self.t_norm[:NB])
                 self._moe_ffn(B, NB, lw, slot + 1)
             else:
-                norm_gemv(lw["w13"], lw["norm1"], slot, self.ff_out)
-                k.swiglu_q80(self.ff_out, self.ff_out[:, c.ff_dim0:],
-                             2 * c.ff_dim0, c.ff_dim0, NB, self.dq.q[:NB],
-                             self.dq.s[:NB], self.dq.bs[:NB],
-                             c.hidden_act == HIDDEN_ACT_GELU)
+                gelu = c.hidden_act == HIDDEN_ACT_GELU
+                if NB == 1 and not fused_norm and c.ff_dim0 % 32 == 0:
+                    # fused W1|W3 GEMV + SwiGLU + Q80 emit (one launch
+                    # replacing gemv + swiglu_q80)
+                    k.norm_quant(x[:NB], lw["norm1"], self.ssq[slot],
+                                 self.xq.q[:NB], self.xq.s[:NB],
+                                 self.xq.bs[:NB], NB, c.norm_eps)
+                    k.q40_gemv_swiglu(lw["w13"].qs, lw["w13"].scales,
+                                      self.xq.q, self.xq.s, self.xq.bs,
+                                      self.dq.q, self.dq.s, self.dq.bs, gelu)
+                else:
+                    norm_gemv(lw["w13"], lw["norm1"], slot, self.ff_out)
+                    k.swiglu_q80(self.ff_out, self.ff_out[:, c.ff_dim0:],
+                                 2 * c.ff_dim0, c.ff_dim0, NB, self.dq.q[:NB],
+                                 self.dq.s[:NB], self.dq.bs[:NB], gelu)
                 self._proj_merge(lw["w2"], self.dq, slot + 1, NB)
             slot += 1
 

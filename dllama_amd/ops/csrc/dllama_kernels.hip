@@ -1553,6 +1553,84 @@ __global__ void k_swiglu_q80(const float *__restrict__ a,
     if (lane == 0) { s[gid] = d; bs[gid] = bsum; }
 }
 
+// fused W1|W3 GEMV + SwiGLU + Q80 quantize (decode B=1): one kernel replaces
+// gemv(w13) + k_swiglu_q80 — per-kernel in-graph overhead is ~3-5 us, so at
+// 32 layers/step the saved launch is worth ~150 us/token. The workgroup
+// owns one Q80 output block (32 ff indices): wave w computes the W1 and W3
+// rows of indices (32*bx+2w, +1) with 4 row streams in flight; the block's
+// silu(a)*g values assemble in LDS and the first 32 lanes quantize.
+// (reference runs matmul w1, matmul w3, silu, mul, cast as 5 ops,
+// llm.cpp:430-448 / nn-cpu-ops.cpp:462-500.)
+template <bool GELU>
+__global__ __launch_bounds__(1024) void k_q40_gemv_swiglu(
+        const uint8_t *__restrict__ qs, const __half *__restrict__ scales,
+        const int8_t *__restrict__ xq, const float *__restrict__ xs,
+        const float *__restrict__ xbs, int ff, int n,
+        int8_t *__restrict__ oq, float *__restrict__ os,
+        float *__restrict__ obs) {
+    const int wave = threadIdx.x / WAVE;  // 16 waves, 2 ff indices each
+    const int lane = threadIdx.x % WAVE;
+    const int i0 = blockIdx.x * 32 + 2 * wave;
+    const int nb = n / QB, nbp = nb >> 1;
+    const int rows[4] = {i0, i0 + 1, ff + i0, ff + i0 + 1};
+    const uint4 *wrow[4];
+    const __half *srow[4];
+    #pragma unroll
+    for (int r = 0; r < 4; r++) {
+        wrow[r] = reinterpret_cast<const uint4 *>(qs + (int64_t)rows[r] * (n >> 1));
+        srow[r] = scales + (int64_t)rows[r] * nb;
+    }
+    float acc[4] = {0.0f, 0.0f, 0.0f, 0.0f};
+    for (int jp = lane; jp < nbp; jp += WAVE) {
+        const int j = jp << 1;
+        const int4 *xr = reinterpret_cast<const int4 *>(xq) + j * 2;
+        const int4 x0 = xr[0], x1 = xr[1], x2 = xr[2], x3 = xr[3];
+        const float2 sx = *reinterpret_cast<const float2 *>(xs + j);
+        const float2 bsum = *reinterpret_cast<const float2 *>(xbs + j);
+        #pragma unroll
+        for (int r = 0; r < 4; r++) {
+            const uint4 w0 = wrow[r][j], w1 = wrow[r][j + 1];
+            const float2 sw = __half22float2(*reinterpret_cast<const __half2 *>(srow[r] + j));
+            acc[r] = fmaf(sw.x * sx.x, (float)q40_block_dot(w0, x0, x1) - 8.0f * bsum.x, acc[r]);
+            acc[r] = fmaf(sw.y * sx.y, (float)q40_block_dot(w1, x2, x3) - 8.0f * bsum.y, acc[r]);
+        }
+    }
+    if ((nb & 1) && lane == 0) {  // odd trailing block
+        const int j = nb - 1;
+        const int4 *xb = reinterpret_cast<const int4 *>(xq) + j * 2;
+        const float sx1 = xs[j], bs1 = xbs[j];
+        #pragma unroll
+        for (int r = 0; r < 4; r++)
+            acc[r] = fmaf(__half2float(srow[r][j]) * sx1,
+                          (float)q40_block_dot(wrow[r][j], xb[0], xb[1]) - 8.0f * bs1,
+                          acc[r]);
+    }
+    #pragma unroll
+    for (int r = 0; r < 4; r++) acc[r] = wave_reduce_sum(acc[r]);
+    __shared__ float sv[32];
+    if (lane == 0) {
+        #pragma unroll
+        for (int t = 0; t < 2; t++) {
+            const float av = acc[t], gv = acc[2 + t];
+            const float act = GELU
+                ? 0.5f * av * (1.0f + tanhf(0.797884560802865f * (av + 0.044715f * av * av * av)))
+                : av / (1.0f + __expf(-av));
+            sv[2 * wave + t] = act * gv;
+        }
+    }
+    __syncthreads();
+    if (threadIdx.x < 32) {
+        const float v = sv[threadIdx.x];
+        const float amax = group32_reduce_max(fabsf(v));
+        const float dd = amax / 127.0f;
+        const float qinv = dd > 0.0f ? 1.0f / dd : 0.0f;
+        const float qf = rintf(v * qinv);
+        oq[(int64_t)blockIdx.x * QB + threadIdx.x] = (int8_t)qf;
+        const float bsum = group32_reduce_sum(qf);
+        if (threadIdx.x == 0) { os[blockIdx.x] = dd; obs[blockIdx.x] = bsum; }
+    }
+}
+
 __global__ void k_silu_mul(const float *__restrict__ a,
                            const float *__restrict__ g,
                            float *__restrict__ out,
@@ -2291,6 +2369,31 @@ void token_from_argmax(torch::Tensor token, torch::Tensor scratch, int64_t count
                        (int)count);
 }
 
+void q40_gemv_swiglu(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
+                     torch::Tensor xs, torch::Tensor xbs, torch::Tensor oq,
+                     torch::Tensor os, torch::Tensor obs, bool gelu = false) {
+    // B=1 decode FFN: W1|W3 GEMV + SwiGLU + Q80 emit, one launch
+    CHECK_CUDA(qs); CHECK_CONT(qs); CHECK_CONT(xq);
+    const int d = qs.size(0);
+    const int ff = d / 2;
+    const int n = qs.size(1) * 2;
+    TORCH_CHECK(ff % 32 == 0, "fused swiglu GEMV needs ff % 32 == 0");
+    TORCH_CHECK(xq.size(-1) == n, "x width mismatch");
+    const dim3 grid(ff / 32);
+    const dim3 block(16 * WAVE);
+    auto launch = [&](auto k) {
+        hipLaunchKernelGGL(k, grid, block, 0, cur_stream(),
+                           qs.data_ptr<uint8_t>(),
+                           reinterpret_cast<const __half *>(scales.data_ptr<at::Half>()),
+                           xq.data_ptr<int8_t>(), xs.data_ptr<float>(),
+                           xbs.data_ptr<float>(), ff, n,
+                           oq.data_ptr<int8_t>(), os.data_ptr<float>(),
+                           obs.data_ptr<float>());
+    };
+    if (gelu) launch(k_q40_gemv_swiglu<true>);
+    else launch(k_q40_gemv_swiglu<false>);
+}
+
 void silu_mul(torch::Tensor a, torch::Tensor g, torch::Tensor out) {
     CHECK_CUDA(a);
     const int64_t n = a.numel();
@@ -2466,6 +2569,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("xq"), py::arg("xs"), py::arg("y"), py::arg("batch"),
           py::arg("part") = py::none(), py::arg("variant") = -1);
     m.def("q40_gemv_rope", &q40_gemv_rope);
+    m.def("q40_gemv_swiglu", &q40_gemv_swiglu, py::arg("qs"), py::arg("scales"),
+          py::arg("xq"), py::arg("xs"), py::arg("xbs"), py::arg("oq"),
+          py::arg("os"), py::arg("obs"), py::arg("gelu") = false);
     m.def("q40_gemv_nq", &q40_gemv_nq, py::arg("qs"), py::arg("scales"),
           py::arg("x"), py::arg("wnorm"), py::arg("ssq"), py::arg("eps"),
           py::arg("y"), py::arg("batch"), py::arg("amax_slot") = py::none());

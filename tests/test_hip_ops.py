@@ -549,3 +549,30 @@ def test_attn_fused_s1_matches_split(k):
         assert torch.allclose(got, want, atol=tol), \
             (hd, (got - want).abs().max().item())
         assert torch.allclose(b1, b8, atol=2.0), (b1 - b8).abs().max().item()
+
+
+def test_q40_gemv_swiglu_fused(k):
+    """Fused W1|W3 GEMV + SwiGLU + Q80 emit vs the two-kernel path."""
+    ff, n = 96, 512  # ff % 32 == 0, 3 workgroups
+    qs, sc, wref = _mk_linear(2 * ff, n, 300)
+    x = rand(1, n, seed=301, scale=0.5)
+    q = torch.zeros(1, n, dtype=torch.int8, device=DEV)
+    s = torch.zeros(1, n // 32, device=DEV)
+    bs = torch.zeros(1, n // 32, device=DEV)
+    k.q80_quantize(x, q, s, bs)
+    # reference: gemv then swiglu_q80
+    y = torch.zeros(1, 2 * ff, device=DEV)
+    k.q40_gemv(qs, sc, q, s, bs, y, 1)
+    wq = torch.zeros(1, ff, dtype=torch.int8, device=DEV)
+    ws = torch.zeros(1, ff // 32, device=DEV)
+    wbs = torch.zeros(1, ff // 32, device=DEV)
+    k.swiglu_q80(y, y[:, ff:], 2 * ff, ff, 1, wq, ws, wbs)
+    fq = torch.zeros(1, ff, dtype=torch.int8, device=DEV)
+    fs = torch.zeros(1, ff // 32, device=DEV)
+    fbs = torch.zeros(1, ff // 32, device=DEV)
+    k.q40_gemv_swiglu(qs, sc, q, s, bs, fq, fs, fbs)
+    got = R.q80_dequantize(fq.cpu(), fs.cpu())
+    want = R.q80_dequantize(wq.cpu(), ws.cpu())
+    tol = want.abs().max().item() / 100 + 1e-6
+    assert torch.allclose(got, want, atol=tol), (got - want).abs().max().item()
+    assert torch.allclose(fbs.cpu(), wbs.cpu(), atol=2.0)
