@@ -386,6 +386,15 @@ class HipTransformer:
                                 self.xq.s, self.xq.bs, self.qkv_out, NB,
                                 self.rope_cache, self.pos, self.k_cache[l],
                                 self.v_cache[l], c.q_dim0, c.kv_dim0, c.head_dim)
+            elif c.is_qwen3 and self.rope_style == 1:
+                # per-head q/k rmsnorm + neox rope + KV write, one launch
+                # (replaces 2x rmsnorm_rows_s + rope_kv = ~9 us/layer of
+                # launch overhead in the captured graph)
+                norm_gemv(lw["qkv"], lw["norm0"], slot, self.qkv_out)
+                k.rope_kv_qknorm(self.qkv_out, self.qkv_ld, c.q_dim0,
+                                 c.kv_dim0, self.rope_cache, self.pos,
+                                 self.k_cache[l], self.v_cache[l], c.head_dim,
+                                 lw["q_norm"], lw["k_norm"], c.norm_eps, B)
             else:
                 norm_gemv(lw["qkv"], lw["norm0"], slot, self.qkv_out)
                 if c.is_qwen3:
@@ -458,6 +467,27 @@ class HipTransformer:
         ka = c.n_active_experts
         S = NB * ka
         k.router_gemv(lw["gate"], self.t_norm, self.moe_router, NB)
+        if c.ff_dim0 % 32 == 0 and ka <= 16:
+            # gate-fused decode path: every consumer recomputes the
+            # deterministic top-k from the router logits in-kernel, so the
+            # FFN is 3 launches (w13+swiglu, w2, scale-merge) instead of 6
+            router = self.moe_router[:NB]
+            k.q40_gemv_grouped_swiglu(lw["w13"].qs, lw["w13"].scales,
+                                      self.xq.q, self.xq.s, self.xq.bs,
+                                      self.moe_dq.q, self.moe_dq.s,
+                                      self.moe_dq.bs, S, router, ka)
+            k.q40_gemv_grouped(lw["w2"].qs, lw["w2"].scales, self.moe_dq.q,
+                               self.moe_dq.s, self.moe_dq.bs,
+                               self.moe_idx[:S], self.moe_y, 1,
+                               router=router, topk=ka, n_slots=S)
+            if not self.tp_path:
+                k.scale_merge_add(self.x[:NB], self.moe_y, router,
+                                  self.ssq[slot], NB, ka, gate=True)
+            else:
+                k.scale_merge(self.partial[:NB], self.moe_y, router, NB, ka,
+                              gate=True)
+                self._sync_partial(NB, slot)
+            return
         k.moe_gate(self.moe_router[:NB], self.moe_idx, self.moe_wts, NB, ka)
         k.q40_gemv_grouped(lw["w13"].qs, lw["w13"].scales, self.xq.q, self.xq.s,
                            self.xq.bs, self.moe_idx[:S], self.moe_out13, ka)
@@ -523,13 +553,13 @@ class HipTransformer:
     # ------------------------------------------------------------ graphs
 
     def _pick_splits(self, pos: int) -> int:
+        # S=8 short context, S=16 past the threshold (tools/attn_kv16_probe).
+        # An S=1 fused single-kernel path exists but measured SLOWER in-model
+        # (13.5 us vs the ~11 us pair: the 32-workgroup grid starves the
+        # dispatcher) — never auto-picked.
         if not self.adaptive_thresh:
             return self.attn_splits
-        if pos < 256:
-            return 1
-        if pos < self.adaptive_thresh:
-            return 8
-        return 16
+        return 8 if pos < self.adaptive_thresh else 16
 
     def _set_attn_splits(self, s: int):
         """Switch the flash-decode K-split count mid-stream (long-context

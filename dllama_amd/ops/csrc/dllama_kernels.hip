@@ -74,6 +74,74 @@ __device__ __forceinline__ float group16_reduce_max(float v) {
     return v;
 }
 
+// whole-wave MoE gate (reference OP_SOFTMAX + OP_MOE_GATE,
+// nn-cpu-ops.cpp:595-720,1462-1492): softmax over the router logits, then
+// iterative top-k with smallest-index tie-break, weights normalized by the
+// top-k sum. Every lane of the executing wave returns the full idx/weight
+// arrays (static-indexed — runtime-indexed locals land in scratch).
+// Callable from consumer kernels so the gate costs ~0.3 us of redundant
+// VALU per wave instead of its own ~10 us single-workgroup launch.
+__device__ __forceinline__ void moe_gate_wave(
+        const float *__restrict__ logits, int n_experts, int topk, int lane,
+        int *out_idx, float *out_w) {
+    const int per = (n_experts + WAVE - 1) / WAVE;
+    // fixed-trip unrolled loops: runtime-indexed arrays go to scratch
+    float v[16];  // per-lane expert logits (supports n_experts <= 1024)
+    float m = -1e30f;
+    #pragma unroll
+    for (int i = 0; i < 16; i++) {
+        const int eidx = lane * per + i;
+        v[i] = (i < per && eidx < n_experts) ? logits[eidx] : -1e30f;
+        m = fmaxf(m, v[i]);
+    }
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        m = fmaxf(m, __shfl_xor(m, off, WAVE));
+    float sum = 0.0f;
+    #pragma unroll
+    for (int i = 0; i < 16; i++) {
+        v[i] = (i < per && (lane * per + i) < n_experts) ? __expf(v[i] - m) : 0.0f;
+        sum += v[i];
+    }
+    sum = wave_reduce_sum(sum);
+    const float inv = 1.0f / sum;
+    // iterative top-k: packed (prob, smallest-index-wins) max per round
+    float wsum = 0.0f;
+    float chosen[16];  // topk <= 16
+    #pragma unroll
+    for (int t = 0; t < 16; t++) {
+        if (t >= topk) continue;  // guarded full unroll (break blocks unrolling)
+        float best = -1.0f;
+        int bi = -1;
+        #pragma unroll
+        for (int i = 0; i < 16; i++) {
+            if (i < per && v[i] > best) { best = v[i]; bi = lane * per + i; }
+        }
+        // wave argmax: (prob, -index) lexicographic via packed compare
+        #pragma unroll
+        for (int off = 32; off > 0; off >>= 1) {
+            const float ob = __shfl_xor(best, off, WAVE);
+            const int oi = __shfl_xor(bi, off, WAVE);
+            if (ob > best || (ob == best && oi >= 0 && (bi < 0 || oi < bi))) {
+                best = ob; bi = oi;
+            }
+        }
+        out_idx[t] = bi;
+        chosen[t] = best * inv;
+        wsum += best * inv;
+        // clear the winner (static-index scan, see scratch note above)
+        #pragma unroll
+        for (int i = 0; i < 16; i++)
+            if (lane * per + i == bi) v[i] = -1.0f;
+    }
+    // normalize by the top-k sum (reference normTopk)
+    const float winv = 1.0f / wsum;
+    #pragma unroll
+    for (int t = 0; t < 16; t++)
+        out_w[t] = t < topk ? chosen[t] * winv : 0.0f;
+}
+
+
 __device__ __forceinline__ float ssq_total(const float *ssq, int b) {
     float t = 0.0f;
     #pragma unroll
@@ -914,7 +982,7 @@ __global__ void k_attn_combine(const float *__restrict__ ml_scratch,
 // row-pair groups so every lane stays busy at MoE shapes where nbp < 64
 // (Qwen3-30B w2: nbp=12 left 52 of 64 lanes idle -> the ~5x-off-stream
 // grouped GEMV; see tools/moe_gemv_probe.hip).
-template <int LPP>
+template <int LPP, bool GATE = false>
 __global__ void k_q40_gemv_grouped(const uint8_t *__restrict__ qs,
                                    const __half *__restrict__ scales,
                                    const int8_t *__restrict__ xq,
@@ -922,7 +990,9 @@ __global__ void k_q40_gemv_grouped(const uint8_t *__restrict__ qs,
                                    const float *__restrict__ xbs,
                                    const int *__restrict__ expert_idx,
                                    float *__restrict__ y,
-                                   int d, int n, int k_slots) {
+                                   int d, int n, int k_slots,
+                                   const float *__restrict__ router,
+                                   int n_experts, int topk) {
     constexpr int NGRP = WAVE / LPP;
     const int wpb = blockDim.x / WAVE;
     const int full_lane = threadIdx.x % WAVE;
@@ -933,7 +1003,22 @@ __global__ void k_q40_gemv_grouped(const uint8_t *__restrict__ qs,
     const int lane = full_lane % LPP;
     const int nb = n / QB;
     const int nbp = nb >> 1;
-    const int e = expert_idx[slot];
+    int e;
+    if constexpr (GATE) {
+        // in-kernel gate: each wave recomputes the deterministic top-k from
+        // the router logits — removes the k_moe_gate launch from the chain
+        int gi[16];
+        float gw[16];
+        const int br = slot / topk, want = slot - br * topk;
+        moe_gate_wave(router + (int64_t)br * n_experts, n_experts, topk,
+                      full_lane, gi, gw);
+        e = 0;
+        #pragma unroll
+        for (int t = 0; t < 16; t++)
+            if (t == want) e = gi[t];
+    } else {
+        e = expert_idx[slot];
+    }
     const int b = slot / k_slots;
     const int row1 = min(row0 + 1, d - 1);
     const uint4 *wrow0 = reinterpret_cast<const uint4 *>(
@@ -1259,83 +1344,51 @@ __global__ void k_moe_gate(const float *__restrict__ logits,
                            int n_experts, int topk) {
     const int b = blockIdx.x;
     const int lane = threadIdx.x;  // blockDim == 64
-    const int per = (n_experts + WAVE - 1) / WAVE;
-    // fixed-trip unrolled loops: runtime-indexed arrays go to scratch
-    float v[16];  // per-lane expert logits (supports n_experts <= 1024)
-    float m = -1e30f;
-    #pragma unroll
-    for (int i = 0; i < 16; i++) {
-        const int eidx = lane * per + i;
-        v[i] = (i < per && eidx < n_experts)
-                   ? logits[(int64_t)b * n_experts + eidx] : -1e30f;
-        m = fmaxf(m, v[i]);
-    }
-    #pragma unroll
-    for (int off = 32; off > 0; off >>= 1)
-        m = fmaxf(m, __shfl_xor(m, off, WAVE));
-    float sum = 0.0f;
-    #pragma unroll
-    for (int i = 0; i < 16; i++) {
-        v[i] = (i < per && (lane * per + i) < n_experts) ? __expf(v[i] - m) : 0.0f;
-        sum += v[i];
-    }
-    sum = wave_reduce_sum(sum);
-    const float inv = 1.0f / sum;
-    // iterative top-k: packed (prob, smallest-index-wins) max per round
-    float wsum = 0.0f;
-    float chosen[16];  // topk <= 16
-    #pragma unroll
-    for (int t = 0; t < 16; t++) {
-        if (t >= topk) continue;  // guarded full unroll (break blocks unrolling)
-        float best = -1.0f;
-        int bi = -1;
-        #pragma unroll
-        for (int i = 0; i < 16; i++) {
-            if (i < per && v[i] > best) { best = v[i]; bi = lane * per + i; }
-        }
-        // wave argmax: (prob, -index) lexicographic via packed compare
-        #pragma unroll
-        for (int off = 32; off > 0; off >>= 1) {
-            const float ob = __shfl_xor(best, off, WAVE);
-            const int oi = __shfl_xor(bi, off, WAVE);
-            if (ob > best || (ob == best && oi >= 0 && (bi < 0 || oi < bi))) {
-                best = ob; bi = oi;
-            }
-        }
-        if (lane == 0) idx[(int64_t)b * topk + t] = bi;
-        chosen[t] = best * inv;
-        wsum += best * inv;
-        // clear the winner (static-index scan, see scratch note above)
-        #pragma unroll
-        for (int i = 0; i < 16; i++)
-            if (lane * per + i == bi) v[i] = -1.0f;
-    }
-    // normalize by the top-k sum (reference normTopk); single write pass
+    int gi[16];
+    float gw[16];
+    moe_gate_wave(logits + (int64_t)b * n_experts, n_experts, topk, lane, gi, gw);
     if (lane == 0) {
-        const float winv = 1.0f / wsum;
         #pragma unroll
         for (int t = 0; t < 16; t++) {
-            if (t < topk)
-                wts[(int64_t)b * topk + t] = chosen[t] * winv;
+            if (t < topk) {
+                idx[(int64_t)b * topk + t] = gi[t];
+                wts[(int64_t)b * topk + t] = gw[t];
+            }
         }
     }
 }
 
 // weighted sum of expert outputs + residual fold + ssq (reference OP_SCALE +
-// OP_MERGE_SUM + merge_add fused): x[b] += sum_s wts[b,s] * y[b*k+s]
+// OP_MERGE_SUM + merge_add fused): x[b] += sum_s wts[b,s] * y[b*k+s].
+// GATE=true recomputes the expert weights from the router logits in-kernel
+// (same deterministic moe_gate_wave as the grouped GEMVs).
+template <bool GATE = false>
 __global__ void k_scale_merge_add(float *__restrict__ x,
                                   const float *__restrict__ y,
                                   const float *__restrict__ wts,
                                   float *__restrict__ ssq,
-                                  int n, int topk) {
+                                  int n, int topk, int n_experts) {
     const int b = blockIdx.y;
+    float gw[16];
+    if constexpr (GATE) {
+        int gi[16];
+        moe_gate_wave(wts + (int64_t)b * n_experts, n_experts, topk,
+                      threadIdx.x % WAVE, gi, gw);
+    }
     float local = 0.0f;
     for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
          i += gridDim.x * blockDim.x) {
         float acc = x[(int64_t)b * n + i];
-        for (int s = 0; s < topk; s++)
-            acc = fmaf(wts[(int64_t)b * topk + s],
-                       y[((int64_t)b * topk + s) * n + i], acc);
+        if constexpr (GATE) {
+            #pragma unroll
+            for (int s = 0; s < 16; s++)
+                if (s < topk)
+                    acc = fmaf(gw[s], y[((int64_t)b * topk + s) * n + i], acc);
+        } else {
+            for (int s = 0; s < topk; s++)
+                acc = fmaf(wts[(int64_t)b * topk + s],
+                           y[((int64_t)b * topk + s) * n + i], acc);
+        }
         x[(int64_t)b * n + i] = acc;
         local += acc * acc;
     }
@@ -1355,17 +1408,31 @@ __global__ void k_scale_merge_add(float *__restrict__ x,
 // (no residual fold, no ssq — those happen after the Q80 sync merge-add;
 // replaces the eager torch.sum fallback so TP MoE decode stays
 // graph-capturable with zero ATen in the step)
+template <bool GATE = false>
 __global__ void k_scale_merge(float *__restrict__ partial,
                               const float *__restrict__ y,
                               const float *__restrict__ wts,
-                              int n, int topk) {
+                              int n, int topk, int n_experts) {
     const int b = blockIdx.y;
+    float gw[16];
+    if constexpr (GATE) {
+        int gi[16];
+        moe_gate_wave(wts + (int64_t)b * n_experts, n_experts, topk,
+                      threadIdx.x % WAVE, gi, gw);
+    }
     for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
          i += gridDim.x * blockDim.x) {
         float acc = 0.0f;
-        for (int s = 0; s < topk; s++)
-            acc = fmaf(wts[(int64_t)b * topk + s],
-                       y[((int64_t)b * topk + s) * n + i], acc);
+        if constexpr (GATE) {
+            #pragma unroll
+            for (int s = 0; s < 16; s++)
+                if (s < topk)
+                    acc = fmaf(gw[s], y[((int64_t)b * topk + s) * n + i], acc);
+        } else {
+            for (int s = 0; s < topk; s++)
+                acc = fmaf(wts[(int64_t)b * topk + s],
+                           y[((int64_t)b * topk + s) * n + i], acc);
+        }
         partial[(int64_t)b * n + i] = acc;
     }
 }
@@ -1628,6 +1695,156 @@ __global__ __launch_bounds__(1024) void k_q40_gemv_swiglu(
         oq[(int64_t)blockIdx.x * QB + threadIdx.x] = (int8_t)qf;
         const float bsum = group32_reduce_sum(qf);
         if (threadIdx.x == 0) { os[blockIdx.x] = dd; obs[blockIdx.x] = bsum; }
+    }
+}
+
+// grouped (MoE) variant of the fused W1|W3 GEMV + SwiGLU + Q80 emit, with
+// the gate computed in-kernel from the router logits: for decode (B=1) this
+// single launch replaces moe_gate + grouped w13 GEMV + swiglu_q80
+// (reference runs repeat_z, gate matmul, softmax, moe_gate, grouped w1/w3,
+// silu, mul, cast — llm.cpp:450-487). Workgroup = one Q80 block of one
+// expert slot's output.
+template <bool GELU>
+__global__ __launch_bounds__(1024) void k_q40_gemv_grouped_swiglu(
+        const uint8_t *__restrict__ qs, const __half *__restrict__ scales,
+        const int8_t *__restrict__ xq, const float *__restrict__ xs,
+        const float *__restrict__ xbs, int ff, int n,
+        const float *__restrict__ router, int n_experts, int topk,
+        int8_t *__restrict__ oq, float *__restrict__ os,
+        float *__restrict__ obs) {
+    const int wave = threadIdx.x / WAVE;  // 16 waves, 2 ff indices each
+    const int lane = threadIdx.x % WAVE;
+    const int slot = blockIdx.y;
+    const int br = slot / topk, want = slot - br * topk;
+    int gi[16];
+    float gw[16];
+    moe_gate_wave(router + (int64_t)br * n_experts, n_experts, topk, lane, gi, gw);
+    int e = 0;
+    #pragma unroll
+    for (int t = 0; t < 16; t++)
+        if (t == want) e = gi[t];
+    const int i0 = blockIdx.x * 32 + 2 * wave;
+    const int nb = n / QB, nbp = nb >> 1;
+    const int rows[4] = {i0, i0 + 1, ff + i0, ff + i0 + 1};
+    const uint4 *wrow[4];
+    const __half *srow[4];
+    #pragma unroll
+    for (int r = 0; r < 4; r++) {
+        const int64_t rbase = (int64_t)e * 2 * ff + rows[r];
+        wrow[r] = reinterpret_cast<const uint4 *>(qs + rbase * (n >> 1));
+        srow[r] = scales + rbase * nb;
+    }
+    const int8_t *xrow = xq + (int64_t)br * n;
+    const float *xsr = xs + (int64_t)br * nb;
+    const float *xbr = xbs + (int64_t)br * nb;
+    float acc[4] = {0.0f, 0.0f, 0.0f, 0.0f};
+    for (int jp = lane; jp < nbp; jp += WAVE) {
+        const int j = jp << 1;
+        const int4 *xr = reinterpret_cast<const int4 *>(xrow) + j * 2;
+        const int4 x0 = xr[0], x1 = xr[1], x2 = xr[2], x3 = xr[3];
+        const float2 sx = *reinterpret_cast<const float2 *>(xsr + j);
+        const float2 bsum = *reinterpret_cast<const float2 *>(xbr + j);
+        #pragma unroll
+        for (int r = 0; r < 4; r++) {
+            const uint4 w0 = wrow[r][j], w1 = wrow[r][j + 1];
+            const float2 sw = __half22float2(*reinterpret_cast<const __half2 *>(srow[r] + j));
+            acc[r] = fmaf(sw.x * sx.x, (float)q40_block_dot(w0, x0, x1) - 8.0f * bsum.x, acc[r]);
+            acc[r] = fmaf(sw.y * sx.y, (float)q40_block_dot(w1, x2, x3) - 8.0f * bsum.y, acc[r]);
+        }
+    }
+    if ((nb & 1) && lane == 0) {  // odd trailing block
+        const int j = nb - 1;
+        const int4 *xb = reinterpret_cast<const int4 *>(xrow) + j * 2;
+        const float sx1 = xsr[j], bs1 = xbr[j];
+        #pragma unroll
+        for (int r = 0; r < 4; r++)
+            acc[r] = fmaf(__half2float(srow[r][j]) * sx1,
+                          (float)q40_block_dot(wrow[r][j], xb[0], xb[1]) - 8.0f * bs1,
+                          acc[r]);
+    }
+    #pragma unroll
+    for (int r = 0; r < 4; r++) acc[r] = wave_reduce_sum(acc[r]);
+    __shared__ float sv[32];
+    if (lane == 0) {
+        #pragma unroll
+        for (int t = 0; t < 2; t++) {
+            const float av = acc[t], gv = acc[2 + t];
+            const float act = GELU
+                ? 0.5f * av * (1.0f + tanhf(0.797884560802865f * (av + 0.044715f * av * av * av)))
+                : av / (1.0f + __expf(-av));
+            sv[2 * wave + t] = act * gv;
+        }
+    }
+    __syncthreads();
+    if (threadIdx.x < 32) {
+        const float v = sv[threadIdx.x];
+        const float amax = group32_reduce_max(fabsf(v));
+        const float dd = amax / 127.0f;
+        const float qinv = dd > 0.0f ? 1.0f / dd : 0.0f;
+        const float qf = rintf(v * qinv);
+        oq[((int64_t)slot * ff) + blockIdx.x * QB + threadIdx.x] = (int8_t)qf;
+        const float bsum = group32_reduce_sum(qf);
+        if (threadIdx.x == 0) {
+            os[(int64_t)slot * (ff / QB) + blockIdx.x] = dd;
+            obs[(int64_t)slot * (ff / QB) + blockIdx.x] = bsum;
+        }
+    }
+}
+
+// Qwen3 attention prologue in one launch: per-head rmsnorm of the q/k heads
+// (reference OP_RMS_NORM nColumns mode, llm.cpp:178-187) + neox rope + KV
+// cache write (reference rope + shift ops). One workgroup (1 wave) per head:
+// q heads normalize+rotate in place, k heads normalize+rotate into the
+// cache, v heads copy to the cache.
+template <typename KVT>
+__global__ void k_rope_kv_qknorm(float *__restrict__ qkv, int ld,
+                                 int q_dim0, int kv_dim0,
+                                 const float *__restrict__ cache,
+                                 const int *__restrict__ pos,
+                                 KVT *__restrict__ kc, KVT *__restrict__ vc,
+                                 int hd,
+                                 const float *__restrict__ wq,
+                                 const float *__restrict__ wk, float eps) {
+    const int b = blockIdx.y;
+    const int p = pos[0] + b;
+    const int h = blockIdx.x;
+    const int qh = q_dim0 / hd, kvh = kv_dim0 / hd;
+    const int half = hd >> 1;
+    const float *pc = cache + (int64_t)p * hd;
+    const int lane = threadIdx.x;  // blockDim == 64
+    if (h >= qh + kvh) {
+        // v head: plain copy into the cache
+        const int hv = h - qh - kvh;
+        const float *src = qkv + (int64_t)b * ld + q_dim0 + kv_dim0 + hv * hd;
+        KVT *dst = vc + (int64_t)p * kv_dim0 + hv * hd;
+        for (int i = lane; i < hd; i += WAVE) dst[i] = kv_c<KVT>(src[i]);
+        return;
+    }
+    const bool is_q = h < qh;
+    float *row = qkv + (int64_t)b * ld + (is_q ? h * hd
+                                               : q_dim0 + (h - qh) * hd);
+    const float *w = is_q ? wq : wk;
+    float acc = 0.0f;
+    for (int i = lane; i < hd; i += WAVE) {
+        const float v = row[i];
+        acc += v * v;
+    }
+    acc = wave_reduce_sum(acc);
+    const float inv = rsqrtf(acc / hd + eps);
+    for (int j = lane; j < half; j += WAVE) {
+        const float v0 = row[j] * inv * w[j];
+        const float v1 = row[j + half] * inv * w[j + half];
+        const float cr = pc[2 * j], ci = pc[2 * j + 1];
+        const float o0 = v0 * cr - v1 * ci;
+        const float o1 = v0 * ci + v1 * cr;
+        if (is_q) {
+            row[j] = o0;
+            row[j + half] = o1;
+        } else {
+            KVT *dst = kc + (int64_t)p * kv_dim0 + (h - qh) * hd;
+            dst[j] = kv_c<KVT>(o0);
+            dst[j + half] = kv_c<KVT>(o1);
+        }
     }
 }
 
@@ -2012,11 +2229,17 @@ int64_t q40_gemv_argmax_blocks(int64_t d) {
 
 void q40_gemv_grouped(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
                       torch::Tensor xs, torch::Tensor xbs, torch::Tensor expert_idx,
-                      torch::Tensor y, int64_t k_slots, int64_t variant = -1) {
+                      torch::Tensor y, int64_t k_slots, int64_t variant = -1,
+                      c10::optional<torch::Tensor> router = c10::nullopt,
+                      int64_t topk = 0, int64_t n_slots_override = 0) {
+    // router given: expert ids come from the in-kernel gate over the router
+    // logits [B, n_experts] (expert_idx is then ignored); n_slots_override
+    // sets the slot count (it can no longer come from expert_idx.numel())
     CHECK_CUDA(qs); CHECK_CONT(qs);
     const int d = qs.size(1);
     const int n = qs.size(2) * 2;
-    const int n_slots = expert_idx.numel();
+    const int n_slots = n_slots_override > 0 ? (int)n_slots_override
+                                             : (int)expert_idx.numel();
     const int waves_per_block = 4;
     // lane-tiled (v2) default since round 2 (validated: 1.4-1.8x at
     // Qwen3-30B shapes); DLLAMA_MOE_V2=0 reverts to the 64-lane layout
@@ -2033,19 +2256,32 @@ void q40_gemv_grouped(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
     const int rows_per_wg = waves_per_block * 2 * (WAVE / lpp);
     const dim3 grid(ceil_div(d, rows_per_wg), n_slots);
     const dim3 block(waves_per_block * WAVE);
+    const bool gate = router.has_value();
+    const float *rp = gate ? router->data_ptr<float>() : nullptr;
+    const int ne = gate ? (int)router->size(-1) : 0;
     auto launch = [&](auto k) {
         hipLaunchKernelGGL(k, grid, block, 0, cur_stream(),
                            qs.data_ptr<uint8_t>(),
                            reinterpret_cast<const __half *>(scales.data_ptr<at::Half>()),
                            xq.data_ptr<int8_t>(), xs.data_ptr<float>(),
                            xbs.data_ptr<float>(), expert_idx.data_ptr<int>(),
-                           y.data_ptr<float>(), d, n, (int)k_slots);
+                           y.data_ptr<float>(), d, n, (int)k_slots,
+                           rp, ne, (int)topk);
     };
-    switch (lpp) {
-        case 8: launch(k_q40_gemv_grouped<8>); break;
-        case 16: launch(k_q40_gemv_grouped<16>); break;
-        case 32: launch(k_q40_gemv_grouped<32>); break;
-        default: launch(k_q40_gemv_grouped<64>); break;
+    if (gate) {
+        switch (lpp) {
+            case 8: launch(k_q40_gemv_grouped<8, true>); break;
+            case 16: launch(k_q40_gemv_grouped<16, true>); break;
+            case 32: launch(k_q40_gemv_grouped<32, true>); break;
+            default: launch(k_q40_gemv_grouped<64, true>); break;
+        }
+    } else {
+        switch (lpp) {
+            case 8: launch(k_q40_gemv_grouped<8>); break;
+            case 16: launch(k_q40_gemv_grouped<16>); break;
+            case 32: launch(k_q40_gemv_grouped<32>); break;
+            default: launch(k_q40_gemv_grouped<64>); break;
+        }
     }
 }
 
@@ -2199,21 +2435,36 @@ void moe_gate(torch::Tensor logits, torch::Tensor idx, torch::Tensor wts,
 }
 
 void scale_merge_add(torch::Tensor x, torch::Tensor y, torch::Tensor wts,
-                     torch::Tensor ssq, int64_t batch, int64_t topk) {
+                     torch::Tensor ssq, int64_t batch, int64_t topk,
+                     bool gate = false) {
+    // gate=true: wts is the [B, n_experts] router-logits buffer and the
+    // expert weights are recomputed in-kernel
     CHECK_CUDA(x);
     const int n = x.size(-1);
-    hipLaunchKernelGGL(k_scale_merge_add, dim3(ceil_div(n, 256), batch), dim3(256),
-                       0, cur_stream(), x.data_ptr<float>(), y.data_ptr<float>(),
-                       wts.data_ptr<float>(), ssq.data_ptr<float>(), n, (int)topk);
+    const int ne = gate ? (int)wts.size(-1) : 0;
+    auto launch = [&](auto k) {
+        hipLaunchKernelGGL(k, dim3(ceil_div(n, 256), batch), dim3(256),
+                           0, cur_stream(), x.data_ptr<float>(), y.data_ptr<float>(),
+                           wts.data_ptr<float>(), ssq.data_ptr<float>(), n,
+                           (int)topk, ne);
+    };
+    if (gate) launch(k_scale_merge_add<true>);
+    else launch(k_scale_merge_add<false>);
 }
 
 void scale_merge(torch::Tensor partial, torch::Tensor y, torch::Tensor wts,
-                 int64_t batch, int64_t topk) {
+                 int64_t batch, int64_t topk, bool gate = false) {
     CHECK_CUDA(partial);
     const int n = partial.size(-1);
-    hipLaunchKernelGGL(k_scale_merge, dim3(ceil_div(n, 256), batch), dim3(256),
-                       0, cur_stream(), partial.data_ptr<float>(),
-                       y.data_ptr<float>(), wts.data_ptr<float>(), n, (int)topk);
+    const int ne = gate ? (int)wts.size(-1) : 0;
+    auto launch = [&](auto k) {
+        hipLaunchKernelGGL(k, dim3(ceil_div(n, 256), batch), dim3(256),
+                           0, cur_stream(), partial.data_ptr<float>(),
+                           y.data_ptr<float>(), wts.data_ptr<float>(), n,
+                           (int)topk, ne);
+    };
+    if (gate) launch(k_scale_merge<true>);
+    else launch(k_scale_merge<false>);
 }
 
 void logits_concat(torch::Tensor dst, torch::Tensor src, int64_t batch) {
@@ -2392,6 +2643,59 @@ void q40_gemv_swiglu(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
     };
     if (gelu) launch(k_q40_gemv_swiglu<true>);
     else launch(k_q40_gemv_swiglu<false>);
+}
+
+void q40_gemv_grouped_swiglu(torch::Tensor qs, torch::Tensor scales,
+                             torch::Tensor xq, torch::Tensor xs,
+                             torch::Tensor xbs, torch::Tensor oq,
+                             torch::Tensor os, torch::Tensor obs,
+                             int64_t n_slots, torch::Tensor router,
+                             int64_t topk, bool gelu = false) {
+    // decode MoE FFN up-projection: gate + grouped W1|W3 GEMV + SwiGLU +
+    // Q80 emit in one launch. qs [E, 2*ff, n/2]; oq/os/obs are the
+    // [n_slots, ff] Q80 triple.
+    CHECK_CUDA(qs); CHECK_CONT(qs); CHECK_CONT(xq);
+    const int ff = (int)qs.size(1) / 2;
+    const int n = (int)qs.size(2) * 2;
+    TORCH_CHECK(ff % 32 == 0, "fused grouped swiglu needs ff % 32 == 0");
+    const dim3 grid(ff / 32, n_slots);
+    const dim3 block(16 * WAVE);
+    auto launch = [&](auto k) {
+        hipLaunchKernelGGL(k, grid, block, 0, cur_stream(),
+                           qs.data_ptr<uint8_t>(),
+                           reinterpret_cast<const __half *>(scales.data_ptr<at::Half>()),
+                           xq.data_ptr<int8_t>(), xs.data_ptr<float>(),
+                           xbs.data_ptr<float>(), ff, n,
+                           router.data_ptr<float>(), (int)router.size(-1),
+                           (int)topk, oq.data_ptr<int8_t>(),
+                           os.data_ptr<float>(), obs.data_ptr<float>());
+    };
+    if (gelu) launch(k_q40_gemv_grouped_swiglu<true>);
+    else launch(k_q40_gemv_grouped_swiglu<false>);
+}
+
+void rope_kv_qknorm(torch::Tensor qkv, int64_t ld, int64_t q_dim0,
+                    int64_t kv_dim0, torch::Tensor cache, torch::Tensor pos,
+                    torch::Tensor kc, torch::Tensor vc, int64_t head_dim,
+                    torch::Tensor wq, torch::Tensor wk, double eps,
+                    int64_t batch) {
+    CHECK_CUDA(qkv);
+    const int heads = (int)((q_dim0 + 2 * kv_dim0) / head_dim);
+    const dim3 grid(heads, batch);
+    auto run = [&](auto kvt, auto kcp, auto vcp) {
+        using KVT = decltype(kvt);
+        hipLaunchKernelGGL((k_rope_kv_qknorm<KVT>), grid, dim3(WAVE), 0,
+                           cur_stream(), qkv.data_ptr<float>(), (int)ld,
+                           (int)q_dim0, (int)kv_dim0, cache.data_ptr<float>(),
+                           pos.data_ptr<int>(), kcp, vcp, (int)head_dim,
+                           wq.data_ptr<float>(), wk.data_ptr<float>(),
+                           (float)eps);
+    };
+    if (kc.scalar_type() == at::kHalf)
+        run(__half{}, reinterpret_cast<__half *>(kc.data_ptr<at::Half>()),
+            reinterpret_cast<__half *>(vc.data_ptr<at::Half>()));
+    else
+        run(0.0f, kc.data_ptr<float>(), vc.data_ptr<float>());
 }
 
 void silu_mul(torch::Tensor a, torch::Tensor g, torch::Tensor out) {
@@ -2580,8 +2884,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("q"), py::arg("s"), py::arg("bs"), py::arg("batch"),
           py::arg("eps"), py::arg("yout") = py::none());
     m.def("moe_gate", &moe_gate);
-    m.def("scale_merge_add", &scale_merge_add);
-    m.def("scale_merge", &scale_merge);
+    m.def("scale_merge_add", &scale_merge_add, py::arg("x"), py::arg("y"),
+          py::arg("wts"), py::arg("ssq"), py::arg("batch"), py::arg("topk"),
+          py::arg("gate") = false);
+    m.def("scale_merge", &scale_merge, py::arg("partial"), py::arg("y"),
+          py::arg("wts"), py::arg("batch"), py::arg("topk"),
+          py::arg("gate") = false);
     m.def("logits_concat", &logits_concat);
     m.def("argmax_token", &argmax_token);
     m.def("router_gemv", &router_gemv);
@@ -2590,7 +2898,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("q40_gemv_grouped", &q40_gemv_grouped, py::arg("qs"),
           py::arg("scales"), py::arg("xq"), py::arg("xs"), py::arg("xbs"),
           py::arg("expert_idx"), py::arg("y"), py::arg("k_slots"),
-          py::arg("variant") = -1);
+          py::arg("variant") = -1, py::arg("router") = py::none(),
+          py::arg("topk") = 0, py::arg("n_slots") = 0);
+    m.def("q40_gemv_grouped_swiglu", &q40_gemv_grouped_swiglu, py::arg("qs"),
+          py::arg("scales"), py::arg("xq"), py::arg("xs"), py::arg("xbs"),
+          py::arg("oq"), py::arg("os"), py::arg("obs"), py::arg("n_slots"),
+          py::arg("router"), py::arg("topk"), py::arg("gelu") = false);
+    m.def("rope_kv_qknorm", &rope_kv_qknorm);
     m.def("rope", &rope);
     m.def("rope_kv", &rope_kv);
     m.def("kv_append", &kv_append);

@@ -115,27 +115,26 @@ def test_hip_prefill_gemm_path(tiny_path):
 
 
 def test_adaptive_splits_recapture(tiny_path):
-    """Adaptive K-split schedule: the first decode below pos 256 re-captures
-    the graph with the fused S=1 attention; logits must keep matching the
-    eager (S=8 two-kernel) path."""
+    """Crossing the adaptive threshold re-captures the decode graph with
+    S=16; logits must keep matching eager decode at S=8."""
     from dllama_amd.models.hip_model import HipTransformer
     m = mf.ModelFile(tiny_path)
     cfg = ModelConfig.from_header(m.header)
     eager = HipTransformer.from_file(m, cfg)
     eager.adaptive_thresh = 0      # pin eager at the default S=8 pair
     adaptive = HipTransformer.from_file(m, cfg)
-    assert adaptive.adaptive_thresh
+    adaptive.adaptive_thresh = 6
     prompt = [1, 2, 3]
     eager.forward(torch.tensor(prompt), torch.arange(3))
     adaptive.forward(torch.tensor(prompt), torch.arange(3))
     adaptive.capture_decode_graph()
-    for step in range(8):
+    for step in range(8):  # crosses the threshold at pos 6
         t = torch.tensor([20 + step])
         p = torch.tensor([3 + step])
         want = eager.forward(t, p).cpu().clone()
         got = adaptive.forward(t, p).cpu().clone()
         assert _rel_err(got[0], want[0]) < 1e-4, f"step {step}"
-    assert adaptive.attn_splits == 1  # short-context fused single kernel
+    assert adaptive.attn_splits == 16
 
 
 def test_tp_path_matches_plain_world1(tiny_path):

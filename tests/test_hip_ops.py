@@ -576,3 +576,86 @@ def test_q40_gemv_swiglu_fused(k):
     tol = want.abs().max().item() / 100 + 1e-6
     assert torch.allclose(got, want, atol=tol), (got - want).abs().max().item()
     assert torch.allclose(fbs.cpu(), wbs.cpu(), atol=2.0)
+
+
+def test_rope_kv_qknorm_fused(k):
+    """Fused per-head q/k rmsnorm + neox rope + KV write vs the 3-kernel
+    path (rmsnorm_rows_s x2 + rope_kv style=1)."""
+    B, hd, qh, kvh, seq = 2, 64, 4, 2, 32
+    q_dim0, kv_dim0 = qh * hd, kvh * hd
+    ld = q_dim0 + 2 * kv_dim0
+    buf = rand(B, ld, seed=400)
+    buf2 = buf.clone()
+    wq = rand(hd, seed=401).abs() + 0.5
+    wk = rand(hd, seed=402).abs() + 0.5
+    cache = R.rope_cache(seq, hd, 10000.0).to(DEV).reshape(seq, hd).contiguous()
+    pos = torch.tensor([5], dtype=torch.int32, device=DEV)
+    eps = 1e-6
+    kc1 = torch.zeros(seq, kv_dim0, dtype=torch.float16, device=DEV)
+    vc1 = torch.zeros(seq, kv_dim0, dtype=torch.float16, device=DEV)
+    k.rmsnorm_rows_s(buf, ld, 0, qh, B, wq, hd, eps)
+    k.rmsnorm_rows_s(buf, ld, q_dim0, kvh, B, wk, hd, eps)
+    k.rope_kv(buf, ld, q_dim0, kv_dim0, cache, pos, kc1, vc1, hd, 1, B)
+    kc2 = torch.zeros(seq, kv_dim0, dtype=torch.float16, device=DEV)
+    vc2 = torch.zeros(seq, kv_dim0, dtype=torch.float16, device=DEV)
+    k.rope_kv_qknorm(buf2, ld, q_dim0, kv_dim0, cache, pos, kc2, vc2, hd,
+                     wq, wk, eps, B)
+    assert torch.allclose(buf2[:, :q_dim0], buf[:, :q_dim0], atol=1e-5), \
+        (buf2[:, :q_dim0] - buf[:, :q_dim0]).abs().max().item()
+    assert torch.allclose(kc2.float(), kc1.float(), atol=1e-3)
+    assert torch.allclose(vc2.float(), vc1.float(), atol=1e-3)
+
+
+def test_moe_gate_fused_consumers(k):
+    """Gate-fused grouped GEMV / grouped swiglu / scale_merge_add must agree
+    with the explicit moe_gate + unfused kernels."""
+    E, B, ka, ff, n = 16, 2, 4, 64, 256
+    S = B * ka
+    gen = torch.Generator(device=DEV).manual_seed(500)
+    w13qs = torch.randint(0, 256, (E, 2 * ff, n // 2), dtype=torch.uint8,
+                          device=DEV, generator=gen)
+    w13sc = (torch.rand((E, 2 * ff, n // 32), device=DEV, generator=gen)
+             * 0.01).to(torch.float16)
+    router = rand(B, E, seed=501)
+    x = rand(B, n, seed=502, scale=0.5)
+    q = torch.zeros(B, n, dtype=torch.int8, device=DEV)
+    s = torch.zeros(B, n // 32, device=DEV)
+    bs = torch.zeros(B, n // 32, device=DEV)
+    k.q80_quantize(x, q, s, bs)
+    # reference: explicit gate + grouped + swiglu
+    idx = torch.zeros(S, dtype=torch.int32, device=DEV)
+    wts = torch.zeros(B, ka, device=DEV)
+    k.moe_gate(router, idx, wts, B, ka)
+    y13 = torch.zeros(S, 2 * ff, device=DEV)
+    k.q40_gemv_grouped(w13qs, w13sc, q, s, bs, idx, y13, ka)
+    dq = torch.zeros(S, ff, dtype=torch.int8, device=DEV)
+    ds = torch.zeros(S, ff // 32, device=DEV)
+    dbs = torch.zeros(S, ff // 32, device=DEV)
+    k.swiglu_q80(y13, y13[:, ff:], 2 * ff, ff, S, dq, ds, dbs)
+    # fused
+    fq = torch.zeros(S, ff, dtype=torch.int8, device=DEV)
+    fs = torch.zeros(S, ff // 32, device=DEV)
+    fbs = torch.zeros(S, ff // 32, device=DEV)
+    k.q40_gemv_grouped_swiglu(w13qs, w13sc, q, s, bs, fq, fs, fbs, S, router, ka)
+    got = R.q80_dequantize(fq.cpu(), fs.cpu())
+    want = R.q80_dequantize(dq.cpu(), ds.cpu())
+    tol = want.abs().max().item() / 100 + 1e-6
+    assert torch.allclose(got, want, atol=tol), (got - want).abs().max().item()
+    # gate-fused w2 grouped GEMV vs explicit-idx
+    w2qs = torch.randint(0, 256, (E, n, ff // 2), dtype=torch.uint8,
+                         device=DEV, generator=gen)
+    w2sc = (torch.rand((E, n, ff // 32), device=DEV, generator=gen)
+            * 0.01).to(torch.float16)
+    y2a = torch.zeros(S, n, device=DEV)
+    y2b = torch.zeros(S, n, device=DEV)
+    k.q40_gemv_grouped(w2qs, w2sc, dq, ds, dbs, idx, y2a, 1)
+    k.q40_gemv_grouped(w2qs, w2sc, dq, ds, dbs, idx, y2b, 1,
+                       router=router, topk=ka, n_slots=S)
+    assert torch.allclose(y2b, y2a, atol=1e-5)
+    # gate-fused scale_merge_add vs explicit wts
+    xa = rand(B, n, seed=503)
+    xb = xa.clone()
+    ssq = torch.zeros(2, B, 16 * 32, device=DEV)
+    k.scale_merge_add(xa, y2a, wts, ssq[0], B, ka)
+    k.scale_merge_add(xb, y2a, router, ssq[1], B, ka, gate=True)
+    assert torch.allclose(xb, xa, atol=1e-5)
