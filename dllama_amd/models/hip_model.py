@@ -245,6 +245,11 @@ class HipTransformer:
         # is recaptured when pos crosses a boundary. DLLAMA_ADAPTIVE_SPLITS=0
         # (or an explicit DLLAMA_ATTN_SPLITS) pins S.
         self.adaptive_thresh = int(_os.environ.get("DLLAMA_ADAPTIVE_SPLITS", "512"))
+        # measured-slower fusions kept for shape experiments (profiles r02):
+        # the 16-wave fused FFN streams worse than the 4-wave GEMV + swiglu
+        # pair, and per-wave gate recompute adds ~8 us to each MoE consumer
+        self.fused_ffn = _os.environ.get("DLLAMA_FUSED_FFN", "0") == "1"
+        self.fused_moe = _os.environ.get("DLLAMA_FUSED_MOE", "0") == "1"
         if ("DLLAMA_ATTN_SPLITS" in _os.environ
                 and "DLLAMA_ADAPTIVE_SPLITS" not in _os.environ):
             self.adaptive_thresh = 0
@@ -423,7 +428,8 @@ class HipTransformer:
                 self._moe_ffn(B, NB, lw, slot + 1)
             else:
                 gelu = c.hidden_act == HIDDEN_ACT_GELU
-                if NB == 1 and not fused_norm and c.ff_dim0 % 32 == 0:
+                if (NB == 1 and not fused_norm and c.ff_dim0 % 32 == 0
+                        and getattr(self, "fused_ffn", False)):
                     # fused W1|W3 GEMV + SwiGLU + Q80 emit (one launch
                     # replacing gemv + swiglu_q80)
                     k.norm_quant(x[:NB], lw["norm1"], self.ssq[slot],
@@ -467,7 +473,8 @@ class HipTransformer:
         ka = c.n_active_experts
         S = NB * ka
         k.router_gemv(lw["gate"], self.t_norm, self.moe_router, NB)
-        if c.ff_dim0 % 32 == 0 and ka <= 16:
+        if (c.ff_dim0 % 32 == 0 and ka <= 16
+                and getattr(self, "fused_moe", False)):
             # gate-fused decode path: every consumer recomputes the
             # deterministic top-k from the router logits in-kernel, so the
             # FFN is 3 launches (w13+swiglu, w2, scale-merge) instead of 6
