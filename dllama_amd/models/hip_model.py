@@ -354,6 +354,11 @@ class HipTransformer:
         stays on device — this function is graph-capturable."""
         c, k = self.cfg, self.k
         NB = _pow2_batch(B)
+        if (B == 1 and not c.is_moe and not self.tp_path
+                and c.dim % 32 == 0
+                and getattr(self, "use_deferred", True)
+                and not getattr(self, "use_fused_norm", False)):
+            return self._forward_dense_deferred()
         x = self.x
         self.ssq.zero_()
         k.embed_gather(self.embedding, self.tokens, x, NB, self.ssq[0])
@@ -464,6 +469,90 @@ class HipTransformer:
                 # logits row; two-kernel argmax, no ATen in the graph
                 k.argmax_token(self.tokens, self.logits_gather[1].view(-1),
                                self.argmax_scratch_full)
+
+    def _forward_dense_deferred(self):
+        """B=1 dense decode with DEFERRED activation quantization: the
+        down-projection GEMVs emit the next matmul's Q80 input in their own
+        epilogue (x*w_norm quantized per wg-local block, scale excluding
+        inv_rms), and every consumer GEMV applies inv = rsqrt(ssq/n + eps)
+        as one multiply per output row. This removes both norm_quant
+        launches per layer — the Q80 codes are scale-invariant, so numerics
+        match the explicit-norm path to fp rounding (reference runs
+        merge_add/inv_rms/rms_norm/cast as 4 ops per half-layer,
+        llm.cpp:263-270)."""
+        c, k = self.cfg, self.k
+        x = self.x
+        eps = c.norm_eps
+        self.ssq.zero_()
+        k.embed_gather(self.embedding, self.tokens, x, 1, self.ssq[0])
+        kv_mul = c.n_heads0 // max(1, c.kv_dim0 // c.head_dim)
+        fused_rope = self.rope_style == 0 and not c.is_qwen3
+        # layer 0's norm0 quant comes from a deferred norm_quant (later
+        # layers get it from the previous w2's EPI_RESID_Q epilogue)
+        k.norm_quant(x[:1], self.layers[0]["norm0"], self.ssq[0],
+                     self.xq.q[:1], self.xq.s[:1], self.xq.bs[:1], 1, eps,
+                     deferred=True)
+        slot = 0
+        last = len(self.layers) - 1
+        for l, lw in enumerate(self.layers):
+            sin = self.ssq[slot]
+            if fused_rope:
+                k.q40_gemv_rope(lw["qkv"].qs, lw["qkv"].scales, self.xq.q,
+                                self.xq.s, self.xq.bs, self.qkv_out, 1,
+                                self.rope_cache, self.pos, self.k_cache[l],
+                                self.v_cache[l], c.q_dim0, c.kv_dim0,
+                                c.head_dim, ssq_in=sin, eps=eps)
+            else:
+                k.q40_gemv(lw["qkv"].qs, lw["qkv"].scales, self.xq.q,
+                           self.xq.s, self.xq.bs, self.qkv_out, 1,
+                           ssq_in=sin, eps=eps)
+                if c.is_qwen3 and self.rope_style == 1:
+                    k.rope_kv_qknorm(self.qkv_out, self.qkv_ld, c.q_dim0,
+                                     c.kv_dim0, self.rope_cache, self.pos,
+                                     self.k_cache[l], self.v_cache[l],
+                                     c.head_dim, lw["q_norm"], lw["k_norm"],
+                                     eps, 1)
+                else:
+                    if c.is_qwen3:
+                        k.rmsnorm_rows_s(self.qkv_out, self.qkv_ld, 0,
+                                         c.q_dim0 // c.head_dim, 1,
+                                         lw["q_norm"], c.head_dim, eps)
+                        k.rmsnorm_rows_s(self.qkv_out, self.qkv_ld, c.q_dim0,
+                                         c.kv_dim0 // c.head_dim, 1,
+                                         lw["k_norm"], c.head_dim, eps)
+                    k.rope_kv(self.qkv_out, self.qkv_ld, c.q_dim0, c.kv_dim0,
+                              self.rope_cache, self.pos, self.k_cache[l],
+                              self.v_cache[l], c.head_dim, self.rope_style, 1)
+            k.attn(self.qkv_out, self.qkv_ld, self.k_cache[l], self.v_cache[l],
+                   self.zbuf[:1], self.pos, 1, c.n_heads0, kv_mul, c.head_dim,
+                   self.attn_splits, self.attn_ml, self.attn_o,
+                   self.attn_counter, self.zq.q, self.zq.s, self.zq.bs)
+            # wo: residual fold + deferred Q80 emit of x*norm1 for w13
+            k.q40_gemv_resid_q(lw["wo"].qs, lw["wo"].scales, self.zq.q,
+                               self.zq.s, self.zq.bs, x, self.ssq[slot + 1],
+                               lw["norm1"], self.xq.q, self.xq.s, self.xq.bs)
+            slot += 1
+            k.q40_gemv(lw["w13"].qs, lw["w13"].scales, self.xq.q, self.xq.s,
+                       self.xq.bs, self.ff_out, 1, ssq_in=self.ssq[slot],
+                       eps=eps)
+            k.swiglu_q80(self.ff_out, self.ff_out[:, c.ff_dim0:],
+                         2 * c.ff_dim0, c.ff_dim0, 1, self.dq.q[:1],
+                         self.dq.s[:1], self.dq.bs[:1],
+                         c.hidden_act == HIDDEN_ACT_GELU)
+            # w2: residual fold + deferred emit for the NEXT layer's norm0
+            # (final_norm for the logits matmul after the last layer)
+            wn = self.final_norm if l == last else self.layers[l + 1]["norm0"]
+            k.q40_gemv_resid_q(lw["w2"].qs, lw["w2"].scales, self.dq.q,
+                               self.dq.s, self.dq.bs, x, self.ssq[slot + 1],
+                               wn, self.xq.q, self.xq.s, self.xq.bs)
+            slot += 1
+        use_amax = self.greedy_feedback
+        k.q40_gemv(self.wcls.qs, self.wcls.scales, self.xq.q, self.xq.s,
+                   self.xq.bs, self.logits0, 1,
+                   self.amax_scratch if use_amax else None,
+                   ssq_in=self.ssq[slot], eps=eps)
+        if use_amax:
+            k.token_from_argmax(self.tokens, self.amax_scratch, self.amax_blocks)
 
     def _moe_ffn(self, B: int, NB: int, lw: dict, slot: int):
         """Router + grouped expert GEMVs (reference llm.cpp:450-487);

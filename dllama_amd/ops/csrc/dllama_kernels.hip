@@ -392,9 +392,12 @@ __global__ void k_norm_quant(const float *__restrict__ x,
                              float *__restrict__ s,
                              float *__restrict__ bs,
                              float *__restrict__ yout,
-                             int n, float eps) {
+                             int n, float eps, int deferred) {
     const int b = blockIdx.y;
-    const float inv = rsqrtf(ssq_total(ssq, b) / n + eps);
+    // deferred: emit x*w with the inv_rms factored out of the scale — the
+    // consuming GEMV applies inv from ssq (PRO==2); codes are identical
+    // either way (q depends only on intra-block ratios)
+    const float inv = deferred ? 1.0f : rsqrtf(ssq_total(ssq, b) / n + eps);
     const int nb = n / QB;
     const int gid = blockIdx.x * blockDim.x + threadIdx.x;
     const int blk = gid / 32;
@@ -523,6 +526,8 @@ __device__ __forceinline__ int q40_block_dot(const uint4 wq, const int4 x0,
 // removes whole kernels from the per-layer chain (the reference runs each
 // as its own op, llm.cpp:263-557).
 #define EPI_NONE 0
+#define EPI_RESID_Q 3  // EPI_RESID + deferred-scale Q80 emit of x*wnorm (one
+                       // block per 16-wave wg; consumer applies inv_rms, PRO==2)
 #define EPI_RESID 1  // x[b,row] += v; accumulate sum(x'^2) into 16-way-spread
                      //  slots -> the next norm becomes a single wide pass
 #define EPI_ROPE 2   // llama rope: rows (2j,2j+1) are a rotation pair held by
@@ -598,7 +603,10 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
                            const float *__restrict__ xf,
                            const float *__restrict__ wnorm,
                            const float *__restrict__ ssq_in,
-                           float eps, int kv_f16) {
+                           float eps, int kv_f16,
+                           int8_t *__restrict__ oqq,
+                           float *__restrict__ oqs,
+                           float *__restrict__ oqbs) {
     const int wpb = blockDim.x / WAVE;
     const int wid = threadIdx.x / WAVE;
     const int row0 = (blockIdx.x * wpb + wid) * RPW;
@@ -607,6 +615,7 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
     const int nbp = nb >> 1;
     const bool wave_valid = row0 < d;
     const int rbase = wave_valid ? row0 : 0;
+    [[maybe_unused]] __shared__ float svq[32];  // EPI_RESID_Q block staging
 
     const uint4 *wrow[RPW];
     const __half *srow[RPW];
@@ -644,7 +653,7 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
         for (int b = 0; b < NB; b++) {
             int4 x0, x1, x2, x3;
             float2 sx, bsum;
-            if (PRO) {
+            if (PRO == 1) {
                 q80_quant_block(xf + (int64_t)b * n + j * QB, wnorm + j * QB,
                                 invb[b], &x0, &x1, &sx.x, &bsum.x);
                 q80_quant_block(xf + (int64_t)b * n + (j + 1) * QB,
@@ -675,7 +684,7 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
             for (int b = 0; b < NB; b++) {
                 int4 xb0, xb1;
                 float sx1, bs1;
-                if (PRO) {
+                if (PRO == 1) {
                     q80_quant_block(xf + (int64_t)b * n + j * QB, wnorm + j * QB,
                                     invb[b], &xb0, &xb1, &sx1, &bs1);
                 } else {
@@ -700,7 +709,10 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
     for (int b = 0; b < NB; b++) {
         float v[RPW];
         #pragma unroll
-        for (int r = 0; r < RPW; r++) v[r] = wave_reduce_sum(acc[r][b]);
+        for (int r = 0; r < RPW; r++) {
+            v[r] = wave_reduce_sum(acc[r][b]);
+            if (PRO == 2) v[r] *= invb[b];  // deferred inv_rms (scalar/row)
+        }
         if (lane != 0 || !wave_valid) continue;
         if (EPI == EPI_ROPE) {
             // RPW==2: rows (rbase, rbase+1) form one llama rotation pair
@@ -749,6 +761,13 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
                     const float xv = x_resid[(int64_t)b * d + row] + v[r];
                     x_resid[(int64_t)b * d + row] = xv;
                     ssq_local[b] += xv * xv;
+                } else if (EPI == EPI_RESID_Q) {
+                    // residual fold + stage the weighted value for the
+                    // wg-local Q80 block (rows of this wg = one block)
+                    const float xv = x_resid[(int64_t)b * d + row] + v[r];
+                    x_resid[(int64_t)b * d + row] = xv;
+                    ssq_local[b] += xv * xv;
+                    svq[wid * RPW + r] = xv * wnorm[row];
                 } else {
                     y[(int64_t)b * d + row] = v[r];
                     if (NB == 1) wave_best = max(wave_best, argmax_pack(v[r], row));
@@ -765,6 +784,29 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
             unsigned long long best = wb[0];
             for (int i = 1; i < wpb; i++) best = max(best, wb[i]);
             amax_scratch[blockIdx.x] = best;
+        }
+    }
+    if constexpr (EPI == EPI_RESID_Q) {
+        __shared__ float sredq[16];
+        if (lane == 0) sredq[wid] = wave_valid ? ssq_local[0] : 0.0f;
+        __syncthreads();  // also publishes svq
+        if (threadIdx.x == 0) {
+            float t = 0.0f;
+            for (int wv = 0; wv < wpb; wv++) t += sredq[wv];
+            atomicAdd(ssq + (blockIdx.x & (SSQ_SPREAD - 1)) * SSQ_PAD, t);
+        }
+        if (threadIdx.x < 32 && (int)(blockIdx.x * 32 + threadIdx.x) < d) {
+            const float vq = svq[threadIdx.x];
+            const float amax = group32_reduce_max(fabsf(vq));
+            const float dd = amax / 127.0f;
+            const float qinv = dd > 0.0f ? 1.0f / dd : 0.0f;
+            const float qf = rintf(vq * qinv);
+            oqq[blockIdx.x * QB + threadIdx.x] = (int8_t)qf;
+            const float bsq = group32_reduce_sum(qf);
+            if (threadIdx.x == 0) {
+                oqs[blockIdx.x] = dd;
+                oqbs[blockIdx.x] = bsq;
+            }
         }
     }
     if (EPI == EPI_RESID) {
@@ -1338,24 +1380,58 @@ __global__ void k_router_gemv(const float *__restrict__ gate,
 // MoE router: softmax over n_experts logits, top-k (first-index ties),
 // normalized weights + int32 expert ids (reference OP_SOFTMAX + OP_MOE_GATE,
 // nn-cpu-ops.cpp:1443-1492). One wave per batch row; n_experts <= 1024.
+// rank-select top-k (replaces the iterative wave-argmax version): expert
+// e's output slot is its rank = #{j: p_j > p_e} + #{j < e: p_j == p_e}
+// (identical ordering and smallest-index tie-break as the reference's
+// insertion sort, nn-cpu-ops.cpp:900-916). All-pairs comparison is pure
+// parallel VALU over LDS — the old 8 serial rounds of 6-step wave argmax
+// were shuffle-latency-bound (~6 us wall for one wave; measured
+// tools/fuse_shape_probe G-vs-F).
 __global__ void k_moe_gate(const float *__restrict__ logits,
                            int *__restrict__ idx,
                            float *__restrict__ wts,
                            int n_experts, int topk) {
     const int b = blockIdx.x;
-    const int lane = threadIdx.x;  // blockDim == 64
-    int gi[16];
-    float gw[16];
-    moe_gate_wave(logits + (int64_t)b * n_experts, n_experts, topk, lane, gi, gw);
-    if (lane == 0) {
-        #pragma unroll
-        for (int t = 0; t < 16; t++) {
-            if (t < topk) {
-                idx[(int64_t)b * topk + t] = gi[t];
-                wts[(int64_t)b * topk + t] = gw[t];
-            }
+    const float *lrow = logits + (int64_t)b * n_experts;
+    __shared__ float p[1024];
+    __shared__ float red[16];
+    __shared__ float topsum;
+    const int t = threadIdx.x;           // blockDim == 128
+    const int wid = t / WAVE, lane = t % WAVE;
+    float m = -1e30f;
+    for (int i = t; i < n_experts; i += blockDim.x) {
+        const float v = lrow[i];
+        p[i] = v;
+        m = fmaxf(m, v);
+    }
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        m = fmaxf(m, __shfl_xor(m, off, WAVE));
+    if (lane == 0) red[wid] = m;
+    __syncthreads();
+    m = fmaxf(red[0], red[1]);
+    // softmax numerator only: the full-softmax denominator cancels in the
+    // top-k normalization (w_t = p_t / sum_topk p)
+    for (int i = t; i < n_experts; i += blockDim.x)
+        p[i] = __expf(p[i] - m);
+    if (t == 0) topsum = 0.0f;
+    __syncthreads();
+    for (int i = t; i < n_experts; i += blockDim.x) {
+        const float mine = p[i];
+        int rank = 0;
+        for (int j = 0; j < n_experts; j++) {
+            const float o = p[j];
+            rank += (o > mine) || (o == mine && j < i);
+        }
+        if (rank < topk) {
+            idx[(int64_t)b * topk + rank] = i;
+            wts[(int64_t)b * topk + rank] = mine;  // normalized below
+            atomicAdd(&topsum, mine);
         }
     }
+    __syncthreads();
+    if (t < topk)
+        wts[(int64_t)b * topk + t] /= topsum;
 }
 
 // weighted sum of expert outputs + residual fold + ssq (reference OP_SCALE +
@@ -2020,6 +2096,10 @@ struct GemvEpi {
     const float *wnorm = nullptr;
     const float *ssq_in = nullptr;
     float eps = 0.0f;
+    // EPI_RESID_Q deferred-quant outputs
+    int8_t *oqq = nullptr;
+    float *oqs = nullptr;
+    float *oqbs = nullptr;
 };
 
 template <int EPI, int PRO>
@@ -2030,7 +2110,8 @@ static void gemv_launch(torch::Tensor &qs, torch::Tensor &scales, torch::Tensor 
     const int n = qs.size(1) * 2;
     TORCH_CHECK(PRO == 1 || xq.size(-1) == n, "x width mismatch");
     TORCH_CHECK(n % QB == 0, "n must be a multiple of 32");
-    const int waves_per_block = 4;
+    // EPI_RESID_Q: 16 waves x RPW2 = 32 output rows per wg = one Q80 block
+    const int waves_per_block = EPI == EPI_RESID_Q ? 16 : 4;
     const dim3 block(waves_per_block * WAVE);
     auto launch = [&](auto nb_const, auto rpw_const) {
         constexpr int RPW = decltype(rpw_const)::value;
@@ -2039,12 +2120,13 @@ static void gemv_launch(torch::Tensor &qs, torch::Tensor &scales, torch::Tensor 
                            block, 0, cur_stream(),
                            qs.data_ptr<uint8_t>(),
                            reinterpret_cast<const __half *>(scales.data_ptr<at::Half>()),
-                           PRO ? nullptr : xq.data_ptr<int8_t>(),
-                           PRO ? nullptr : xs.data_ptr<float>(),
-                           PRO ? nullptr : xbs.data_ptr<float>(), y, d, n,
+                           PRO == 1 ? nullptr : xq.data_ptr<int8_t>(),
+                           PRO == 1 ? nullptr : xs.data_ptr<float>(),
+                           PRO == 1 ? nullptr : xbs.data_ptr<float>(), y, d, n,
                            e.slot, e.x_resid, e.ssq,
                            e.cache, e.pos, e.kc, e.vc, e.q_dim0, e.kv_dim0, e.hd,
-                           e.xf, e.wnorm, e.ssq_in, e.eps, e.kv_f16);
+                           e.xf, e.wnorm, e.ssq_in, e.eps, e.kv_f16,
+                           e.oqq, e.oqs, e.oqbs);
     };
     std::integral_constant<int, 1> r1;
     std::integral_constant<int, 2> r2;
@@ -2105,12 +2187,43 @@ static void gemv_launch(torch::Tensor &qs, torch::Tensor &scales, torch::Tensor 
 
 void q40_gemv(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
               torch::Tensor xs, torch::Tensor xbs, torch::Tensor y, int64_t batch,
-              c10::optional<torch::Tensor> amax_slot = c10::nullopt) {
+              c10::optional<torch::Tensor> amax_slot = c10::nullopt,
+              c10::optional<torch::Tensor> ssq_in = c10::nullopt,
+              double eps = 0.0) {
+    // ssq_in given: the xq triple is DEFERRED (scales exclude inv_rms, see
+    // EPI_RESID_Q); the kernel applies inv = rsqrt(ssq/n + eps) per row
     CHECK_CUDA(qs); CHECK_CONT(qs); CHECK_CONT(xq);
     GemvEpi e;
     if (amax_slot.has_value())
         e.slot = reinterpret_cast<unsigned long long *>(amax_slot->data_ptr<int64_t>());
-    gemv_launch<EPI_NONE, 0>(qs, scales, xq, xs, xbs, y.data_ptr<float>(), batch, e);
+    if (ssq_in.has_value()) {
+        e.ssq_in = ssq_in->data_ptr<float>();
+        e.eps = (float)eps;
+        gemv_launch<EPI_NONE, 2>(qs, scales, xq, xs, xbs, y.data_ptr<float>(), batch, e);
+    } else {
+        gemv_launch<EPI_NONE, 0>(qs, scales, xq, xs, xbs, y.data_ptr<float>(), batch, e);
+    }
+}
+
+void q40_gemv_resid_q(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
+                      torch::Tensor xs, torch::Tensor xbs, torch::Tensor x,
+                      torch::Tensor ssq, torch::Tensor wnorm, torch::Tensor oq,
+                      torch::Tensor os, torch::Tensor obs) {
+    // decode down-projection with fused residual + ssq + DEFERRED Q80 emit
+    // of x*wnorm: replaces gemv_resid + norm_quant (the consumer runs with
+    // ssq_in to apply inv_rms). B=1 only; 16-wave wgs (one block per wg).
+    CHECK_CUDA(qs); CHECK_CONT(qs); CHECK_CONT(xq);
+    const int d = qs.size(0);
+    TORCH_CHECK(d % 32 == 0, "resid_q needs d % 32 == 0");
+    GemvEpi e;
+    e.x_resid = x.data_ptr<float>();
+    e.ssq = ssq.data_ptr<float>();
+    e.wnorm = wnorm.data_ptr<float>();
+    e.oqq = oq.data_ptr<int8_t>();
+    e.oqs = os.data_ptr<float>();
+    e.oqbs = obs.data_ptr<float>();
+    e.force_rpw2 = true;
+    gemv_launch<EPI_RESID_Q, 0>(qs, scales, xq, xs, xbs, x.data_ptr<float>(), 1, e);
 }
 
 void q40_gemv_resid(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
@@ -2128,7 +2241,9 @@ void q40_gemv_rope(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
                    torch::Tensor xs, torch::Tensor xbs, torch::Tensor y,
                    int64_t batch, torch::Tensor cache, torch::Tensor pos,
                    torch::Tensor kc, torch::Tensor vc, int64_t q_dim0,
-                   int64_t kv_dim0, int64_t head_dim) {
+                   int64_t kv_dim0, int64_t head_dim,
+                   c10::optional<torch::Tensor> ssq_in = c10::nullopt,
+                   double eps = 0.0) {
     CHECK_CUDA(qs); CHECK_CONT(qs); CHECK_CONT(xq);
     GemvEpi e;
     e.cache = cache.data_ptr<float>();
@@ -2140,7 +2255,13 @@ void q40_gemv_rope(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
     e.kv_dim0 = (int)kv_dim0;
     e.hd = (int)head_dim;
     e.force_rpw2 = true;  // rotation pairs live in one RPW=2 wave
-    gemv_launch<EPI_ROPE, 0>(qs, scales, xq, xs, xbs, y.data_ptr<float>(), batch, e);
+    if (ssq_in.has_value()) {  // deferred-scale input (see q40_gemv)
+        e.ssq_in = ssq_in->data_ptr<float>();
+        e.eps = (float)eps;
+        gemv_launch<EPI_ROPE, 2>(qs, scales, xq, xs, xbs, y.data_ptr<float>(), batch, e);
+    } else {
+        gemv_launch<EPI_ROPE, 0>(qs, scales, xq, xs, xbs, y.data_ptr<float>(), batch, e);
+    }
 }
 
 void q40_gemv_nq(torch::Tensor qs, torch::Tensor scales, torch::Tensor x,
@@ -2413,7 +2534,8 @@ void embed_gather(torch::Tensor table, torch::Tensor tokens, torch::Tensor x,
 void norm_quant(torch::Tensor x, torch::Tensor w, torch::Tensor ssq,
                 torch::Tensor q, torch::Tensor s, torch::Tensor bs,
                 int64_t batch, double eps,
-                c10::optional<torch::Tensor> yout = c10::nullopt) {
+                c10::optional<torch::Tensor> yout = c10::nullopt,
+                bool deferred = false) {
     CHECK_CUDA(x);
     const int n = x.size(-1);
     const dim3 grid(ceil_div(n, 256), batch);
@@ -2421,7 +2543,7 @@ void norm_quant(torch::Tensor x, torch::Tensor w, torch::Tensor ssq,
                        x.data_ptr<float>(), w.data_ptr<float>(), ssq.data_ptr<float>(),
                        q.data_ptr<int8_t>(), s.data_ptr<float>(), bs.data_ptr<float>(),
                        yout.has_value() ? yout->data_ptr<float>() : nullptr,
-                       n, (float)eps);
+                       n, (float)eps, deferred ? 1 : 0);
 }
 
 void moe_gate(torch::Tensor logits, torch::Tensor idx, torch::Tensor wts,
@@ -2429,7 +2551,7 @@ void moe_gate(torch::Tensor logits, torch::Tensor idx, torch::Tensor wts,
     CHECK_CUDA(logits);
     const int n_experts = logits.size(-1);
     TORCH_CHECK(n_experts <= 1024, "moe_gate supports <=1024 experts");
-    hipLaunchKernelGGL(k_moe_gate, dim3(batch), dim3(64), 0, cur_stream(),
+    hipLaunchKernelGGL(k_moe_gate, dim3(batch), dim3(128), 0, cur_stream(),
                        logits.data_ptr<float>(), idx.data_ptr<int>(),
                        wts.data_ptr<float>(), n_experts, (int)topk);
 }
@@ -2867,12 +2989,19 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("add_rmsnorm_q80", &add_rmsnorm_q80);
     m.def("q40_gemv", &q40_gemv, py::arg("qs"), py::arg("scales"), py::arg("xq"),
           py::arg("xs"), py::arg("xbs"), py::arg("y"), py::arg("batch"),
-          py::arg("amax_slot") = py::none());
+          py::arg("amax_slot") = py::none(), py::arg("ssq_in") = py::none(),
+          py::arg("eps") = 0.0);
     m.def("q40_gemv_resid", &q40_gemv_resid);
+    m.def("q40_gemv_resid_q", &q40_gemv_resid_q);
     m.def("q40_gemm", &q40_gemm, py::arg("qs"), py::arg("scales"),
           py::arg("xq"), py::arg("xs"), py::arg("y"), py::arg("batch"),
           py::arg("part") = py::none(), py::arg("variant") = -1);
-    m.def("q40_gemv_rope", &q40_gemv_rope);
+    m.def("q40_gemv_rope", &q40_gemv_rope, py::arg("qs"), py::arg("scales"),
+          py::arg("xq"), py::arg("xs"), py::arg("xbs"), py::arg("y"),
+          py::arg("batch"), py::arg("cache"), py::arg("pos"), py::arg("kc"),
+          py::arg("vc"), py::arg("q_dim0"), py::arg("kv_dim0"),
+          py::arg("head_dim"), py::arg("ssq_in") = py::none(),
+          py::arg("eps") = 0.0);
     m.def("q40_gemv_swiglu", &q40_gemv_swiglu, py::arg("qs"), py::arg("scales"),
           py::arg("xq"), py::arg("xs"), py::arg("xbs"), py::arg("oq"),
           py::arg("os"), py::arg("obs"), py::arg("gelu") = false);
@@ -2882,7 +3011,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("q40_gemv_nq_rope", &q40_gemv_nq_rope);
     m.def("norm_quant", &norm_quant, py::arg("x"), py::arg("w"), py::arg("ssq"),
           py::arg("q"), py::arg("s"), py::arg("bs"), py::arg("batch"),
-          py::arg("eps"), py::arg("yout") = py::none());
+          py::arg("eps"), py::arg("yout") = py::none(),
+          py::arg("deferred") = false);
     m.def("moe_gate", &moe_gate);
     m.def("scale_merge_add", &scale_merge_add, py::arg("x"), py::arg("y"),
           py::arg("wts"), py::arg("ssq"), py::arg("batch"), py::arg("topk"),

@@ -184,3 +184,46 @@ def test_tp_path_moe_world1(tmp_path):
     got = tp.forward(tokens, torch.arange(3)).cpu().clone()
     assert _rel_err(got, want) < 0.02, _rel_err(got, want)
     assert torch.equal(got.argmax(-1), want.argmax(-1))
+
+
+def test_deferred_quant_decode_matches_explicit(tiny_path):
+    """The deferred-quant decode path (EPI_RESID_Q producers + PRO=2
+    consumers, no norm_quant launches) must match the explicit-norm path:
+    Q80 codes are scale-invariant, so only fp rounding differs."""
+    from dllama_amd.models.hip_model import HipTransformer
+    m = mf.ModelFile(tiny_path)
+    cfg = ModelConfig.from_header(m.header)
+    expl = HipTransformer.from_file(m, cfg)
+    expl.use_deferred = False
+    deferred = HipTransformer.from_file(m, cfg)
+    prompt = [3, 17, 101]
+    expl.forward(torch.tensor(prompt), torch.arange(3))
+    deferred.forward(torch.tensor(prompt), torch.arange(3))
+    for step in range(5):
+        t = torch.tensor([20 + step])
+        p = torch.tensor([3 + step])
+        want = expl.forward(t, p).cpu().clone()
+        got = deferred.forward(t, p).cpu().clone()
+        assert _rel_err(got[0], want[0]) < 2e-3, (step, _rel_err(got[0], want[0]))
+        assert torch.equal(got.argmax(-1), want.argmax(-1)), f"step {step}"
+
+
+def test_deferred_quant_qwen3_dense(tmp_path):
+    """Deferred path through the qwen3 qk-norm variant (non-fused-rope
+    consumers with PRO=2) vs the CPU oracle."""
+    from dllama_amd.models.hip_model import HipTransformer
+    p = str(tmp_path / "q3.m")
+    make_tiny_qwen3(p, moe=False)
+    m = mf.ModelFile(p)
+    cfg = ModelConfig.from_header(m.header)
+    cpu = CpuTransformer(m, cfg)
+    hip = HipTransformer.from_file(m, cfg)
+    pr = [1, 2, 3]
+    hip.forward(torch.tensor(pr), torch.arange(3))
+    cpu.forward(torch.tensor(pr), torch.arange(3))
+    for step in range(3):
+        t = torch.tensor([9 + step])
+        pp = torch.tensor([3 + step])
+        want = cpu.forward(t, pp)
+        got = hip.forward(t, pp).cpu()
+        assert _rel_err(got[0], want[0]) < 0.03, f"step {step}"
