@@ -250,6 +250,9 @@ class HipTransformer:
         # pair, and per-wave gate recompute adds ~8 us to each MoE consumer
         self.fused_ffn = _os.environ.get("DLLAMA_FUSED_FFN", "0") == "1"
         self.fused_moe = _os.environ.get("DLLAMA_FUSED_MOE", "0") == "1"
+        # deferred-quant dense decode (EPI_RESID_Q + PRO2); =0 reverts to
+        # explicit norm_quant launches
+        self.use_deferred = _os.environ.get("DLLAMA_DEFERRED", "1") == "1"
         if ("DLLAMA_ATTN_SPLITS" in _os.environ
                 and "DLLAMA_ADAPTIVE_SPLITS" not in _os.environ):
             self.adaptive_thresh = 0
