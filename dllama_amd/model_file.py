@@ -381,9 +381,15 @@ def write_tensor(f, x: np.ndarray, float_type: int) -> int:
 
 
 def write_synthetic_model(path: str, h: LlmHeader, seed: int = 1234,
-                          scale: float = 0.02) -> None:
+                          scale: float = 0.02, fast: bool = False) -> None:
     """Random-init model in the canonical walk order (for tests/benches;
-    there is no network to download real checkpoints)."""
+    there is no network to download real checkpoints).
+
+    fast=True tiles one pre-quantized random block instead of drawing fresh
+    values per tensor — content is still well-formed Q40/F32 with sane
+    magnitudes, but writing becomes disk-bound (a 40 GB 70B file in tens of
+    seconds instead of many minutes of RNG+quantize). Used by the
+    big-model load-path harness (tools/load_bigmodel.py)."""
     if h.head_dim == 0:
         h.head_dim = h.dim // h.n_heads
     rng = np.random.default_rng(seed)
@@ -391,10 +397,31 @@ def write_synthetic_model(path: str, h: LlmHeader, seed: int = 1234,
         write_header(f, h)
     # re-read so header_size/offsets come from the file itself
     hdr = read_header(path)
+    pattern: dict[int, bytes] = {}
+    if fast:
+        import io
+        block = rng.standard_normal(size=1 << 22, dtype=np.float32) * scale
+        for ft in (F32, Q40):
+            buf = io.BytesIO()
+            write_tensor(buf, block, ft)
+            pattern[ft] = buf.getvalue()
     with open(path, "ab") as f:
         for e in tensor_walk(hdr):
-            x = rng.standard_normal(size=int(np.prod(e.shape)), dtype=np.float32) * scale
-            write_tensor(f, x, e.float_type)
+            n = int(np.prod(e.shape))
+            if fast:
+                pat = pattern[e.float_type]
+                want = _tensor_bytes(n, e.float_type)
+                reps = -(-want // len(pat))
+                f.write((pat * reps)[:want])
+            else:
+                x = rng.standard_normal(size=n, dtype=np.float32) * scale
+                write_tensor(f, x, e.float_type)
+
+
+def _tensor_bytes(n: int, float_type: int) -> int:
+    if float_type == Q40:
+        return n // 32 * 18
+    return n * 4
 
 
 # ------------------------------------------------------------- presets

@@ -150,6 +150,21 @@ __device__ __forceinline__ float ssq_total(const float *ssq, int b) {
     return t;
 }
 
+// wave-parallel variant: the 16 spread slots live on 16 different
+// cachelines (by design — atomics), so the serial loop above costs ~16
+// dependent-ish loads at every wave's kernel start; here lanes 0..15 each
+// load one slot and a 16-group reduce + broadcast distributes the sum
+// (used by the PRO consumers, which run on every wave of 3.5k workgroups)
+__device__ __forceinline__ float ssq_total_wave(const float *ssq, int b,
+                                                int lane) {
+    float t = (lane & 63) < SSQ_SPREAD
+                  ? ssq[(b * SSQ_SPREAD + (lane & 15)) * SSQ_PAD] : 0.0f;
+    #pragma unroll
+    for (int off = 8; off > 0; off >>= 1)
+        t += __shfl_xor(t, off, 16);
+    return __shfl(t, 0, WAVE);
+}
+
 // ------------------------------------------------------------------ q80 quantize
 // f32 [rows, n] -> int8 q [rows, n], f32 scale [rows, n/32], f32 bsum [rows, n/32]
 // (role of reference quantizeF32toQ80, nn-quants.cpp:67 and the
@@ -636,7 +651,7 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
     if (PRO) {
         #pragma unroll
         for (int b = 0; b < NB; b++)
-            invb[b] = rsqrtf(ssq_total(ssq_in, b) / n + eps);
+            invb[b] = rsqrtf(ssq_total_wave(ssq_in, b, lane) / n + eps);
     }
 
     for (int jp = lane; jp < nbp; jp += WAVE) {

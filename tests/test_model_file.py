@@ -97,3 +97,31 @@ def test_unknown_preset():
     import pytest
     with pytest.raises(KeyError, match="unknown preset"):
         mf.preset_header("llama-9000b")
+
+
+def test_write_synthetic_fast_matches_layout(tmp_path):
+    """fast=True produces a byte-layout-identical file structure (same
+    header, same tensor byte extents) as the slow writer."""
+    from dllama_amd import model_file as mf
+    h1 = mf.LlmHeader(arch_type=mf.ARCH_LLAMA, dim=64, hidden_dim=128,
+                      n_layers=2, n_heads=4, n_kv_heads=2, head_dim=16,
+                      vocab_size=128, seq_len=64, rope_theta=10000,
+                      rope_type=mf.ROPE_LLAMA)
+    h1.finalize()
+    p_slow = str(tmp_path / "slow.m")
+    p_fast = str(tmp_path / "fast.m")
+    mf.write_synthetic_model(p_slow, h1, seed=3)
+    h2 = mf.LlmHeader(arch_type=mf.ARCH_LLAMA, dim=64, hidden_dim=128,
+                      n_layers=2, n_heads=4, n_kv_heads=2, head_dim=16,
+                      vocab_size=128, seq_len=64, rope_theta=10000,
+                      rope_type=mf.ROPE_LLAMA)
+    h2.finalize()
+    mf.write_synthetic_model(p_fast, h2, seed=3, fast=True)
+    import os
+    assert os.path.getsize(p_fast) == os.path.getsize(p_slow)
+    # both parse and walk identically
+    a = mf.ModelFile(p_slow)
+    b = mf.ModelFile(p_fast)
+    for ea, eb in zip(mf.tensor_walk(a.header), mf.tensor_walk(b.header)):
+        assert ea.name == eb.name and ea.shape == eb.shape
+        assert ea.offset == eb.offset
