@@ -203,12 +203,25 @@ def main(argv=None) -> int:
     args = parser.parse_args(["inference"] + (argv if argv is not None
                                               else sys.argv[1:]))
     STATE = ApiState(args)
+    comm = STATE.comm
+    if comm.world > 1:
+        # TP serving: rank 0 owns HTTP + sampling; ranks > 0 replay control
+        # packets (reference root/worker split, app.cpp:168-230) — without
+        # this every rank would bind the port
+        from ..parallel.lockstep import RootModel, follower_loop
+        if comm.rank > 0:
+            follower_loop(STATE.engine.model, comm)
+            return 0
+        STATE.engine.model = RootModel(STATE.engine.model, comm)
     server = HTTPServer((args.host, args.port), Handler)
     print(f"⭐ dllama-api listening on {args.host}:{args.port}")
     try:
         server.serve_forever()
     except KeyboardInterrupt:
         pass
+    finally:
+        if comm.world > 1 and comm.rank == 0:
+            STATE.engine.model.stop_followers()
     return 0
 
 

@@ -189,3 +189,77 @@ def test_bench_driver_contract_tp2_cpu():
     assert j["n_gpus"] == 2 and j["steps"] == 4
     assert j["config"]["parallelism"] == "tp2"
     assert j["value"] > 0 and j["ms_per_step"] > 0
+
+
+def _api_tp_worker(rank, world, mp_path, tp_path, port, out_q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        os.environ["WORLD_SIZE"] = str(world)
+        os.environ["RANK"] = str(rank)
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from dllama_amd.apps.api import ApiState
+        from dllama_amd.apps.main import build_parser
+        from dllama_amd.parallel.lockstep import RootModel, follower_loop
+        args = build_parser().parse_args(
+            ["inference", "--model", mp_path, "--tokenizer", tp_path,
+             "--gpu-index", "-1", "--temperature", "0", "--seed", "1",
+             "--buffer-float-type", "f32"])
+        state = ApiState(args)
+        body = {"messages": [{"role": "user", "content": "abc"}],
+                "max_tokens": 6, "temperature": 0}
+        if rank == 0:
+            state.engine.model = RootModel(state.engine.model, state.comm)
+            text, _, n_gen = state.complete(body, lambda d: None)
+            state.engine.model.stop_followers()
+            out_q.put((0, (text, n_gen)))
+        else:
+            n = follower_loop(state.engine.model, state.comm)
+            out_q.put((1, n))
+        dist.destroy_process_group()
+    except Exception as e:  # noqa: BLE001
+        out_q.put((rank, f"ERROR: {e!r}"))
+
+
+@pytest.mark.timeout(240)
+def test_api_tp2_root_follower(tmp_path):
+    """dllama-api under TP: rank 0 serves + samples, rank 1 replays control
+    packets (reference root/worker split, app.cpp:168-230). The TP=2
+    completion must equal a single-rank completion of the same request."""
+    from dllama_amd.utils.testing import make_byte_tokenizer
+    mp_path = str(tmp_path / "tiny.m")
+    tp_path = str(tmp_path / "tiny.t")
+    make_byte_tokenizer(tp_path)
+    from dllama_amd.tokenizer import Tokenizer
+    vocab = Tokenizer(tp_path).vocab_size
+    make_tiny_llama(mp_path, vocab_size=vocab + (32 - vocab % 32) % 32)
+
+    # single-rank reference completion
+    from dllama_amd.apps.api import ApiState
+    from dllama_amd.apps.main import build_parser
+    os.environ.pop("WORLD_SIZE", None)
+    args = build_parser().parse_args(
+        ["inference", "--model", mp_path, "--tokenizer", tp_path,
+         "--gpu-index", "-1", "--temperature", "0", "--seed", "1",
+         "--buffer-float-type", "f32"])
+    ref_state = ApiState(args)
+    body = {"messages": [{"role": "user", "content": "abc"}],
+            "max_tokens": 6, "temperature": 0}
+    want_text, _, want_n = ref_state.complete(body, lambda d: None)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_api_tp_worker,
+                         args=(r, 2, mp_path, tp_path, 29573, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    res = dict(q.get(timeout=200) for _ in range(2))
+    for p in procs:
+        p.join(timeout=60)
+    assert not isinstance(res[0], str), res[0]
+    assert not isinstance(res[1], str), res[1]
+    text, n_gen = res[0]
+    assert (text, n_gen) == (want_text, want_n)
+    assert res[1] >= n_gen  # follower replayed prefill + decode forwards
