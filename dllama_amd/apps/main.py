@@ -42,10 +42,12 @@ def build_parser() -> argparse.ArgumentParser:
                    help="TP sync buffer quantization (reference --buffer-float-type)")
     p.add_argument("--nthreads", type=int, default=os.cpu_count(),
                    help="CPU backend threads (reference --nthreads)")
-    p.add_argument("--cpu-dtype", default="f32", choices=["f32", "f16"],
-                   help="CPU backend weight dtype: f16 halves resident weight "
-                        "memory (same speed; torch CPU f16 GEMV is not faster); "
-                        "f32 is the exact-oracle default")
+    p.add_argument("--cpu-dtype", default="f32", choices=["f32", "f16", "q40"],
+                   help="CPU backend weight handling: q40 streams the Q40 "
+                        "planes natively (quantized-weight RAM, native C++ "
+                        "matmul — role of the reference AVX512/NEON path); "
+                        "f16 halves resident memory via torch; f32 is the "
+                        "exact-oracle default")
     p.add_argument("--temperature", type=float, default=0.8)
     p.add_argument("--topp", type=float, default=0.9)
     p.add_argument("--seed", type=int, default=None)
@@ -100,7 +102,8 @@ def load_engine(args):
     else:
         from ..models.cpu_model import CpuTransformer
         torch.set_num_threads(max(1, args.nthreads))
-        wdt = torch.float16 if args.cpu_dtype == "f16" else torch.float32
+        wdt = ("q40" if args.cpu_dtype == "q40"
+               else torch.float16 if args.cpu_dtype == "f16" else torch.float32)
         model = CpuTransformer(m, cfg, comm, weight_dtype=wdt)
     tok = Tokenizer(args.tokenizer) if args.tokenizer else None
     seed = args.seed if args.seed is not None else int(time.time())

@@ -312,6 +312,20 @@ class ModelFile:
                 rows[:, node * row0_bytes: (node + 1) * row0_bytes]).reshape(-1)
         raise ValueError(e.kind)
 
+    def slice_q40_planes(self, name: str, layer: int, node: int,
+                         n_nodes: int, expert: int = -1):
+        """This rank's Q40 shard as device-layout planes (nibble uint8
+        [d0, n0/2] + f16 scales [d0, n0/32]) without dequantizing — the
+        native CPU matmul streams these directly."""
+        e = self.entry(name, layer, expert)
+        if e.float_type != Q40:
+            raise ValueError(f"{name} is not Q40")
+        d, n = e.shape
+        d0 = d // n_nodes if e.kind == "row" else d
+        n0 = n // n_nodes if e.kind == "col" else n
+        raw = self.slice_bytes(e, node, n_nodes)
+        return quants.q40_to_planes(raw, d0, n0)
+
     def slice_f32(self, name: str, layer: int, node: int, n_nodes: int,
                   expert: int = -1) -> np.ndarray:
         """This rank's shard dequantized to f32 with its sliced shape."""

@@ -21,6 +21,20 @@ import torch
 
 from ..model_file import HIDDEN_ACT_GELU, ModelFile, ROPE_FALCON
 from ..ops import reference as R
+from ..quants import Q40
+
+
+class Q40W:
+    """A Q40 weight kept in plane layout on CPU (uint8 nibbles + f16
+    scales); the native extension op streams it without a f32 copy."""
+
+    def __init__(self, qs: torch.Tensor, scales: torch.Tensor):
+        self.qs = qs
+        self.scales = scales
+        self.d = qs.shape[-2]
+
+    def __getitem__(self, e: int) -> "Q40W":  # stacked MoE experts
+        return Q40W(self.qs[e], self.scales[e])
 from ..parallel.comm import Comm, SingleComm
 from ..quants import Q80
 from .config import ModelConfig
@@ -38,10 +52,21 @@ class CpuTransformer:
         # decode: 0.92x). f32 stays the default: it is the numerics oracle
         # and the reference-parity path.
         self.weight_dtype = weight_dtype
+        self.q40_native = weight_dtype == "q40"
+        if self.q40_native:
+            # native Q40 streaming: quantized-weight RAM instead of the 8x
+            # f32 copy (reference CPU path, nn-cpu-ops.cpp:231-449)
+            if m.header.weight_type != Q40:
+                raise ValueError("--cpu-dtype q40 needs a Q40 .m file")
+            from ..ops import hip_ops
+            self._k = hip_ops()  # the in-tree extension (CPU entry point)
         c = self.cfg
         r, w = c.rank, c.world
 
         def t(name, layer=-1, expert=-1):
+            if self.q40_native:
+                qs, sc = m.slice_q40_planes(name, layer, r, w, expert)
+                return Q40W(torch.from_numpy(qs), torch.from_numpy(sc))
             return torch.from_numpy(
                 np.array(m.slice_f32(name, layer, r, w, expert))).to(weight_dtype)
 
@@ -58,9 +83,15 @@ class CpuTransformer:
             }
             if c.is_moe:
                 lw["gate"] = torch.from_numpy(np.array(m.f32("block_moe_gate", l)))
-                lw["w1"] = torch.stack([t("block_matmul_w1", l, e) for e in range(c.n_experts)])
-                lw["w2"] = torch.stack([t("block_matmul_w2", l, e) for e in range(c.n_experts)])
-                lw["w3"] = torch.stack([t("block_matmul_w3", l, e) for e in range(c.n_experts)])
+                def stack(name):
+                    parts = [t(name, l, e) for e in range(c.n_experts)]
+                    if self.q40_native:
+                        return Q40W(torch.stack([p.qs for p in parts]),
+                                    torch.stack([p.scales for p in parts]))
+                    return torch.stack(parts)
+                lw["w1"] = stack("block_matmul_w1")
+                lw["w2"] = stack("block_matmul_w2")
+                lw["w3"] = stack("block_matmul_w3")
             else:
                 lw["w1"] = t("block_matmul_w1", l)
                 lw["w2"] = t("block_matmul_w2", l)
@@ -81,7 +112,13 @@ class CpuTransformer:
             partial = R.q80_roundtrip(partial)
         return self.comm.allreduce_(partial)
 
-    def _matmul(self, x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    def _matmul(self, x: torch.Tensor, w) -> torch.Tensor:
+        if isinstance(w, Q40W):
+            if self.activation_quant:
+                x = R.q80_roundtrip(x)
+            y = torch.zeros(x.shape[0], w.d)
+            self._k.q40_matmul_cpu(w.qs, w.scales, x.contiguous().float(), y)
+            return y
         if self.weight_dtype != torch.float32:
             # fast serving path: Q80-roundtripped activations (same wire
             # semantics), f16 weight stream, f32 output
@@ -152,17 +189,13 @@ class CpuTransformer:
         B = t1.shape[0]
         router = t1 @ lw["gate"].t()
         idx, wts = R.moe_gate(router, c.n_active_experts)  # [B,k]
-        tq = R.q80_roundtrip(t1) if self.activation_quant else t1
-        tq = tq.to(self.weight_dtype)
         partial = torch.zeros(B, c.dim)
         for b in range(B):
             for s in range(c.n_active_experts):
                 e = int(idx[b, s])
-                a = (tq[b] @ lw["w1"][e].t()).float()
-                g = (tq[b] @ lw["w3"][e].t()).float()
+                a = self._matmul(t1[b:b + 1], lw["w1"][e])[0]
+                g = self._matmul(t1[b:b + 1], lw["w3"][e])[0]
                 d = R.swiglu(a, g)
-                if self.activation_quant:
-                    d = R.q80_roundtrip(d)
-                partial[b] += wts[b, s] * (d.to(self.weight_dtype)
-                                           @ lw["w2"][e].t()).float()
+                partial[b] += wts[b, s] * self._matmul(d.reshape(1, -1),
+                                                       lw["w2"][e])[0]
         return partial

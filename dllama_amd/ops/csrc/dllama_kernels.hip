@@ -2842,6 +2842,50 @@ void silu_mul(torch::Tensor a, torch::Tensor g, torch::Tensor out) {
                        a.data_ptr<float>(), g.data_ptr<float>(), out.data_ptr<float>(), n);
 }
 
+// ------------------------------------------------------------ CPU Q40 GEMM
+// Native CPU matmul streaming Q40 planes directly (no f32 dequant copy):
+// y[b, row] = sum_blk scale * sum_k (q-8) * x — keeps CPU serving at
+// quantized-weight RAM (role of the reference's AVX512/NEON CPU kernels,
+// src/nn/nn-cpu-ops.cpp:231-449). Parallel over rows via at::parallel_for
+// (torch threads, the CLI's --nthreads). The inner 32-wide block loop
+// auto-vectorizes; activations stay f32 (more precise than the reference's
+// Q80 activations, same wire format compatibility).
+void q40_matmul_cpu(torch::Tensor qs, torch::Tensor scales, torch::Tensor x,
+                    torch::Tensor y) {
+    TORCH_CHECK(!qs.is_cuda() && !x.is_cuda(), "q40_matmul_cpu is CPU-only");
+    CHECK_CONT(qs); CHECK_CONT(x);
+    const int d = qs.size(0);
+    const int n = qs.size(1) * 2;
+    const int B = x.size(0);
+    const int nb = n / QB;
+    TORCH_CHECK(x.size(-1) == n, "x width mismatch");
+    const uint8_t *W = qs.data_ptr<uint8_t>();
+    const at::Half *S = scales.data_ptr<at::Half>();
+    const float *X = x.data_ptr<float>();
+    float *Y = y.data_ptr<float>();
+    at::parallel_for(0, d, 16, [&](int64_t r0, int64_t r1) {
+        for (int64_t row = r0; row < r1; row++) {
+            const uint8_t *w = W + row * (n >> 1);
+            const at::Half *sc = S + row * nb;
+            for (int b = 0; b < B; b++) {
+                const float *xr = X + (int64_t)b * n;
+                float acc = 0.0f;
+                for (int blk = 0; blk < nb; blk++) {
+                    const uint8_t *wb = w + blk * 16;
+                    const float *xb = xr + blk * QB;
+                    float a = 0.0f;
+                    for (int k = 0; k < 16; k++) {
+                        a += (float)((int)(wb[k] & 15) - 8) * xb[k]
+                           + (float)((int)(wb[k] >> 4) - 8) * xb[k + 16];
+                    }
+                    acc += (float)sc[blk] * a;
+                }
+                Y[(int64_t)b * d + row] = acc;
+            }
+        }
+    });
+}
+
 void sync_pack(torch::Tensor q, torch::Tensor s, torch::Tensor buf) {
     CHECK_CUDA(q);
     const int n = q.size(-1);
@@ -3008,6 +3052,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("eps") = 0.0);
     m.def("q40_gemv_resid", &q40_gemv_resid);
     m.def("q40_gemv_resid_q", &q40_gemv_resid_q);
+    m.def("q40_matmul_cpu", &q40_matmul_cpu);
     m.def("q40_gemm", &q40_gemm, py::arg("qs"), py::arg("scales"),
           py::arg("xq"), py::arg("xs"), py::arg("y"), py::arg("batch"),
           py::arg("part") = py::none(), py::arg("variant") = -1);

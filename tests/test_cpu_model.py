@@ -288,3 +288,42 @@ def test_cpu_f16_weights_close_to_f32(tmp_path):
     # tight relative agreement and same greedy tokens on a decisive model
     err = (f16 - ref).abs().max() / ref.abs().max()
     assert float(err) < 0.05, float(err)
+
+
+def test_cpu_q40_native_matches_f32(tmp_path):
+    """--cpu-dtype q40: native C++ Q40-plane matmul (quantized-weight RAM)
+    vs the f32-dequant oracle (reference CPU kernels,
+    nn-cpu-ops.cpp:231-449). Activations f32 here, so differences come only
+    from the weight stream being identical Q40 values."""
+    from dllama_amd.models.config import ModelConfig
+    from dllama_amd.models.cpu_model import CpuTransformer
+    from dllama_amd.utils.testing import make_tiny_llama
+    p = str(tmp_path / "q.m")
+    make_tiny_llama(p, vocab_size=256)
+    m = mf.ModelFile(p)
+    cfg = ModelConfig.from_header(m.header)
+    ref = CpuTransformer(m, cfg)
+    nat = CpuTransformer(m, cfg, weight_dtype="q40")
+    tokens = torch.tensor([3, 17, 101])
+    want = ref.forward(tokens, torch.arange(3))
+    got = nat.forward(tokens, torch.arange(3))
+    err = (got - want).abs().max().item() / (want.abs().max().item() + 1e-9)
+    assert err < 1e-4, err
+    assert torch.equal(got.argmax(-1), want.argmax(-1))
+
+
+def test_cpu_q40_native_moe(tmp_path):
+    from dllama_amd.models.config import ModelConfig
+    from dllama_amd.models.cpu_model import CpuTransformer
+    from dllama_amd.utils.testing import make_tiny_qwen3
+    p = str(tmp_path / "qm.m")
+    make_tiny_qwen3(p, moe=True)
+    m = mf.ModelFile(p)
+    cfg = ModelConfig.from_header(m.header)
+    ref = CpuTransformer(m, cfg)
+    nat = CpuTransformer(m, cfg, weight_dtype="q40")
+    tokens = torch.tensor([1, 2, 3])
+    want = ref.forward(tokens, torch.arange(3))
+    got = nat.forward(tokens, torch.arange(3))
+    err = (got - want).abs().max().item() / (want.abs().max().item() + 1e-9)
+    assert err < 5e-3, err
