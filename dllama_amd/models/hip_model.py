@@ -253,6 +253,11 @@ class HipTransformer:
         # deferred-quant dense decode (EPI_RESID_Q + PRO2); =0 reverts to
         # explicit norm_quant launches
         self.use_deferred = _os.environ.get("DLLAMA_DEFERRED", "1") == "1"
+        # bf16 prefill: dense prompt chunks run hipBLASLt GEMMs (torch.mm)
+        # over a bf16 copy of the weights — MFMA-bound instead of the
+        # VALU-bound Q40 int8 GEMM (~1.3 TB/s ceiling). Costs a 2x-weight
+        # bf16 shadow; =0 keeps the hand-written int8-MFMA GEMM prefill.
+        self.prefill_bf16 = _os.environ.get("DLLAMA_PREFILL_BF16", "1") == "1"
         if ("DLLAMA_ATTN_SPLITS" in _os.environ
                 and "DLLAMA_ADAPTIVE_SPLITS" not in _os.environ):
             self.adaptive_thresh = 0
@@ -285,8 +290,37 @@ class HipTransformer:
             self.moe_dq = QuantBuf(S, c.ff_dim0, dev)
             self.moe_y = torch.zeros(S, c.dim, device=dev)
 
+    def _dequant_bf16(self, lin: Linear) -> torch.Tensor:
+        """Q40 planes -> bf16 weight matrix on device (prefill GEMM shadow;
+        same dequantized values the int8 GEMV streams, rounded to bf16)."""
+        qs = lin.qs
+        d, nb2 = qs.shape[-2], qs.shape[-1]
+        n = nb2 * 2
+        lo = (qs & 15).to(torch.float32) - 8.0
+        hi = (qs >> 4).to(torch.float32) - 8.0
+        lo = lo.view(*qs.shape[:-1], n // 32, 16)
+        hi = hi.view(*qs.shape[:-1], n // 32, 16)
+        w = torch.cat([lo, hi], dim=-1)  # elems j | j+16 per block
+        sc = lin.scales.view(*qs.shape[:-1], n // 32, 1).float()
+        return (w * sc).view(*qs.shape[:-2], d, n).to(torch.bfloat16)
+
+    def _build_bf16_shadow(self):
+        """bf16 copies of the dense matmul weights for hipBLASLt prefill
+        (skipped for MoE/TP where the Q40 GEMM path stays)."""
+        c = self.cfg
+        if not self.prefill_bf16 or c.is_moe or self.tp_path:
+            self.prefill_bf16 = False
+            return
+        self.wcls_bf16 = self._dequant_bf16(self.wcls)
+        for lw in self.layers:
+            lw["qkv_bf16"] = self._dequant_bf16(lw["qkv"])
+            lw["wo_bf16"] = self._dequant_bf16(lw["wo"])
+            lw["w13_bf16"] = self._dequant_bf16(lw["w13"])
+            lw["w2_bf16"] = self._dequant_bf16(lw["w2"])
+
     def _finish_init(self):
         c, dev = self.cfg, self.device
+        self._build_bf16_shadow()
         cache = R.rope_cache(c.seq_len, c.head_dim, c.rope_theta, c.rope_scaling)
         self.rope_cache = cache.reshape(c.seq_len, c.head_dim).contiguous().to(dev)
         # f16 KV cache by default: halves the attention HBM stream (measured
@@ -362,6 +396,8 @@ class HipTransformer:
                 and getattr(self, "use_deferred", True)
                 and not getattr(self, "use_fused_norm", False)):
             return self._forward_dense_deferred()
+        if B >= 8 and getattr(self, "prefill_bf16", False):
+            return self._forward_prefill_bf16(B)
         x = self.x
         self.ssq.zero_()
         k.embed_gather(self.embedding, self.tokens, x, NB, self.ssq[0])
@@ -472,6 +508,71 @@ class HipTransformer:
                 # logits row; two-kernel argmax, no ATen in the graph
                 k.argmax_token(self.tokens, self.logits_gather[1].view(-1),
                                self.argmax_scratch_full)
+
+    def _forward_prefill_bf16(self, B: int):
+        """Dense prompt chunks as hipBLASLt bf16 GEMMs over the weight
+        shadow: prefill is batch>=8 plain GEMM work, which belongs on the
+        matrix cores — the hand-written Q40 int8 GEMM is VALU-bound at
+        ~1.3 TB/s weight stream while the bf16 library GEMM is MFMA-bound
+        (the reference's llamafile sgemm plays this exact role for batch>1,
+        src/nn/llamafile/sgemm.cpp:819-986). rope/KV/attention/norm stay in
+        the HIP kernels; elementwise glue is torch (prefill runs once per
+        32-token chunk, not per token)."""
+        import torch.nn.functional as F
+        c, k = self.cfg, self.k
+        NB = _pow2_batch(B)
+        x = self.x
+        eps = c.norm_eps
+        self.ssq.zero_()
+        k.embed_gather(self.embedding, self.tokens, x, NB, self.ssq[0])
+        kv_mul = c.n_heads0 // max(1, c.kv_dim0 // c.head_dim)
+        slot = 0
+        for l, lw in enumerate(self.layers):
+            k.norm_f32(x[:NB], lw["norm0"], self.ssq[slot], self.t_norm[:NB],
+                       NB, eps)
+            self.qkv_out[:NB].copy_(
+                torch.mm(self.t_norm[:NB].bfloat16(), lw["qkv_bf16"].t()))
+            if c.is_qwen3 and self.rope_style == 1:
+                k.rope_kv_qknorm(self.qkv_out, self.qkv_ld, c.q_dim0,
+                                 c.kv_dim0, self.rope_cache, self.pos,
+                                 self.k_cache[l], self.v_cache[l], c.head_dim,
+                                 lw["q_norm"], lw["k_norm"], eps, B)
+            else:
+                if c.is_qwen3:
+                    k.rmsnorm_rows_s(self.qkv_out, self.qkv_ld, 0,
+                                     c.q_dim0 // c.head_dim, B, lw["q_norm"],
+                                     c.head_dim, eps)
+                    k.rmsnorm_rows_s(self.qkv_out, self.qkv_ld, c.q_dim0,
+                                     c.kv_dim0 // c.head_dim, B, lw["k_norm"],
+                                     c.head_dim, eps)
+                k.rope_kv(self.qkv_out, self.qkv_ld, c.q_dim0, c.kv_dim0,
+                          self.rope_cache, self.pos, self.k_cache[l],
+                          self.v_cache[l], c.head_dim, self.rope_style, B)
+            k.attn(self.qkv_out, self.qkv_ld, self.k_cache[l], self.v_cache[l],
+                   self.zbuf[:B], self.pos, B, c.n_heads0, kv_mul, c.head_dim,
+                   self.attn_splits, self.attn_ml, self.attn_o,
+                   self.attn_counter)
+            partial = torch.mm(self.zbuf[:NB].bfloat16(),
+                               lw["wo_bf16"].t()).float()
+            k.add_ssq(x[:NB], partial, self.ssq[slot + 1], NB)
+            slot += 1
+            k.norm_f32(x[:NB], lw["norm1"], self.ssq[slot], self.t_norm[:NB],
+                       NB, eps)
+            ff = torch.mm(self.t_norm[:NB].bfloat16(), lw["w13_bf16"].t()).float()
+            a, g = ff[:, :c.ff_dim0], ff[:, c.ff_dim0:]
+            if c.hidden_act == HIDDEN_ACT_GELU:
+                d = F.gelu(a, approximate="tanh") * g
+            else:
+                d = F.silu(a) * g
+            partial = torch.mm(d.bfloat16(), lw["w2_bf16"].t()).float()
+            k.add_ssq(x[:NB], partial, self.ssq[slot + 1], NB)
+            slot += 1
+        if self.skip_logits and B > 1:
+            return
+        k.norm_f32(x[:NB], self.final_norm, self.ssq[slot], self.t_norm[:NB],
+                   NB, eps)
+        self.logits0[:NB].copy_(
+            torch.mm(self.t_norm[:NB].bfloat16(), self.wcls_bf16.t()))
 
     def _forward_dense_deferred(self):
         """B=1 dense decode with DEFERRED activation quantization: the
