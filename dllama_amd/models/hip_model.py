@@ -254,10 +254,11 @@ class HipTransformer:
         # explicit norm_quant launches
         self.use_deferred = _os.environ.get("DLLAMA_DEFERRED", "1") == "1"
         # bf16 prefill: dense prompt chunks run hipBLASLt GEMMs (torch.mm)
-        # over a bf16 copy of the weights — MFMA-bound instead of the
-        # VALU-bound Q40 int8 GEMM (~1.3 TB/s ceiling). Costs a 2x-weight
-        # bf16 shadow; =0 keeps the hand-written int8-MFMA GEMM prefill.
-        self.prefill_bf16 = _os.environ.get("DLLAMA_PREFILL_BF16", "1") == "1"
+        # over a bf16 copy of the weights. Measured SLOWER than the
+        # hand-written int8-MFMA GEMM at the 32-token chunk shape (5355 vs
+        # 5651 tok/s prefill) and costs a 2x-weight shadow — off by default,
+        # kept as the library baseline for GEMM tuning A/Bs.
+        self.prefill_bf16 = _os.environ.get("DLLAMA_PREFILL_BF16", "0") == "1"
         if ("DLLAMA_ATTN_SPLITS" in _os.environ
                 and "DLLAMA_ADAPTIVE_SPLITS" not in _os.environ):
             self.adaptive_thresh = 0
