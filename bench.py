@@ -189,8 +189,9 @@ def main():
     toks = args.steps / elapsed
     base = BASELINE_TOKS.get(n_gpus)
     if rank == 0:
+        mname = "Llama-3.1-8B" if args.model == "llama-3.1-8b" else args.model
         print(json.dumps({
-            "metric": "decode tokens/s (Llama-3.1-8B Q40)",
+            "metric": f"decode tokens/s ({mname} Q40)",
             "value": round(toks, 2),
             "unit": "tokens/s",
             "n_gpus": n_gpus,
