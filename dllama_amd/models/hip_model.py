@@ -392,8 +392,9 @@ class HipTransformer:
         stays on device — this function is graph-capturable."""
         c, k = self.cfg, self.k
         NB = _pow2_batch(B)
-        if (B == 1 and not c.is_moe and not self.tp_path
-                and c.dim % 32 == 0
+        if (B == 1 and not self.tp_path and c.dim % 32 == 0
+                and (not c.is_moe
+                     or (c.dim % 256 == 0 and c.n_active_experts <= 16))
                 and getattr(self, "use_deferred", True)
                 and not getattr(self, "use_fused_norm", False)):
             return self._forward_dense_deferred()
@@ -632,24 +633,49 @@ class HipTransformer:
                    self.zbuf[:1], self.pos, 1, c.n_heads0, kv_mul, c.head_dim,
                    self.attn_splits, self.attn_ml, self.attn_o,
                    self.attn_counter, self.zq.q, self.zq.s, self.zq.bs)
-            # wo: residual fold + deferred Q80 emit of x*norm1 for w13
+            # wo: residual fold + deferred Q80 emit of x*norm1 for the FFN
             k.q40_gemv_resid_q(lw["wo"].qs, lw["wo"].scales, self.zq.q,
                                self.zq.s, self.zq.bs, x, self.ssq[slot + 1],
                                lw["norm1"], self.xq.q, self.xq.s, self.xq.bs)
             slot += 1
-            k.q40_gemv(lw["w13"].qs, lw["w13"].scales, self.xq.q, self.xq.s,
-                       self.xq.bs, self.ff_out, 1, ssq_in=self.ssq[slot],
-                       eps=eps)
-            k.swiglu_q80(self.ff_out, self.ff_out[:, c.ff_dim0:],
-                         2 * c.ff_dim0, c.ff_dim0, 1, self.dq.q[:1],
-                         self.dq.s[:1], self.dq.bs[:1],
-                         c.hidden_act == HIDDEN_ACT_GELU)
-            # w2: residual fold + deferred emit for the NEXT layer's norm0
-            # (final_norm for the logits matmul after the last layer)
             wn = self.final_norm if l == last else self.layers[l + 1]["norm0"]
-            k.q40_gemv_resid_q(lw["w2"].qs, lw["w2"].scales, self.dq.q,
-                               self.dq.s, self.dq.bs, x, self.ssq[slot + 1],
-                               wn, self.xq.q, self.xq.s, self.xq.bs)
+            if c.is_moe:
+                # MoE deferred FFN: router reads x*norm1*inv directly (no
+                # t_norm buffer), grouped w13 consumes the deferred xq, and
+                # the scale-merge epilogue emits the NEXT layer's deferred
+                # quant — both per-layer norm_quant launches gone here too
+                ka = c.n_active_experts
+                k.router_gemv_norm(lw["gate"], x, lw["norm1"],
+                                   self.ssq[slot], eps, self.moe_router, 1)
+                k.moe_gate(self.moe_router[:1], self.moe_idx, self.moe_wts,
+                           1, ka)
+                k.q40_gemv_grouped(lw["w13"].qs, lw["w13"].scales, self.xq.q,
+                                   self.xq.s, self.xq.bs, self.moe_idx[:ka],
+                                   self.moe_out13, ka,
+                                   ssq_in=self.ssq[slot], eps=eps)
+                k.swiglu_q80(self.moe_out13, self.moe_out13[:, c.ff_dim0:],
+                             2 * c.ff_dim0, c.ff_dim0, ka, self.moe_dq.q[:ka],
+                             self.moe_dq.s[:ka], self.moe_dq.bs[:ka])
+                k.q40_gemv_grouped(lw["w2"].qs, lw["w2"].scales, self.moe_dq.q,
+                                   self.moe_dq.s, self.moe_dq.bs,
+                                   self.moe_idx[:ka], self.moe_y, 1)
+                k.scale_merge_add_q(x[:1], self.moe_y, self.moe_wts,
+                                    self.ssq[slot + 1], wn, self.xq.q,
+                                    self.xq.s, self.xq.bs, 1, ka)
+            else:
+                k.q40_gemv(lw["w13"].qs, lw["w13"].scales, self.xq.q,
+                           self.xq.s, self.xq.bs, self.ff_out, 1,
+                           ssq_in=self.ssq[slot], eps=eps)
+                k.swiglu_q80(self.ff_out, self.ff_out[:, c.ff_dim0:],
+                             2 * c.ff_dim0, c.ff_dim0, 1, self.dq.q[:1],
+                             self.dq.s[:1], self.dq.bs[:1],
+                             c.hidden_act == HIDDEN_ACT_GELU)
+                # w2: residual fold + deferred emit for the NEXT layer's
+                # norm0 (final_norm for logits after the last layer)
+                k.q40_gemv_resid_q(lw["w2"].qs, lw["w2"].scales, self.dq.q,
+                                   self.dq.s, self.dq.bs, x,
+                                   self.ssq[slot + 1], wn, self.xq.q,
+                                   self.xq.s, self.xq.bs)
             slot += 1
         use_amax = self.greedy_feedback
         k.q40_gemv(self.wcls.qs, self.wcls.scales, self.xq.q, self.xq.s,

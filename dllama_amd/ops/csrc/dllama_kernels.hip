@@ -1049,7 +1049,9 @@ __global__ void k_q40_gemv_grouped(const uint8_t *__restrict__ qs,
                                    float *__restrict__ y,
                                    int d, int n, int k_slots,
                                    const float *__restrict__ router,
-                                   int n_experts, int topk) {
+                                   int n_experts, int topk,
+                                   const float *__restrict__ ssq_in,
+                                   float eps) {
     constexpr int NGRP = WAVE / LPP;
     const int wpb = blockDim.x / WAVE;
     const int full_lane = threadIdx.x % WAVE;
@@ -1114,6 +1116,13 @@ __global__ void k_q40_gemv_grouped(const uint8_t *__restrict__ qs,
     for (int o = LPP / 2; o > 0; o >>= 1) {
         acc0 += __shfl_down(acc0, o, WAVE);
         acc1 += __shfl_down(acc1, o, WAVE);
+    }
+    if (ssq_in != nullptr) {
+        // deferred-scale input (see EPI_RESID_Q): apply inv_rms per row
+        const float inv = rsqrtf(
+            ssq_total_wave(ssq_in, b, threadIdx.x % WAVE) / n + eps);
+        acc0 *= inv;
+        acc1 *= inv;
     }
     if (lane == 0) {
         y[(int64_t)slot * d + row0] = acc0;
@@ -1371,6 +1380,36 @@ __global__ void k_gemm_reduce(const float *__restrict__ part,
     }
 }
 
+// router GEMV with the FFN rmsnorm fused: logits[b][e] =
+// inv_rms(b) * sum_i gate[e,i] * x[b,i] * wnorm[i] — the MoE deferred-quant
+// path never materializes t_norm (reference computes rms_norm then the gate
+// matmul as separate ops, llm.cpp:430-450)
+__global__ void k_router_gemv_norm(const float *__restrict__ gate,
+                                   const float *__restrict__ x,
+                                   const float *__restrict__ wnorm,
+                                   const float *__restrict__ ssq,
+                                   float *__restrict__ logits,
+                                   int n_experts, int dim, float eps) {
+    const int e = blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
+    const int b = blockIdx.y;
+    if (e >= n_experts) return;
+    const int lane = threadIdx.x % WAVE;
+    const float inv = rsqrtf(ssq_total_wave(ssq, b, lane) / dim + eps);
+    const float4 *g4 = reinterpret_cast<const float4 *>(gate + (int64_t)e * dim);
+    const float4 *x4 = reinterpret_cast<const float4 *>(x + (int64_t)b * dim);
+    const float4 *w4 = reinterpret_cast<const float4 *>(wnorm);
+    float acc = 0.0f;
+    for (int i = lane; i < dim / 4; i += WAVE) {
+        const float4 gv = g4[i];
+        const float4 xv = x4[i];
+        const float4 wv = w4[i];
+        acc += gv.x * xv.x * wv.x + gv.y * xv.y * wv.y
+             + gv.z * xv.z * wv.z + gv.w * xv.w * wv.w;
+    }
+    acc = wave_reduce_sum(acc);
+    if (lane == 0) logits[(int64_t)b * n_experts + e] = acc * inv;
+}
+
 // f32 router GEMV: logits[b][e] = gate[e,:] . t[b,:]
 __global__ void k_router_gemv(const float *__restrict__ gate,
                               const float *__restrict__ t,
@@ -1565,6 +1604,52 @@ __global__ void k_argmax_stage1(const float *__restrict__ x, int64_t n,
     if (threadIdx.x == 0) {
         for (int i = 1; i < blockDim.x / WAVE; i++) best = max(best, red[i]);
         scratch[blockIdx.x] = best;
+    }
+}
+
+// MoE deferred-quant epilogue: weighted expert sum + residual fold + ssq +
+// DEFERRED Q80 emit of x*wnorm (the next matmul's input; consumer applies
+// inv_rms). Each 256-thread wg owns 256 consecutive elements = 8 wg-local
+// Q80 blocks, so the quantization is barrier-free per 32-lane group.
+__global__ void k_scale_merge_add_q(float *__restrict__ x,
+                                    const float *__restrict__ y,
+                                    const float *__restrict__ wts,
+                                    float *__restrict__ ssq,
+                                    const float *__restrict__ wnorm,
+                                    int8_t *__restrict__ oq,
+                                    float *__restrict__ os,
+                                    float *__restrict__ obs,
+                                    int n, int topk) {
+    const int b = blockIdx.y;
+    const int i = blockIdx.x * 256 + threadIdx.x;
+    float acc = x[(int64_t)b * n + i];
+    for (int s = 0; s < topk; s++)
+        acc = fmaf(wts[(int64_t)b * topk + s],
+                   y[((int64_t)b * topk + s) * n + i], acc);
+    x[(int64_t)b * n + i] = acc;
+    // ssq: per-wg reduce then one spread-slot atomic
+    float local = acc * acc;
+    local = wave_reduce_sum(local);
+    __shared__ float red[4];
+    const int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    if (lane == 0) red[wid] = local;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        atomicAdd(ssq + (b * SSQ_SPREAD + (blockIdx.x & (SSQ_SPREAD - 1))) * SSQ_PAD,
+                  red[0] + red[1] + red[2] + red[3]);
+    }
+    // deferred quant of x*wnorm: block = this thread's 32-lane group
+    const float v = acc * wnorm[i];
+    const float amax = group32_reduce_max(fabsf(v));
+    const float dd = amax / 127.0f;
+    const float qinv = dd > 0.0f ? 1.0f / dd : 0.0f;
+    const float qf = rintf(v * qinv);
+    oq[(int64_t)b * n + i] = (int8_t)qf;
+    const float bsum = group32_reduce_sum(qf);
+    if ((threadIdx.x & 31) == 0) {
+        const int blk = i / QB;
+        os[(int64_t)b * (n / QB) + blk] = dd;
+        obs[(int64_t)b * (n / QB) + blk] = bsum;
     }
 }
 
@@ -2367,7 +2452,9 @@ void q40_gemv_grouped(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
                       torch::Tensor xs, torch::Tensor xbs, torch::Tensor expert_idx,
                       torch::Tensor y, int64_t k_slots, int64_t variant = -1,
                       c10::optional<torch::Tensor> router = c10::nullopt,
-                      int64_t topk = 0, int64_t n_slots_override = 0) {
+                      int64_t topk = 0, int64_t n_slots_override = 0,
+                      c10::optional<torch::Tensor> ssq_in = c10::nullopt,
+                      double eps = 0.0) {
     // router given: expert ids come from the in-kernel gate over the router
     // logits [B, n_experts] (expert_idx is then ignored); n_slots_override
     // sets the slot count (it can no longer come from expert_idx.numel())
@@ -2395,6 +2482,7 @@ void q40_gemv_grouped(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
     const bool gate = router.has_value();
     const float *rp = gate ? router->data_ptr<float>() : nullptr;
     const int ne = gate ? (int)router->size(-1) : 0;
+    const float *sqp = ssq_in.has_value() ? ssq_in->data_ptr<float>() : nullptr;
     auto launch = [&](auto k) {
         hipLaunchKernelGGL(k, grid, block, 0, cur_stream(),
                            qs.data_ptr<uint8_t>(),
@@ -2402,7 +2490,7 @@ void q40_gemv_grouped(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
                            xq.data_ptr<int8_t>(), xs.data_ptr<float>(),
                            xbs.data_ptr<float>(), expert_idx.data_ptr<int>(),
                            y.data_ptr<float>(), d, n, (int)k_slots,
-                           rp, ne, (int)topk);
+                           rp, ne, (int)topk, sqp, (float)eps);
     };
     if (gate) {
         switch (lpp) {
@@ -2628,6 +2716,33 @@ void argmax_token(torch::Tensor token, torch::Tensor x, torch::Tensor scratch) {
                        x.data_ptr<float>(), n, sp);
     hipLaunchKernelGGL(k_token_from_argmax, dim3(1), dim3(1024), 0, cur_stream(),
                        (long *)token.data_ptr<int64_t>(), sp, blocks);
+}
+
+void scale_merge_add_q(torch::Tensor x, torch::Tensor y, torch::Tensor wts,
+                       torch::Tensor ssq, torch::Tensor wnorm,
+                       torch::Tensor oq, torch::Tensor os, torch::Tensor obs,
+                       int64_t batch, int64_t topk) {
+    CHECK_CUDA(x);
+    const int n = x.size(-1);
+    TORCH_CHECK(n % 256 == 0, "scale_merge_add_q needs dim % 256 == 0");
+    hipLaunchKernelGGL(k_scale_merge_add_q, dim3(n / 256, batch), dim3(256),
+                       0, cur_stream(), x.data_ptr<float>(), y.data_ptr<float>(),
+                       wts.data_ptr<float>(), ssq.data_ptr<float>(),
+                       wnorm.data_ptr<float>(), oq.data_ptr<int8_t>(),
+                       os.data_ptr<float>(), obs.data_ptr<float>(), n, (int)topk);
+}
+
+void router_gemv_norm(torch::Tensor gate, torch::Tensor x, torch::Tensor wnorm,
+                      torch::Tensor ssq, double eps, torch::Tensor logits,
+                      int64_t batch) {
+    CHECK_CUDA(gate);
+    const int n_experts = gate.size(0);
+    const int dim = gate.size(1);
+    hipLaunchKernelGGL(k_router_gemv_norm, dim3(ceil_div(n_experts, 4), batch),
+                       dim3(256), 0, cur_stream(), gate.data_ptr<float>(),
+                       x.data_ptr<float>(), wnorm.data_ptr<float>(),
+                       ssq.data_ptr<float>(), logits.data_ptr<float>(),
+                       n_experts, dim, (float)eps);
 }
 
 void router_gemv(torch::Tensor gate, torch::Tensor t, torch::Tensor logits,
@@ -3083,13 +3198,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("logits_concat", &logits_concat);
     m.def("argmax_token", &argmax_token);
     m.def("router_gemv", &router_gemv);
+    m.def("router_gemv_norm", &router_gemv_norm);
+    m.def("scale_merge_add_q", &scale_merge_add_q);
     m.def("norm_f32", &norm_f32);
     m.def("add_ssq", &add_ssq);
     m.def("q40_gemv_grouped", &q40_gemv_grouped, py::arg("qs"),
           py::arg("scales"), py::arg("xq"), py::arg("xs"), py::arg("xbs"),
           py::arg("expert_idx"), py::arg("y"), py::arg("k_slots"),
           py::arg("variant") = -1, py::arg("router") = py::none(),
-          py::arg("topk") = 0, py::arg("n_slots") = 0);
+          py::arg("topk") = 0, py::arg("n_slots") = 0,
+          py::arg("ssq_in") = py::none(), py::arg("eps") = 0.0);
     m.def("q40_gemv_grouped_swiglu", &q40_gemv_grouped_swiglu, py::arg("qs"),
           py::arg("scales"), py::arg("xq"), py::arg("xs"), py::arg("xbs"),
           py::arg("oq"), py::arg("os"), py::arg("obs"), py::arg("n_slots"),

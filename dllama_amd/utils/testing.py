@@ -39,9 +39,10 @@ def make_tiny_llama(path: str, seed: int = 7, vocab_size: int = 512,
     return h
 
 
-def make_tiny_qwen3(path: str, seed: int = 9, moe: bool = False) -> mf.LlmHeader:
+def make_tiny_qwen3(path: str, seed: int = 9, moe: bool = False,
+                    dim: int = 64) -> mf.LlmHeader:
     if moe:
-        h = mf.LlmHeader(arch_type=mf.ARCH_QWEN3_MOE, dim=64, hidden_dim=96,
+        h = mf.LlmHeader(arch_type=mf.ARCH_QWEN3_MOE, dim=dim, hidden_dim=96,
                          n_layers=2, n_heads=4, n_kv_heads=2, head_dim=64,
                          n_experts=4, n_active_experts=2, moe_hidden_dim=64,
                          vocab_size=256, seq_len=64, rope_theta=10000,

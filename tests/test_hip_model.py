@@ -227,3 +227,31 @@ def test_deferred_quant_qwen3_dense(tmp_path):
         want = cpu.forward(t, pp)
         got = hip.forward(t, pp).cpu()
         assert _rel_err(got[0], want[0]) < 0.03, f"step {step}"
+
+
+def test_deferred_quant_moe(tmp_path):
+    """MoE deferred-quant decode (fused router+norm, deferred grouped w13
+    input, scale-merge quant emit) vs the explicit-norm path and the CPU
+    oracle. Needs dim % 256 == 0 to activate."""
+    from dllama_amd.models.hip_model import HipTransformer
+    p = str(tmp_path / "moe256.m")
+    make_tiny_qwen3(p, moe=True, dim=256)
+    m = mf.ModelFile(p)
+    cfg = ModelConfig.from_header(m.header)
+    expl = HipTransformer.from_file(m, cfg)
+    expl.use_deferred = False
+    deferred = HipTransformer.from_file(m, cfg)
+    cpu = CpuTransformer(m, cfg)
+    prompt = [1, 2, 3]
+    expl.forward(torch.tensor(prompt), torch.arange(3))
+    deferred.forward(torch.tensor(prompt), torch.arange(3))
+    cpu.forward(torch.tensor(prompt), torch.arange(3))
+    for step in range(4):
+        t = torch.tensor([9 + step])
+        pp = torch.tensor([3 + step])
+        want = expl.forward(t, pp).cpu().clone()
+        got = deferred.forward(t, pp).cpu().clone()
+        oracle = cpu.forward(t, pp)
+        assert _rel_err(got[0], want[0]) < 2e-3, (step, _rel_err(got[0], want[0]))
+        assert _rel_err(got[0], oracle[0]) < 0.03, f"step {step}"
+        assert torch.equal(got.argmax(-1), want.argmax(-1)), f"step {step}"
