@@ -1007,6 +1007,108 @@ __global__ void k_attn_split(const float *__restrict__ q, int q_ld,
 
 }
 
+// GQA-grouped split variant: one workgroup owns ALL kv_mul query heads of
+// a kv head (wave w = query head kvh*kv_mul+w), so the K/V rows stream
+// through one CU's L1 once instead of kv_mul times through different XCD
+// L2s (4x HBM amplification for 8B's 32q/8kv at long context). Waves are
+// independent heads — no cross-wave merge; each writes its split partial
+// directly. The model scales S by kv_mul to keep the grid full.
+template <int VEC, typename KVT>
+__global__ void k_attn_split_gqa(const float *__restrict__ q, int q_ld,
+                                 const KVT *__restrict__ kc,
+                                 const KVT *__restrict__ vc,
+                                 const int *__restrict__ pos,
+                                 int n_heads0, int kv_mul, int kv_dim0,
+                                 float scale,
+                                 float *__restrict__ ml_scratch,
+                                 float *__restrict__ o_scratch) {
+    const int kvh = blockIdx.x;
+    const int b = blockIdx.y;
+    const int sp = blockIdx.z;
+    const int S = gridDim.z;
+    const int hd = VEC * WAVE;
+    const int plen = pos[0] + b + 1;
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int h0 = kvh * kv_mul + wave;
+    const int kv_off = kvh * hd;
+    const int VEC16 = VEC * 4;
+    const int lane16 = lane & 15;
+    const int group = lane >> 4;
+
+    float qreg[VEC16];
+    #pragma unroll
+    for (int v = 0; v < VEC16; v++)
+        qreg[v] = q[(int64_t)b * q_ld + h0 * hd + lane16 * VEC16 + v] * scale;
+
+    float m = -1e30f, l = 0.0f, o[VEC];
+    #pragma unroll
+    for (int v = 0; v < VEC; v++) o[v] = 0.0f;
+
+    // 16 consecutive timesteps per round (4 group-chunks x 4 unrolled);
+    // all waves walk the same rounds, so the wg's K/V reads coalesce in L1
+    for (int tb0 = sp * 16; tb0 < plen; tb0 += 16 * S) {
+        float su[4];
+        #pragma unroll
+        for (int u = 0; u < 4; u++) {
+            const int tg = tb0 + u * 4 + group;
+            float partial = 0.0f;
+            if (tg < plen) {
+                const KVT *krow = kc + (int64_t)tg * kv_dim0 + kv_off + lane16 * VEC16;
+                #pragma unroll
+                for (int v = 0; v < VEC16; v++)
+                    partial = fmaf(qreg[v], kv_f(krow[v]), partial);
+            }
+            su[u] = group16_reduce_sum(partial);
+        }
+        float s16[16];
+        float mn = m;
+        #pragma unroll
+        for (int u = 0; u < 4; u++) {
+            #pragma unroll
+            for (int gg = 0; gg < 4; gg++) {
+                float v = __shfl(su[u], gg * 16, WAVE);
+                if (tb0 + u * 4 + gg >= plen) v = -1e30f;
+                s16[4 * u + gg] = v;
+                mn = fmaxf(mn, v);
+            }
+        }
+        const float f = __expf(m - mn);
+        float w16[16];
+        float lsum = 0.0f;
+        #pragma unroll
+        for (int i = 0; i < 16; i++) {
+            w16[i] = __expf(s16[i] - mn);
+            lsum += w16[i];
+        }
+        l = l * f + lsum;
+        #pragma unroll
+        for (int v = 0; v < VEC; v++) o[v] *= f;
+        #pragma unroll
+        for (int u = 0; u < 4; u++) {
+            #pragma unroll
+            for (int gg = 0; gg < 4; gg++) {
+                const int t = tb0 + u * 4 + gg;
+                if (t >= plen) continue;
+                const KVT *vrow = vc + (int64_t)t * kv_dim0 + kv_off + lane * VEC;
+                #pragma unroll
+                for (int v = 0; v < VEC; v++)
+                    o[v] = fmaf(w16[4 * u + gg], kv_f(vrow[v]), o[v]);
+            }
+        }
+        m = mn;
+    }
+
+    const int64_t slot = ((int64_t)b * n_heads0 + h0) * S + sp;
+    if (lane == 0) {
+        ml_scratch[slot * 2] = m;
+        ml_scratch[slot * 2 + 1] = l;
+    }
+    #pragma unroll
+    for (int v = 0; v < VEC; v++)
+        o_scratch[slot * hd + lane * VEC + v] = o[v];
+}
+
 // Stage 2: combine the S split partials. QUANT=true additionally emits the
 // Q80 triple of the attention output directly (each workgroup owns a whole
 // head = hd/32 quant blocks), eliminating the separate cast kernel the
@@ -2581,6 +2683,51 @@ void attn(torch::Tensor q, int64_t q_ld, torch::Tensor kc, torch::Tensor vc,
     const dim3 cgrid(n_heads0, batch);
     const bool quant = zq.has_value();
     const bool kv16 = kc.scalar_type() == at::kHalf;
+    // GQA grouping: one wg per (kv head, split) with kv_mul waves — the
+    // K/V rows stream once through that CU's L1 instead of kv_mul times
+    // through different XCDs' L2 (4x HBM amplification at 8B's 32q/8kv)
+    if (kv_mul > 1 && kv_mul <= 8 && n_heads0 % kv_mul == 0 && splits > 1) {
+        const dim3 ggrid(n_heads0 / kv_mul, batch, splits);
+        const dim3 gblock(kv_mul * WAVE);
+        auto grun = [&](auto vec_const, auto kvt, auto kcp, auto vcp) {
+            constexpr int V = decltype(vec_const)::value;
+            using KVT = decltype(kvt);
+            hipLaunchKernelGGL((k_attn_split_gqa<V, KVT>), ggrid, gblock, 0,
+                               cur_stream(), q.data_ptr<float>(), (int)q_ld,
+                               kcp, vcp, pos.data_ptr<int>(), (int)n_heads0,
+                               (int)kv_mul, kv_dim0, scale,
+                               ml_scratch.data_ptr<float>(),
+                               o_scratch.data_ptr<float>());
+        };
+        auto gcomb = [&](auto vec_const) {
+            constexpr int V = decltype(vec_const)::value;
+            if (quant)
+                hipLaunchKernelGGL((k_attn_combine<V, true>), cgrid, dim3(V * WAVE),
+                                   0, cur_stream(), ml_scratch.data_ptr<float>(),
+                                   o_scratch.data_ptr<float>(), nullptr,
+                                   zq->data_ptr<int8_t>(), zs->data_ptr<float>(),
+                                   zbs->data_ptr<float>(), (int)n_heads0, (int)splits);
+            else
+                hipLaunchKernelGGL((k_attn_combine<V, false>), cgrid, dim3(V * WAVE),
+                                   0, cur_stream(), ml_scratch.data_ptr<float>(),
+                                   o_scratch.data_ptr<float>(), y.data_ptr<float>(),
+                                   nullptr, nullptr, nullptr, (int)n_heads0, (int)splits);
+        };
+        auto gdisp = [&](auto vec_const) {
+            if (kv16)
+                grun(vec_const, __half{},
+                     reinterpret_cast<const __half *>(kc.data_ptr<at::Half>()),
+                     reinterpret_cast<const __half *>(vc.data_ptr<at::Half>()));
+            else
+                grun(vec_const, 0.0f, (const float *)kc.data_ptr<float>(),
+                     (const float *)vc.data_ptr<float>());
+            gcomb(vec_const);
+        };
+        if (head_dim == 128) gdisp(std::integral_constant<int, 2>{});
+        else if (head_dim == 64) gdisp(std::integral_constant<int, 1>{});
+        else TORCH_CHECK(false, "unsupported head_dim ", head_dim);
+        return;
+    }
     // splits==1 + quant output: single fused kernel (normalize + Q80 emit
     // in the split kernel), no combine launch
     if (splits == 1 && quant) {
