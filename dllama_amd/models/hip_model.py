@@ -780,17 +780,13 @@ class HipTransformer:
     # ------------------------------------------------------------ graphs
 
     def _pick_splits(self, pos: int) -> int:
-        # S=8 short context, S=16 past the threshold (tools/attn_kv16_probe),
-        # scaled by kv_mul when the GQA-grouped split kernel applies (its
-        # grid is kv_heads x S, so S grows to keep the chip full). An S=1
-        # fused single-kernel path exists but measured SLOWER in-model
-        # (13.5 us vs the ~11 us pair: 32-workgroup grid) — never auto-picked.
-        c = self.cfg
-        kv_mul = c.n_heads0 // max(1, c.kv_dim0 // c.head_dim)
-        mult = kv_mul if 1 < kv_mul <= 8 and c.n_heads0 % kv_mul == 0 else 1
+        # S=8 short context, S=16 past the threshold (tools/attn_kv16_probe).
+        # Two measured-slower variants exist behind env flags and are never
+        # auto-picked: the S=1 fused single kernel (13.5 us vs the ~11 us
+        # pair) and the GQA-grouped split (DLLAMA_GQA_ATTN).
         if not self.adaptive_thresh:
             return self.attn_splits
-        return (8 if pos < self.adaptive_thresh else 16) * mult
+        return 8 if pos < self.adaptive_thresh else 16
 
     def _set_attn_splits(self, s: int):
         """Switch the flash-decode K-split count mid-stream (long-context

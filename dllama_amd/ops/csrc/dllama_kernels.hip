@@ -2683,10 +2683,15 @@ void attn(torch::Tensor q, int64_t q_ld, torch::Tensor kc, torch::Tensor vc,
     const dim3 cgrid(n_heads0, batch);
     const bool quant = zq.has_value();
     const bool kv16 = kc.scalar_type() == at::kHalf;
-    // GQA grouping: one wg per (kv head, split) with kv_mul waves — the
-    // K/V rows stream once through that CU's L1 instead of kv_mul times
-    // through different XCDs' L2 (4x HBM amplification at 8B's 32q/8kv)
-    if (kv_mul > 1 && kv_mul <= 8 && n_heads0 % kv_mul == 0 && splits > 1) {
+    // GQA grouping experiment: one wg per (kv head, split) with kv_mul
+    // waves sharing the K/V stream through L1. Measured ~20% SLOWER
+    // end-to-end (448 vs 545 tok/s short-ctx 8B; MoE 366->255) — the L2
+    // already absorbs the GQA re-reads and the grown combine fan-in costs
+    // more than the saved traffic. Kept behind DLLAMA_GQA_ATTN=1.
+    static const bool env_gqa =
+        std::getenv("DLLAMA_GQA_ATTN") && atoi(std::getenv("DLLAMA_GQA_ATTN")) == 1;
+    if (env_gqa && kv_mul > 1 && kv_mul <= 8 && n_heads0 % kv_mul == 0
+        && splits > 1) {
         const dim3 ggrid(n_heads0 / kv_mul, batch, splits);
         const dim3 gblock(kv_mul * WAVE);
         auto grun = [&](auto vec_const, auto kvt, auto kcp, auto vcp) {

@@ -134,9 +134,7 @@ def test_adaptive_splits_recapture(tiny_path):
         want = eager.forward(t, p).cpu().clone()
         got = adaptive.forward(t, p).cpu().clone()
         assert _rel_err(got[0], want[0]) < 1e-4, f"step {step}"
-    kv_mul = cfg.n_heads0 // max(1, cfg.kv_dim0 // cfg.head_dim)
-    mult = kv_mul if 1 < kv_mul <= 8 else 1
-    assert adaptive.attn_splits == 16 * mult
+    assert adaptive.attn_splits == 16
 
 
 def test_tp_path_matches_plain_world1(tiny_path):
