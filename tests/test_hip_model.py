@@ -255,3 +255,53 @@ def test_deferred_quant_moe(tmp_path):
         assert _rel_err(got[0], want[0]) < 2e-3, (step, _rel_err(got[0], want[0]))
         assert _rel_err(got[0], oracle[0]) < 0.03, f"step {step}"
         assert torch.equal(got.argmax(-1), want.argmax(-1)), f"step {step}"
+
+
+def test_rccl_collectives_capture_in_graph(tiny_path):
+    """Real RCCL (nccl backend) collectives inside a captured hipGraph on
+    hardware: world=1 process group, force_sync TP path (all_gather of the
+    Q80 wire + logits gather are genuine RCCL calls even at world 1).
+    De-risks the driver's 8-GPU graph-captured decode before it ever runs."""
+    import os
+    import subprocess
+    import sys
+    code = f'''
+import os, torch
+os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+os.environ.setdefault("MASTER_PORT", "29581")
+os.environ["WORLD_SIZE"] = "1"
+os.environ["RANK"] = "0"
+os.environ["LOCAL_RANK"] = "0"
+import torch.distributed as dist
+dist.init_process_group("nccl", rank=0, world_size=1)
+from dllama_amd import model_file as mf
+from dllama_amd.models.config import ModelConfig
+from dllama_amd.models.hip_model import HipTransformer
+from dllama_amd.parallel.comm import DistComm
+m = mf.ModelFile({tiny_path!r})
+cfg = ModelConfig.from_header(m.header)
+plain = HipTransformer.from_file(m, cfg)
+tp = HipTransformer.from_file(m, cfg, comm=DistComm(), force_sync=True)
+prompt = [3, 17, 101]
+import torch as T
+want = plain.forward(T.tensor(prompt), T.arange(3)).cpu()
+got = tp.forward(T.tensor(prompt), T.arange(3)).cpu()
+assert T.equal(got.argmax(-1), want.argmax(-1))
+tp.greedy_feedback = True
+tp.capture_decode_graph()   # RCCL all-gathers captured inside the graph
+plain.greedy_feedback = True
+plain.capture_decode_graph()
+plain.pos.fill_(3); tp.pos.fill_(3)
+plain.tokens[0] = 7; tp.tokens[0] = 7
+for step in range(4):
+    plain._graph.replay()
+    tp._graph.replay()
+    T.cuda.synchronize()
+    assert int(plain.tokens[0]) == int(tp.tokens[0]), step
+dist.destroy_process_group()
+print("RCCL_GRAPH_OK")
+'''
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=300,
+                       env={**os.environ, "HSA_ENABLE_IPC_MODE_LEGACY": "0"})
+    assert "RCCL_GRAPH_OK" in r.stdout, (r.stdout[-2000:], r.stderr[-2000:])
