@@ -347,8 +347,8 @@ class HipTransformer:
             row_bytes = c.dim + 2 * nb_dim
             out = self.sync_out[: NB * row_bytes]
             if getattr(self, "fused_sync", False):
-                # round-2 experimental: quantize straight into the wire
-                # buffer, one pass (DLLAMA_FUSED_SYNC=1)
+                # quantize straight into the wire buffer, one pass
+                # (default; DLLAMA_FUSED_SYNC=0 splits it)
                 self.k.sync_quant_pack(self.partial[:NB], out)
             else:
                 q = self.xq  # reuse dim-sized quant buffer

@@ -1159,11 +1159,11 @@ __global__ void k_attn_combine(const float *__restrict__ ml_scratch,
 // Grouped (MoE) variant: weights [n_experts, d, ...]; slot s uses expert
 // expert_idx[s] and input row slot_batch[s]; y [S, d]
 // (reference 3-D expert matmul with index indirection, nn-core.hpp:209-213).
-// LPP = lanes per 2-row pair. 64 is the production default; smaller LPP
-// (round-2 experimental, DLLAMA_MOE_V2=1) tiles a wave into 64/LPP
-// row-pair groups so every lane stays busy at MoE shapes where nbp < 64
-// (Qwen3-30B w2: nbp=12 left 52 of 64 lanes idle -> the ~5x-off-stream
-// grouped GEMV; see tools/moe_gemv_probe.hip).
+// LPP = lanes per 2-row pair. The lane-tiled default (since round 2)
+// picks LPP = smallest pow2 >= nbp so every lane stays busy at MoE shapes
+// where nbp < 64 (Qwen3-30B w2: nbp=12 left 52 of 64 lanes idle -> the
+// ~5x-off-stream round-1 grouped GEMV; see tools/moe_gemv_probe.hip).
+// DLLAMA_MOE_V2=0 reverts to the whole-wave (LPP=64) layout.
 template <int LPP, bool GATE = false>
 __global__ void k_q40_gemv_grouped(const uint8_t *__restrict__ qs,
                                    const __half *__restrict__ scales,
@@ -1366,8 +1366,9 @@ k_q40_gemm(const uint8_t *__restrict__ qs,
     }
 }
 
-// EXPERIMENTAL round-2 GEMM (DLLAMA_GEMM_V2=1; see tools/gemm_v2_probe.hip
-// for the standalone A/B harness). Differences vs k_q40_gemm:
+// The default prefill GEMM since round 2 (DLLAMA_GEMM_V2=0 reverts; see
+// tools/gemm_v2_probe.hip for the standalone A/B harness). Differences vs
+// the round-1 k_q40_gemm:
 //   - activation fragments + x-scales staged in LDS per 8-block chunk,
 //     loaded once per workgroup and shared by all 4 waves,
 //   - weight uint4 tiles prefetched through a 4-deep register ring,
@@ -2166,6 +2167,7 @@ __global__ void k_silu_mul(const float *__restrict__ a,
 //  k_merge_add, the reference merge-add-forward-q80-f32.comp equivalent).
 // fused quantize-into-wire: one pass replaces k_q80_quantize + k_sync_pack
 // for the TP sync (the wire consumer k_merge_add needs no blocksum).
+// Default since round 2 (DLLAMA_FUSED_SYNC=0 splits it again).
 // EXPERIMENTAL round-2 path: dispatched only under DLLAMA_FUSED_SYNC=1.
 __global__ void k_sync_quant_pack(const float *__restrict__ x,
                                   uint8_t *__restrict__ buf,
