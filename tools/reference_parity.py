@@ -131,8 +131,14 @@ def run_ours(model, tok_path, prompt, steps) -> str:
     from dllama_amd.tokenizer import Sampler, Tokenizer
     m = mf.ModelFile(model, max_seq_len=256)
     tok = Tokenizer(tok_path)
-    eng = InferenceEngine(CpuTransformer(m, ModelConfig.from_header(m.header)),
-                          tok, Sampler(m.header.vocab_size, 0.0, 0.9, 1))
+    cfg = ModelConfig.from_header(m.header)
+    if USE_HIP:
+        from dllama_amd.models.hip_model import HipTransformer
+        net = HipTransformer.from_file(m, cfg)
+    else:
+        net = CpuTransformer(m, cfg)
+    eng = InferenceEngine(net, tok,
+                          Sampler(m.header.vocab_size, 0.0, 0.9, 1))
     tokens = tok.encode(prompt)
     emulated = tokens[:-1] + [0]
     out, _ = eng.generate(emulated, steps - len(tokens) + 1)
@@ -239,8 +245,12 @@ def run_ours_chat(model, tok_path, sys_prompt, users):
     return turns
 
 
+USE_HIP = "--hip" in sys.argv  # greedy-compare the GPU path itself
+
+
 def main():
-    workdir = sys.argv[1] if len(sys.argv) > 1 else tempfile.mkdtemp(prefix="parity")
+    argv = [a for a in sys.argv[1:] if a != "--hip"]
+    workdir = argv[0] if argv else tempfile.mkdtemp(prefix="parity")
     os.makedirs(workdir, exist_ok=True)
     binary = build_reference(workdir)
     rc = 0
@@ -265,6 +275,8 @@ def main():
             print(f"❌ [{arch}] MISMATCH after {n} matching chars")
             rc = 1
 
+    if USE_HIP:
+        return rc  # chat-mode parity is the CPU harness's job
     # chat-mode parity (llama): template generation, multi-turn KV
     # continuity, EOS detection — vs the reference's interactive loop
     model, tok = make_ascii_assets(workdir, arch="llama")
