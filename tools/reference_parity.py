@@ -25,6 +25,12 @@ REF_SRC = "/root/reference"
 
 
 def build_reference(workdir: str) -> str:
+    # a prebuilt binary (tools/_refbin/dllama, gitignored) lets the harness
+    # run on GPU boxes where /root/reference is not present
+    prebuilt = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                            "_refbin", "dllama")
+    if not os.path.isdir(REF_SRC) and os.path.exists(prebuilt):
+        return prebuilt
     ref = os.path.join(workdir, "refbuild")
     if not os.path.exists(ref):
         shutil.copytree(REF_SRC, ref)
