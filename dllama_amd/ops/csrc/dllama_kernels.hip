@@ -648,7 +648,10 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
         for (int b = 0; b < NB; b++) acc[r][b] = 0.0f;
 
     float invb[NB];
-    if (PRO) {
+    if (PRO == 1) {
+        // PRO1 needs inv for the in-loop requantize; PRO2 only applies it
+        // in the epilogue — computing it there keeps the ssq load off the
+        // kernel's critical start (ahead of the first weight loads)
         #pragma unroll
         for (int b = 0; b < NB; b++)
             invb[b] = rsqrtf(ssq_total_wave(ssq_in, b, lane) / n + eps);
@@ -723,6 +726,8 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
     #pragma unroll
     for (int b = 0; b < NB; b++) {
         float v[RPW];
+        if (PRO == 2)
+            invb[b] = rsqrtf(ssq_total_wave(ssq_in, b, lane) / n + eps);
         #pragma unroll
         for (int r = 0; r < RPW; r++) {
             v[r] = wave_reduce_sum(acc[r][b]);
