@@ -114,7 +114,7 @@ for d, n, tag in ((6144, 4096, "qkv"), (4096, 14336, "w2"), (28672, 4096, "w13")
     qs, sc = mklin(d, n)
     xq, xs, _ = mkx(32, n)
     y = torch.zeros(32, d, device=dev)
-    t1 = bench(lambda: k.q40_gemm(qs, sc, xq, xs, y, 32, part), n=32)
+    t1 = bench(lambda: k.q40_gemm(qs, sc, xq, xs, y, 32, part, variant=0), n=32)
     t2 = bench(lambda: k.q40_gemm(qs, sc, xq, xs, y, 32, part, variant=1), n=32)
     ideal = (d * n * 0.5625) / 6.3e3 / 1000
     print(f"gemm {tag:4s} v1={t1:7.2f}us v2={t2:7.2f}us ideal={ideal:6.2f}us "
@@ -130,7 +130,7 @@ for d, n, ks, tag in ((1536, 2048, 8, "w13"), (2048, 768, 1, "w2")):
     xq, xs, xbs = mkx(max(1, nx), n)
     idx = torch.arange(8, dtype=torch.int32, device=dev) % E
     y = torch.zeros(8, d, device=dev)
-    t1 = bench(lambda: k.q40_gemv_grouped(qs, sc, xq, xs, xbs, idx, y, ks), n=64)
+    t1 = bench(lambda: k.q40_gemv_grouped(qs, sc, xq, xs, xbs, idx, y, ks, variant=0), n=64)
     t2 = bench(lambda: k.q40_gemv_grouped(qs, sc, xq, xs, xbs, idx, y, ks, variant=1), n=64)
     ideal = (8 * d * n * 0.5625) / 6.3e3 / 1000
     print(f"grouped {tag:4s} v1={t1:7.2f}us v2={t2:7.2f}us ideal={ideal:6.2f}us "
