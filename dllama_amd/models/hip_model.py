@@ -392,7 +392,10 @@ class HipTransformer:
         stays on device — this function is graph-capturable."""
         c, k = self.cfg, self.k
         NB = _pow2_batch(B)
-        if (B == 1 and not self.tp_path and c.dim % 32 == 0
+        tp_def_ok = (not self.tp_path
+                     or (not c.is_moe and c.sync_type == Q80
+                         and c.dim % 256 == 0))
+        if (B == 1 and c.dim % 32 == 0 and tp_def_ok
                 and (not c.is_moe
                      or (c.dim % 256 == 0 and c.n_active_experts <= 16))
                 and getattr(self, "use_deferred", True)
@@ -634,9 +637,18 @@ class HipTransformer:
                    self.attn_splits, self.attn_ml, self.attn_o,
                    self.attn_counter, self.zq.q, self.zq.s, self.zq.bs)
             # wo: residual fold + deferred Q80 emit of x*norm1 for the FFN
-            k.q40_gemv_resid_q(lw["wo"].qs, lw["wo"].scales, self.zq.q,
-                               self.zq.s, self.zq.bs, x, self.ssq[slot + 1],
-                               lw["norm1"], self.xq.q, self.xq.s, self.xq.bs)
+            # (TP: the epilogue packs the Q80 wire instead; merge-add after
+            # the all-gather folds the residual and emits the deferred quant
+            # — reference SYNC_NODE_SLICES + OP_MERGE_ADD, one fewer launch
+            # on each side of the collective)
+            if self.tp_path:
+                self._tp_proj_deferred(lw["wo"], self.zq, self.ssq[slot + 1],
+                                       lw["norm1"])
+            else:
+                k.q40_gemv_resid_q(lw["wo"].qs, lw["wo"].scales, self.zq.q,
+                                   self.zq.s, self.zq.bs, x,
+                                   self.ssq[slot + 1], lw["norm1"],
+                                   self.xq.q, self.xq.s, self.xq.bs)
             slot += 1
             wn = self.final_norm if l == last else self.layers[l + 1]["norm0"]
             if c.is_moe:
@@ -672,18 +684,43 @@ class HipTransformer:
                              c.hidden_act == HIDDEN_ACT_GELU)
                 # w2: residual fold + deferred emit for the NEXT layer's
                 # norm0 (final_norm for logits after the last layer)
-                k.q40_gemv_resid_q(lw["w2"].qs, lw["w2"].scales, self.dq.q,
-                                   self.dq.s, self.dq.bs, x,
-                                   self.ssq[slot + 1], wn, self.xq.q,
-                                   self.xq.s, self.xq.bs)
+                if self.tp_path:
+                    self._tp_proj_deferred(lw["w2"], self.dq,
+                                           self.ssq[slot + 1], wn)
+                else:
+                    k.q40_gemv_resid_q(lw["w2"].qs, lw["w2"].scales,
+                                       self.dq.q, self.dq.s, self.dq.bs, x,
+                                       self.ssq[slot + 1], wn, self.xq.q,
+                                       self.xq.s, self.xq.bs)
             slot += 1
-        use_amax = self.greedy_feedback
+        use_amax = self.greedy_feedback and not self.tp_path
         k.q40_gemv(self.wcls.qs, self.wcls.scales, self.xq.q, self.xq.s,
                    self.xq.bs, self.logits0, 1,
                    self.amax_scratch if use_amax else None,
                    ssq_in=self.ssq[slot], eps=eps)
-        if use_amax:
-            k.token_from_argmax(self.tokens, self.amax_scratch, self.amax_blocks)
+        if self.tp_path:
+            self.comm.all_gather(self.logits_gather[1], self.logits0[:1])
+        if self.greedy_feedback:
+            if use_amax:
+                k.token_from_argmax(self.tokens, self.amax_scratch,
+                                    self.amax_blocks)
+            else:
+                k.argmax_token(self.tokens, self.logits_gather[1].view(-1),
+                               self.argmax_scratch_full)
+
+    def _tp_proj_deferred(self, lin: Linear, qb: QuantBuf, ssq_slot,
+                          wnorm: torch.Tensor):
+        """TP down-projection, deferred form: the GEMV epilogue emits the
+        Q80 wire directly, the all-gathered slices merge into x with the
+        next matmul's deferred quant emitted in the same kernel."""
+        c, k = self.cfg, self.k
+        row_bytes = c.dim + 2 * (c.dim // QB)
+        k.q40_gemv_pack(lin.qs, lin.scales, qb.q, qb.s, qb.bs,
+                        self.sync_out[:row_bytes])
+        inb = self.sync_in[1]
+        self.comm.all_gather(inb, self.sync_out[:row_bytes])
+        k.merge_add_q(self.x[:1], inb.view(c.world, 1, row_bytes), ssq_slot,
+                      wnorm, self.xq.q, self.xq.s, self.xq.bs)
 
     def _moe_ffn(self, B: int, NB: int, lw: dict, slot: int):
         """Router + grouped expert GEMVs (reference llm.cpp:450-487);

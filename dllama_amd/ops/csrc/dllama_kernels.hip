@@ -541,6 +541,8 @@ __device__ __forceinline__ int q40_block_dot(const uint4 wq, const int4 x0,
 // removes whole kernels from the per-layer chain (the reference runs each
 // as its own op, llm.cpp:263-557).
 #define EPI_NONE 0
+#define EPI_PACK 4     // TP: plain dot + Q80 WIRE emit (codes then f16 scales,
+                       // the all-gather payload) straight from the epilogue
 #define EPI_RESID_Q 3  // EPI_RESID + deferred-scale Q80 emit of x*wnorm (one
                        // block per 16-wave wg; consumer applies inv_rms, PRO==2)
 #define EPI_RESID 1  // x[b,row] += v; accumulate sum(x'^2) into 16-way-spread
@@ -788,6 +790,8 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
                     x_resid[(int64_t)b * d + row] = xv;
                     ssq_local[b] += xv * xv;
                     svq[wid * RPW + r] = xv * wnorm[row];
+                } else if (EPI == EPI_PACK) {
+                    svq[wid * RPW + r] = v[r];  // partial, packed below
                 } else {
                     y[(int64_t)b * d + row] = v[r];
                     if (NB == 1) wave_best = max(wave_best, argmax_pack(v[r], row));
@@ -804,6 +808,27 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
             unsigned long long best = wb[0];
             for (int i = 1; i < wpb; i++) best = max(best, wb[i]);
             amax_scratch[blockIdx.x] = best;
+        }
+    }
+    if constexpr (EPI == EPI_PACK) {
+        // wire layout (all-gather payload, B=1): int8 codes [d] then f16
+        // block scales [d/16 bytes] — what k_sync_quant_pack produces and
+        // k_merge_add consumes (no blocksum on the wire)
+        __syncthreads();  // publish svq
+        if (threadIdx.x < 32 && (int)(blockIdx.x * 32 + threadIdx.x) < d) {
+            const float vq = svq[threadIdx.x];
+            const float amax = group32_reduce_max(fabsf(vq));
+            const float dd = amax / 127.0f;
+            const float qinv = dd > 0.0f ? 1.0f / dd : 0.0f;
+            uint8_t *buf = reinterpret_cast<uint8_t *>(oqq);
+            buf[blockIdx.x * QB + threadIdx.x] =
+                (uint8_t)(int8_t)rintf(vq * qinv);
+            if (threadIdx.x == 0) {
+                const __half h = __float2half(dd);
+                const uint16_t u = *reinterpret_cast<const uint16_t *>(&h);
+                buf[d + 2 * blockIdx.x] = (uint8_t)(u & 0xFF);
+                buf[d + 2 * blockIdx.x + 1] = (uint8_t)(u >> 8);
+            }
         }
     }
     if constexpr (EPI == EPI_RESID_Q) {
@@ -2253,6 +2278,54 @@ __global__ void k_merge_add(float *__restrict__ x,
     }
 }
 
+// merge_add + DEFERRED Q80 emit of x*wnorm (the next matmul's input; the
+// PRO==2 consumer applies inv_rms): completes the TP all-reduce AND
+// replaces the following norm_quant launch. One 256-thread wg per 256
+// elements = 8 wg-local quant blocks.
+__global__ void k_merge_add_q(float *__restrict__ x,
+                              const uint8_t *__restrict__ bufs,
+                              float *__restrict__ ssq,
+                              const float *__restrict__ wnorm,
+                              int8_t *__restrict__ oq,
+                              float *__restrict__ os,
+                              float *__restrict__ obs,
+                              int world, int n, int rows) {
+    const int nb = n / QB;
+    const int row_bytes = n + 2 * nb;
+    const int r = blockIdx.y;
+    const int c = blockIdx.x * 256 + threadIdx.x;
+    float acc = x[(int64_t)r * n + c];
+    for (int w = 0; w < world; w++) {
+        const uint8_t *row = bufs + ((int64_t)w * rows + r) * row_bytes;
+        const int8_t qv = (int8_t)row[c];
+        const uint16_t u = (uint16_t)row[n + 2 * (c / QB)]
+                         | ((uint16_t)row[n + 2 * (c / QB) + 1] << 8);
+        const __half h = *reinterpret_cast<const __half *>(&u);
+        acc = fmaf((float)qv, __half2float(h), acc);
+    }
+    x[(int64_t)r * n + c] = acc;
+    float local = acc * acc;
+    local = wave_reduce_sum(local);
+    __shared__ float red[4];
+    const int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    if (lane == 0) red[wid] = local;
+    __syncthreads();
+    if (threadIdx.x == 0)
+        atomicAdd(ssq + (r * SSQ_SPREAD + (blockIdx.x & (SSQ_SPREAD - 1))) * SSQ_PAD,
+                  red[0] + red[1] + red[2] + red[3]);
+    const float v = acc * wnorm[c];
+    const float amax = group32_reduce_max(fabsf(v));
+    const float dd = amax / 127.0f;
+    const float qinv = dd > 0.0f ? 1.0f / dd : 0.0f;
+    const float qf = rintf(v * qinv);
+    oq[(int64_t)r * n + c] = (int8_t)qf;
+    const float bsum = group32_reduce_sum(qf);
+    if ((threadIdx.x & 31) == 0) {
+        os[(int64_t)r * nb + c / QB] = dd;
+        obs[(int64_t)r * nb + c / QB] = bsum;
+    }
+}
+
 // x += y (residual merge for the f32/TP=1 path)
 __global__ void k_add(float *__restrict__ x, const float *__restrict__ y, int64_t n) {
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
@@ -2342,8 +2415,8 @@ static void gemv_launch(torch::Tensor &qs, torch::Tensor &scales, torch::Tensor 
     const int n = qs.size(1) * 2;
     TORCH_CHECK(PRO == 1 || xq.size(-1) == n, "x width mismatch");
     TORCH_CHECK(n % QB == 0, "n must be a multiple of 32");
-    // EPI_RESID_Q: 16 waves x RPW2 = 32 output rows per wg = one Q80 block
-    const int waves_per_block = EPI == EPI_RESID_Q ? 16 : 4;
+    // EPI_RESID_Q / EPI_PACK: 16 waves x RPW2 = 32 rows per wg = one block
+    const int waves_per_block = (EPI == EPI_RESID_Q || EPI == EPI_PACK) ? 16 : 4;
     const dim3 block(waves_per_block * WAVE);
     auto launch = [&](auto nb_const, auto rpw_const) {
         constexpr int RPW = decltype(rpw_const)::value;
@@ -2435,6 +2508,19 @@ void q40_gemv(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
     } else {
         gemv_launch<EPI_NONE, 0>(qs, scales, xq, xs, xbs, y.data_ptr<float>(), batch, e);
     }
+}
+
+void q40_gemv_pack(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
+                   torch::Tensor xs, torch::Tensor xbs, torch::Tensor wire) {
+    // TP down-projection: plain partial dot + Q80 WIRE emit in one launch
+    // (replaces gemv-into-partial + sync_quant_pack). B=1 decode only.
+    CHECK_CUDA(qs); CHECK_CONT(qs); CHECK_CONT(xq);
+    const int d = qs.size(0);
+    TORCH_CHECK(d % 32 == 0, "gemv_pack needs d % 32 == 0");
+    GemvEpi e;
+    e.oqq = reinterpret_cast<int8_t *>(wire.data_ptr<uint8_t>());
+    e.force_rpw2 = true;
+    gemv_launch<EPI_PACK, 0>(qs, scales, xq, xs, xbs, nullptr, 1, e);
 }
 
 void q40_gemv_resid_q(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
@@ -3202,6 +3288,23 @@ void sync_quant_pack(torch::Tensor x, torch::Tensor buf) {
                        buf.data_ptr<uint8_t>(), n, blocks);
 }
 
+void merge_add_q(torch::Tensor x, torch::Tensor bufs, torch::Tensor ssq,
+                 torch::Tensor wnorm, torch::Tensor oq, torch::Tensor os,
+                 torch::Tensor obs) {
+    // bufs [world, rows, row_bytes] (all-gather output); x [rows, n]
+    CHECK_CUDA(x); CHECK_CONT(bufs);
+    const int world = bufs.size(0);
+    const int rows = bufs.size(1);
+    const int n = x.size(-1);
+    TORCH_CHECK(n % 256 == 0, "merge_add_q needs dim % 256 == 0");
+    hipLaunchKernelGGL(k_merge_add_q, dim3(n / 256, rows), dim3(256), 0,
+                       cur_stream(), x.data_ptr<float>(),
+                       bufs.data_ptr<uint8_t>(), ssq.data_ptr<float>(),
+                       wnorm.data_ptr<float>(), oq.data_ptr<int8_t>(),
+                       os.data_ptr<float>(), obs.data_ptr<float>(),
+                       world, n, rows);
+}
+
 void merge_add(torch::Tensor x, torch::Tensor bufs,
                c10::optional<torch::Tensor> ssq = c10::nullopt) {
     CHECK_CUDA(x);
@@ -3349,6 +3452,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("eps") = 0.0);
     m.def("q40_gemv_resid", &q40_gemv_resid);
     m.def("q40_gemv_resid_q", &q40_gemv_resid_q);
+    m.def("q40_gemv_pack", &q40_gemv_pack);
+    m.def("merge_add_q", &merge_add_q);
     m.def("q40_matmul_cpu", &q40_matmul_cpu);
     m.def("q40_gemm", &q40_gemm, py::arg("qs"), py::arg("scales"),
           py::arg("xq"), py::arg("xs"), py::arg("y"), py::arg("batch"),

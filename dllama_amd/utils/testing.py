@@ -30,8 +30,8 @@ def make_byte_tokenizer(path: str, chat_template: str | None = None) -> None:
 
 
 def make_tiny_llama(path: str, seed: int = 7, vocab_size: int = 512,
-                    seq_len: int = 128) -> mf.LlmHeader:
-    h = mf.LlmHeader(arch_type=mf.ARCH_LLAMA, dim=64, hidden_dim=128, n_layers=2,
+                    seq_len: int = 128, dim: int = 64) -> mf.LlmHeader:
+    h = mf.LlmHeader(arch_type=mf.ARCH_LLAMA, dim=dim, hidden_dim=128, n_layers=2,
                      n_heads=4, n_kv_heads=2, head_dim=64, vocab_size=vocab_size,
                      seq_len=seq_len, rope_theta=10000, rope_type=mf.ROPE_LLAMA)
     h.finalize()

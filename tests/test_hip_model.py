@@ -305,3 +305,33 @@ print("RCCL_GRAPH_OK")
                        text=True, timeout=300,
                        env={**os.environ, "HSA_ENABLE_IPC_MODE_LEGACY": "0"})
     assert "RCCL_GRAPH_OK" in r.stdout, (r.stdout[-2000:], r.stderr[-2000:])
+
+
+def test_tp_deferred_path_world1(tmp_path):
+    """TP deferred decode (EPI_PACK wire emit -> all-gather -> merge_add_q
+    with deferred quant) vs the plain world=1 path, at a dim that activates
+    it (dim % 256 == 0)."""
+    from dllama_amd.models.hip_model import HipTransformer
+    p = str(tmp_path / "l256.m")
+    make_tiny_llama(p, vocab_size=256, dim=256)
+    m = mf.ModelFile(p)
+    cfg = ModelConfig.from_header(m.header)
+    plain = HipTransformer.from_file(m, cfg)
+    tp = HipTransformer.from_file(m, cfg, force_sync=True)
+    prompt = [3, 17, 101]
+    want = plain.forward(torch.tensor(prompt), torch.arange(3)).cpu().clone()
+    got = tp.forward(torch.tensor(prompt), torch.arange(3)).cpu().clone()
+    assert _rel_err(got, want) < 0.02, _rel_err(got, want)
+    assert torch.equal(got.argmax(-1), want.argmax(-1))
+    # graph-captured deferred TP decode with greedy feedback
+    tp.greedy_feedback = True
+    tp.capture_decode_graph()
+    plain.greedy_feedback = True
+    plain.capture_decode_graph()
+    for mdl in (plain, tp):
+        mdl.pos.fill_(3)
+        mdl.tokens[0] = 7
+    for step in range(5):
+        plain._graph.replay()
+        tp._graph.replay()
+        assert int(plain.tokens[0]) == int(tp.tokens[0]), f"step {step}"
