@@ -393,8 +393,7 @@ class HipTransformer:
         c, k = self.cfg, self.k
         NB = _pow2_batch(B)
         tp_def_ok = (not self.tp_path
-                     or (not c.is_moe and c.sync_type == Q80
-                         and c.dim % 256 == 0))
+                     or (c.sync_type == Q80 and c.dim % 256 == 0))
         if (B == 1 and c.dim % 32 == 0 and tp_def_ok
                 and (not c.is_moe
                      or (c.dim % 256 == 0 and c.n_active_experts <= 16))
@@ -671,9 +670,22 @@ class HipTransformer:
                 k.q40_gemv_grouped(lw["w2"].qs, lw["w2"].scales, self.moe_dq.q,
                                    self.moe_dq.s, self.moe_dq.bs,
                                    self.moe_idx[:ka], self.moe_y, 1)
-                k.scale_merge_add_q(x[:1], self.moe_y, self.moe_wts,
-                                    self.ssq[slot + 1], wn, self.xq.q,
-                                    self.xq.s, self.xq.bs, 1, ka)
+                if self.tp_path:
+                    # expert-weighted sum packed straight into the Q80 wire;
+                    # merge-add after the gather folds residual + emits the
+                    # next deferred quant
+                    rb = c.dim + 2 * (c.dim // QB)
+                    k.scale_merge_pack(self.moe_y, self.moe_wts,
+                                       self.sync_out[:rb], 1, ka, c.dim)
+                    inb = self.sync_in[1]
+                    self.comm.all_gather(inb, self.sync_out[:rb])
+                    k.merge_add_q(x[:1], inb.view(c.world, 1, rb),
+                                  self.ssq[slot + 1], wn, self.xq.q,
+                                  self.xq.s, self.xq.bs)
+                else:
+                    k.scale_merge_add_q(x[:1], self.moe_y, self.moe_wts,
+                                        self.ssq[slot + 1], wn, self.xq.q,
+                                        self.xq.s, self.xq.bs, 1, ka)
             else:
                 k.q40_gemv(lw["w13"].qs, lw["w13"].scales, self.xq.q,
                            self.xq.s, self.xq.bs, self.ff_out, 1,

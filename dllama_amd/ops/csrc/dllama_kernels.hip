@@ -2278,6 +2278,34 @@ __global__ void k_merge_add(float *__restrict__ x,
     }
 }
 
+// TP MoE epilogue: expert-weighted sum of the grouped-GEMV outputs packed
+// straight into the Q80 wire (the all-gather payload) — replaces
+// scale_merge + sync_quant_pack on the TP critical path. One 256-thread wg
+// per 256 elements (8 wire blocks), B=1 decode.
+__global__ void k_scale_merge_pack(const float *__restrict__ y,
+                                   const float *__restrict__ wts,
+                                   uint8_t *__restrict__ buf,
+                                   int n, int topk) {
+    const int b = blockIdx.y;
+    const int i = blockIdx.x * 256 + threadIdx.x;
+    float acc = 0.0f;
+    for (int s = 0; s < topk; s++)
+        acc = fmaf(wts[(int64_t)b * topk + s],
+                   y[((int64_t)b * topk + s) * n + i], acc);
+    const float amax = group32_reduce_max(fabsf(acc));
+    const float dd = amax / 127.0f;
+    const float qinv = dd > 0.0f ? 1.0f / dd : 0.0f;
+    const int row_bytes = n + 2 * (n / QB);
+    uint8_t *row = buf + (int64_t)b * row_bytes;
+    row[i] = (uint8_t)(int8_t)rintf(acc * qinv);
+    if ((threadIdx.x & 31) == 0) {
+        const __half h = __float2half(dd);
+        const uint16_t u = *reinterpret_cast<const uint16_t *>(&h);
+        row[n + 2 * (i / QB)] = (uint8_t)(u & 0xFF);
+        row[n + 2 * (i / QB) + 1] = (uint8_t)(u >> 8);
+    }
+}
+
 // merge_add + DEFERRED Q80 emit of x*wnorm (the next matmul's input; the
 // PRO==2 consumer applies inv_rms): completes the TP all-reduce AND
 // replaces the following norm_quant launch. One 256-thread wg per 256
@@ -3288,6 +3316,15 @@ void sync_quant_pack(torch::Tensor x, torch::Tensor buf) {
                        buf.data_ptr<uint8_t>(), n, blocks);
 }
 
+void scale_merge_pack(torch::Tensor y, torch::Tensor wts, torch::Tensor wire,
+                      int64_t batch, int64_t topk, int64_t n) {
+    CHECK_CUDA(y);
+    TORCH_CHECK(n % 256 == 0, "scale_merge_pack needs dim % 256 == 0");
+    hipLaunchKernelGGL(k_scale_merge_pack, dim3(n / 256, batch), dim3(256), 0,
+                       cur_stream(), y.data_ptr<float>(), wts.data_ptr<float>(),
+                       wire.data_ptr<uint8_t>(), (int)n, (int)topk);
+}
+
 void merge_add_q(torch::Tensor x, torch::Tensor bufs, torch::Tensor ssq,
                  torch::Tensor wnorm, torch::Tensor oq, torch::Tensor os,
                  torch::Tensor obs) {
@@ -3454,6 +3491,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("q40_gemv_resid_q", &q40_gemv_resid_q);
     m.def("q40_gemv_pack", &q40_gemv_pack);
     m.def("merge_add_q", &merge_add_q);
+    m.def("scale_merge_pack", &scale_merge_pack);
     m.def("q40_matmul_cpu", &q40_matmul_cpu);
     m.def("q40_gemm", &q40_gemm, py::arg("qs"), py::arg("scales"),
           py::arg("xq"), py::arg("xs"), py::arg("y"), py::arg("batch"),

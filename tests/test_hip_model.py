@@ -335,3 +335,25 @@ def test_tp_deferred_path_world1(tmp_path):
         plain._graph.replay()
         tp._graph.replay()
         assert int(plain.tokens[0]) == int(tp.tokens[0]), f"step {step}"
+
+
+def test_tp_deferred_moe_world1(tmp_path):
+    """TP deferred MoE decode (router+norm fused, scale_merge_pack wire
+    emit, merge_add_q) vs the plain world=1 MoE path at dim=256."""
+    from dllama_amd.models.hip_model import HipTransformer
+    p = str(tmp_path / "m256.m")
+    make_tiny_qwen3(p, moe=True, dim=256)
+    m = mf.ModelFile(p)
+    cfg = ModelConfig.from_header(m.header)
+    plain = HipTransformer.from_file(m, cfg)
+    tp = HipTransformer.from_file(m, cfg, force_sync=True)
+    prompt = [1, 2, 3]
+    plain.forward(torch.tensor(prompt), torch.arange(3))
+    tp.forward(torch.tensor(prompt), torch.arange(3))
+    for step in range(4):
+        t = torch.tensor([9 + step])
+        pp = torch.tensor([3 + step])
+        want = plain.forward(t, pp).cpu().clone()
+        got = tp.forward(t, pp).cpu().clone()
+        assert _rel_err(got[0], want[0]) < 0.02, (step, _rel_err(got[0], want[0]))
+        assert torch.equal(got.argmax(-1), want.argmax(-1)), f"step {step}"
