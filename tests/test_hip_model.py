@@ -357,3 +357,20 @@ def test_tp_deferred_moe_world1(tmp_path):
         got = tp.forward(t, pp).cpu().clone()
         assert _rel_err(got[0], want[0]) < 0.02, (step, _rel_err(got[0], want[0]))
         assert torch.equal(got.argmax(-1), want.argmax(-1)), f"step {step}"
+
+
+def test_tp_path_f32_sync_world1(tiny_path):
+    """force_sync with the f32 sync buffer (--buffer-float-type f32):
+    all-reduce + add_ssq TP branch on GPU, vs the plain path."""
+    from dllama_amd.models.hip_model import HipTransformer
+    from dllama_amd.quants import F32
+    m = mf.ModelFile(tiny_path, sync_type=F32)
+    cfg = ModelConfig.from_header(m.header)
+    cfg.sync_type = F32
+    plain = HipTransformer.from_file(m, cfg)
+    tp = HipTransformer.from_file(m, cfg, force_sync=True)
+    prompt = [3, 17, 101]
+    want = plain.forward(torch.tensor(prompt), torch.arange(3)).cpu().clone()
+    got = tp.forward(torch.tensor(prompt), torch.arange(3)).cpu().clone()
+    assert _rel_err(got, want) < 1e-4, _rel_err(got, want)
+    assert torch.equal(got.argmax(-1), want.argmax(-1))
