@@ -262,7 +262,8 @@ class HipTransformer:
         if ("DLLAMA_ATTN_SPLITS" in _os.environ
                 and "DLLAMA_ADAPTIVE_SPLITS" not in _os.environ):
             self.adaptive_thresh = 0
-        # S=8 measured best at decode (16/32: combine reads S partials)
+        # split scratch sized for the current S (the adaptive schedule
+        # reallocates on recapture: 8 short ctx, 16 past 512, 32 past 2k)
         self.attn_ml = torch.zeros(NB * c.n_heads0 * self.attn_splits * 2, device=dev)
         self.attn_o = torch.zeros(NB * c.n_heads0 * self.attn_splits * c.head_dim,
                                   device=dev)
