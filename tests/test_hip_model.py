@@ -257,14 +257,18 @@ def test_deferred_quant_moe(tmp_path):
         assert torch.equal(got.argmax(-1), want.argmax(-1)), f"step {step}"
 
 
-def test_rccl_collectives_capture_in_graph(tiny_path):
+def test_rccl_collectives_capture_in_graph(tmp_path):
     """Real RCCL (nccl backend) collectives inside a captured hipGraph on
     hardware: world=1 process group, force_sync TP path (all_gather of the
     Q80 wire + logits gather are genuine RCCL calls even at world 1).
-    De-risks the driver's 8-GPU graph-captured decode before it ever runs."""
+    dim=256 activates the DEFERRED TP path (EPI_PACK wire emit +
+    merge_add_q), the exact kernels the driver's 8-GPU decode replays.
+    De-risks the 8-GPU graph-captured decode before it ever runs."""
     import os
     import subprocess
     import sys
+    mpath = str(tmp_path / "l256r.m")
+    make_tiny_llama(mpath, vocab_size=256, dim=256)
     code = f'''
 import os, torch
 os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
@@ -278,7 +282,7 @@ from dllama_amd import model_file as mf
 from dllama_amd.models.config import ModelConfig
 from dllama_amd.models.hip_model import HipTransformer
 from dllama_amd.parallel.comm import DistComm
-m = mf.ModelFile({tiny_path!r})
+m = mf.ModelFile({mpath!r})
 cfg = ModelConfig.from_header(m.header)
 plain = HipTransformer.from_file(m, cfg)
 tp = HipTransformer.from_file(m, cfg, comm=DistComm(), force_sync=True)
