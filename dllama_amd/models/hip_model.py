@@ -283,10 +283,6 @@ class HipTransformer:
                 self.sync_in = {n: torch.zeros(c.world, n * row_bytes,
                                                dtype=torch.uint8, device=dev)
                                 for n in nbs}
-        # single-slot expert index for routing small-n dense GEMVs through
-        # the lane-tiled grouped kernel (nbp < 64 leaves wave lanes idle in
-        # the plain GEMV; the LPP layout keeps them busy)
-        self.idx0 = torch.zeros(1, dtype=torch.int32, device=dev)
         if c.is_moe:
             S = NB * c.n_active_experts
             self.moe_idx = torch.zeros(S, dtype=torch.int32, device=dev)
@@ -616,19 +612,9 @@ class HipTransformer:
                                 self.v_cache[l], c.q_dim0, c.kv_dim0,
                                 c.head_dim, ssq_in=sin, eps=eps)
             else:
-                if c.dim < 4096:
-                    # nbp < 64: the plain GEMV idles (64 - nbp) lanes per
-                    # wave; the lane-tiled grouped kernel with one slot is
-                    # the same math at full occupancy
-                    k.q40_gemv_grouped(lw["qkv"].qs.unsqueeze(0),
-                                       lw["qkv"].scales.unsqueeze(0),
-                                       self.xq.q, self.xq.s, self.xq.bs,
-                                       self.idx0, self.qkv_out, 1,
-                                       ssq_in=sin, eps=eps)
-                else:
-                    k.q40_gemv(lw["qkv"].qs, lw["qkv"].scales, self.xq.q,
-                               self.xq.s, self.xq.bs, self.qkv_out, 1,
-                               ssq_in=sin, eps=eps)
+                k.q40_gemv(lw["qkv"].qs, lw["qkv"].scales, self.xq.q,
+                           self.xq.s, self.xq.bs, self.qkv_out, 1,
+                           ssq_in=sin, eps=eps)
                 if c.is_qwen3 and self.rope_style == 1:
                     k.rope_kv_qknorm(self.qkv_out, self.qkv_ld, c.q_dim0,
                                      c.kv_dim0, self.rope_cache, self.pos,
