@@ -83,14 +83,15 @@ class HipTransformer:
         self.device = torch.device(device or "cuda")
         self.k = hip_ops()
         # buffers are indexed with NB=_pow2_batch(B) rows and the prefill
-        # GEMM pads to 32 rows: a non-pow2 or <32 n_batches would launch
-        # grids over short buffers (OOB). Round up instead of trusting it.
-        nb = max(32, _pow2_batch(n_batches))
-        if nb != n_batches:
+        # GEMM's MFMA fragment covers exactly 32 batch rows: a non-pow2 or
+        # <32 n_batches would launch grids over short buffers (OOB), and
+        # >32 rows would silently never be written by the GEMM. Pin to 32.
+        nb = 32
+        if n_batches != 32:
             import warnings
             warnings.warn(
-                f"HIP backend rounds --n-batches {n_batches} up to {nb} "
-                "(power of two >= 32 required by buffer/GEMM layout)")
+                f"HIP backend pins --n-batches to 32 (got {n_batches}): the "
+                "prefill GEMM's MFMA fragment is 32 batch rows")
         self.n_batches = nb
         self.layers: list[dict] = []
         self.embedding = None

@@ -2661,6 +2661,8 @@ void q40_gemm(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
     const int n = qs.size(1) * 2;
     TORCH_CHECK(xq.size(-1) == n, "x width mismatch");
     TORCH_CHECK(xq.size(0) >= 32, "gemm needs 32 padded batch rows");
+    TORCH_CHECK(batch <= 32, "the MFMA fragment covers 32 batch rows; "
+                "chunk larger prompts (rows past 32 would be dropped)");
     TORCH_CHECK(n % QB == 0, "n must be a multiple of 32");
     const int mtiles = ceil_div(d, 128);
     int ksplit = 1;
