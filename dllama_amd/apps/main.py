@@ -252,8 +252,9 @@ def run_perplexity(args) -> int:
         return 1
     nll, count = 0.0, 0
     pos = 0
-    for i in range(0, len(tokens) - 1, args.n_batches):
-        chunk = tokens[i: i + args.n_batches]
+    nb = engine.n_batches  # the backend may pin/round the CLI value
+    for i in range(0, len(tokens) - 1, nb):
+        chunk = tokens[i: i + nb]
         t = torch.tensor(chunk, dtype=torch.int64)
         p = torch.arange(pos, pos + len(chunk), dtype=torch.int64)
         logits = engine.model.forward(t, p)  # [B, vocab]
