@@ -260,6 +260,9 @@ class HipTransformer:
         # 5651 tok/s prefill) and costs a 2x-weight shadow — off by default,
         # kept as the library baseline for GEMM tuning A/Bs.
         self.prefill_bf16 = _os.environ.get("DLLAMA_PREFILL_BF16", "0") == "1"
+        # K-split the deferred down-projections (wo/w2 underfill at 1 wg per
+        # 32 rows when dim <= 4096); merged by add_ssq_q. A/B experiment.
+        self.ksplit_resid = int(_os.environ.get("DLLAMA_KSPLIT_RESID", "0"))
         if ("DLLAMA_ATTN_SPLITS" in _os.environ
                 and "DLLAMA_ADAPTIVE_SPLITS" not in _os.environ):
             self.adaptive_thresh = 0
@@ -645,6 +648,13 @@ class HipTransformer:
             if self.tp_path:
                 self._tp_proj_deferred(lw["wo"], self.zq, self.ssq[slot + 1],
                                        lw["norm1"])
+            elif self.ksplit_resid:
+                k.q40_gemv_ksplit(lw["wo"].qs, lw["wo"].scales, self.zq.q,
+                                  self.zq.s, self.zq.bs, self.gemm_part,
+                                  self.ksplit_resid)
+                k.add_ssq_q(x[:1], self.gemm_part, self.ssq[slot + 1],
+                            lw["norm1"], self.xq.q[:1], self.xq.s[:1],
+                            self.xq.bs[:1], self.ksplit_resid)
             else:
                 k.q40_gemv_resid_q(lw["wo"].qs, lw["wo"].scales, self.zq.q,
                                    self.zq.s, self.zq.bs, x,
@@ -701,6 +711,13 @@ class HipTransformer:
                 if self.tp_path:
                     self._tp_proj_deferred(lw["w2"], self.dq,
                                            self.ssq[slot + 1], wn)
+                elif self.ksplit_resid:
+                    k.q40_gemv_ksplit(lw["w2"].qs, lw["w2"].scales, self.dq.q,
+                                      self.dq.s, self.dq.bs, self.gemm_part,
+                                      self.ksplit_resid)
+                    k.add_ssq_q(x[:1], self.gemm_part, self.ssq[slot + 1],
+                                wn, self.xq.q[:1], self.xq.s[:1],
+                                self.xq.bs[:1], self.ksplit_resid)
                 else:
                     k.q40_gemv_resid_q(lw["w2"].qs, lw["w2"].scales,
                                        self.dq.q, self.dq.s, self.dq.bs, x,

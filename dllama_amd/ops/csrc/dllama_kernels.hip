@@ -630,6 +630,12 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
     const int lane = threadIdx.x % WAVE;
     const int nb = n / QB;
     const int nbp = nb >> 1;
+    // K-split (EPI_NONE, B=1 only): gridDim.y slices of the block-pair
+    // range; the small-d down-projections underfill the chip at 1 wg per
+    // 8 rows, so slicing K multiplies the resident workgroups
+    const int nks = gridDim.y;
+    const int j0p = (int)((int64_t)nbp * blockIdx.y / nks);
+    const int j1p = (int)((int64_t)nbp * (blockIdx.y + 1) / nks);
     const bool wave_valid = row0 < d;
     const int rbase = wave_valid ? row0 : 0;
     [[maybe_unused]] __shared__ float svq[32];  // EPI_RESID_Q block staging
@@ -659,7 +665,7 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
             invb[b] = rsqrtf(ssq_total_wave(ssq_in, b, lane) / n + eps);
     }
 
-    for (int jp = lane; jp < nbp; jp += WAVE) {
+    for (int jp = j0p + lane; jp < j1p; jp += WAVE) {
         const int j = jp << 1;
         uint4 wq0[RPW], wq1[RPW];
         float2 sw[RPW];
@@ -694,7 +700,7 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
             }
         }
     }
-    if ((nb & 1) && lane == 0) {  // odd trailing block
+    if ((nb & 1) && lane == 0 && blockIdx.y == nks - 1) {  // odd trailing block
         const int j = nb - 1;
         #pragma unroll
         for (int r = 0; r < RPW; r++) {
@@ -793,7 +799,7 @@ __global__ void k_q40_gemv(const uint8_t *__restrict__ qs,
                 } else if (EPI == EPI_PACK) {
                     svq[wid * RPW + r] = v[r];  // partial, packed below
                 } else {
-                    y[(int64_t)b * d + row] = v[r];
+                    y[((int64_t)b + (int64_t)blockIdx.y * NB) * d + row] = v[r];
                     if (NB == 1) wave_best = max(wave_best, argmax_pack(v[r], row));
                 }
             }
@@ -2354,6 +2360,45 @@ __global__ void k_merge_add_q(float *__restrict__ x,
     }
 }
 
+// K-split completion for the deferred down-projections: x += sum of the
+// K-slices' partials, ssq, and the DEFERRED Q80 emit of x*wnorm — the
+// same epilogue EPI_RESID_Q provides, but fed by a fully-filled K-split
+// GEMV instead of a 1-wg-per-32-rows underfilled one. 256-elem wgs.
+__global__ void k_add_ssq_q(float *__restrict__ x,
+                            const float *__restrict__ parts,
+                            float *__restrict__ ssq,
+                            const float *__restrict__ wnorm,
+                            int8_t *__restrict__ oq,
+                            float *__restrict__ os,
+                            float *__restrict__ obs,
+                            int n, int nks) {
+    const int i = blockIdx.x * 256 + threadIdx.x;
+    float acc = x[i];
+    for (int k = 0; k < nks; k++)
+        acc += parts[(int64_t)k * n + i];
+    x[i] = acc;
+    float local = acc * acc;
+    local = wave_reduce_sum(local);
+    __shared__ float red[4];
+    const int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    if (lane == 0) red[wid] = local;
+    __syncthreads();
+    if (threadIdx.x == 0)
+        atomicAdd(ssq + (blockIdx.x & (SSQ_SPREAD - 1)) * SSQ_PAD,
+                  red[0] + red[1] + red[2] + red[3]);
+    const float v = acc * wnorm[i];
+    const float amax = group32_reduce_max(fabsf(v));
+    const float dd = amax / 127.0f;
+    const float qinv = dd > 0.0f ? 1.0f / dd : 0.0f;
+    const float qf = rintf(v * qinv);
+    oq[i] = (int8_t)qf;
+    const float bsum = group32_reduce_sum(qf);
+    if ((threadIdx.x & 31) == 0) {
+        os[i / QB] = dd;
+        obs[i / QB] = bsum;
+    }
+}
+
 // x += y (residual merge for the f32/TP=1 path)
 __global__ void k_add(float *__restrict__ x, const float *__restrict__ y, int64_t n) {
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
@@ -2549,6 +2594,38 @@ void q40_gemv_pack(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
     e.oqq = reinterpret_cast<int8_t *>(wire.data_ptr<uint8_t>());
     e.force_rpw2 = true;
     gemv_launch<EPI_PACK, 0>(qs, scales, xq, xs, xbs, nullptr, 1, e);
+}
+
+void q40_gemv_ksplit(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
+                     torch::Tensor xs, torch::Tensor xbs, torch::Tensor part,
+                     int64_t ks) {
+    // B=1 K-split GEMV into part[ks, d] (see k_add_ssq_q for the merge)
+    CHECK_CUDA(qs); CHECK_CONT(qs); CHECK_CONT(xq);
+    const int d = qs.size(0);
+    const int n = qs.size(1) * 2;
+    TORCH_CHECK((int64_t)ks * d <= part.numel(), "part buffer too small");
+    const dim3 grid(ceil_div(d, 8), ks);
+    hipLaunchKernelGGL((k_q40_gemv<1, 2, EPI_NONE, 0>), grid, dim3(256), 0,
+                       cur_stream(), qs.data_ptr<uint8_t>(),
+                       reinterpret_cast<const __half *>(scales.data_ptr<at::Half>()),
+                       xq.data_ptr<int8_t>(), xs.data_ptr<float>(),
+                       xbs.data_ptr<float>(), part.data_ptr<float>(), d, n,
+                       nullptr, nullptr, nullptr, nullptr, nullptr, nullptr,
+                       nullptr, 0, 0, 0, nullptr, nullptr, nullptr, 0.0f, 0,
+                       nullptr, nullptr, nullptr);
+}
+
+void add_ssq_q(torch::Tensor x, torch::Tensor parts, torch::Tensor ssq,
+               torch::Tensor wnorm, torch::Tensor oq, torch::Tensor os,
+               torch::Tensor obs, int64_t nks) {
+    CHECK_CUDA(x);
+    const int n = x.size(-1);
+    TORCH_CHECK(n % 256 == 0, "add_ssq_q needs dim % 256 == 0");
+    hipLaunchKernelGGL(k_add_ssq_q, dim3(n / 256), dim3(256), 0, cur_stream(),
+                       x.data_ptr<float>(), parts.data_ptr<float>(),
+                       ssq.data_ptr<float>(), wnorm.data_ptr<float>(),
+                       oq.data_ptr<int8_t>(), os.data_ptr<float>(),
+                       obs.data_ptr<float>(), n, (int)nks);
 }
 
 void q40_gemv_resid_q(torch::Tensor qs, torch::Tensor scales, torch::Tensor xq,
@@ -3492,6 +3569,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("q40_gemv_resid", &q40_gemv_resid);
     m.def("q40_gemv_resid_q", &q40_gemv_resid_q);
     m.def("q40_gemv_pack", &q40_gemv_pack);
+    m.def("q40_gemv_ksplit", &q40_gemv_ksplit);
+    m.def("add_ssq_q", &add_ssq_q);
     m.def("merge_add_q", &merge_add_q);
     m.def("scale_merge_pack", &scale_merge_pack);
     m.def("q40_matmul_cpu", &q40_matmul_cpu);
