@@ -211,4 +211,5 @@ def test_tokenizer_writer_byte_exact_vs_reference_converter(tmp_path):
     ours = str(tmp_path / "t.t")
     tok.write_tokenizer(ours, vocab, scores, 3, True, [3],
                         template.decode())
-    assert open(ours, "rb").read() == buf.getvalue()
+    with open(ours, "rb") as f:
+        assert f.read() == buf.getvalue()
