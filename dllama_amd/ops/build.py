@@ -25,8 +25,14 @@ def load_extension(verbose: bool = False):
     from torch.utils.cpp_extension import load
     src, build_dir = _paths()
     os.makedirs(build_dir, exist_ok=True)
+    # -mavx2/-mfma vectorize the HOST-side CPU Q40 matmul (hipcc applies
+    # x86 flags to the host pass only); AVX2 is safe on every EPYC this
+    # project can land on
     return load(name=_EXT_NAME, sources=[src], build_directory=build_dir,
-                extra_cuda_cflags=["-O3", "-std=c++17"],
+                extra_cuda_cflags=["-O3", "-std=c++17", "-mavx2", "-mfma",
+                                   "-mf16c", "-fopenmp"],
+                extra_ldflags=["-L/opt/rocm/lib/llvm/lib", "-lomp",
+                               "-Wl,-rpath,/opt/rocm/lib/llvm/lib"],
                 verbose=verbose, with_cuda=True)
 
 

@@ -14,6 +14,14 @@
 // All f32 accumulation (parity with the reference numerics).
 
 #include <hip/hip_runtime.h>
+#if defined(__AVX2__)
+#include <immintrin.h>
+#endif
+#include <vector>
+#include <thread>
+#if defined(_OPENMP)
+#include <omp.h>
+#endif
 #include <hip/hip_fp16.h>
 #include <torch/extension.h>
 #include <cstdlib>
@@ -3353,27 +3361,105 @@ void q40_matmul_cpu(torch::Tensor qs, torch::Tensor scales, torch::Tensor x,
     const at::Half *S = scales.data_ptr<at::Half>();
     const float *X = x.data_ptr<float>();
     float *Y = y.data_ptr<float>();
-    at::parallel_for(0, d, 16, [&](int64_t r0, int64_t r1) {
+    // Q80-quantize the activations once (the same per-32-block wire the GPU
+    // path and the reference use, nn-quants.cpp:67), then int8 block dots —
+    // 8-bit integer MACs are what x86 actually vectorizes for this format
+    // (role of the reference's AVX2/AVX512 matmul, nn-cpu-ops.cpp:231-449).
+    std::vector<int8_t> xq((size_t)B * n);
+    std::vector<float> xsc((size_t)B * nb), xbs((size_t)B * nb);
+    for (int b = 0; b < B; b++) {
+        const float *xr = X + (int64_t)b * n;
+        for (int blk = 0; blk < nb; blk++) {
+            float amax = 0.0f;
+            for (int j = 0; j < QB; j++)
+                amax = std::max(amax, std::fabs(xr[blk * QB + j]));
+            const float dd = amax / 127.0f;
+            const float qinv = dd > 0.0f ? 1.0f / dd : 0.0f;
+            float bsum = 0.0f;
+            for (int j = 0; j < QB; j++) {
+                const float qf = std::rint(xr[blk * QB + j] * qinv);
+                xq[(size_t)b * n + blk * QB + j] = (int8_t)qf;
+                bsum += qf;
+            }
+            xsc[(size_t)b * nb + blk] = dd;
+            xbs[(size_t)b * nb + blk] = bsum;
+        }
+    }
+    // OpenMP over rows: at::parallel_for silently serialized here (measured
+    // 1-thread == 8-thread) and per-call std::thread spawn costs ~0.3 ms;
+    // omp's persistent pool respects torch's thread count via num_threads
+    const int nt = std::max(1, (int)at::get_num_threads());
+    auto worker = [&](int64_t r0, int64_t r1) {
         for (int64_t row = r0; row < r1; row++) {
             const uint8_t *w = W + row * (n >> 1);
             const at::Half *sc = S + row * nb;
             for (int b = 0; b < B; b++) {
-                const float *xr = X + (int64_t)b * n;
+                const int8_t *xr = xq.data() + (size_t)b * n;
+                const float *sx = xsc.data() + (size_t)b * nb;
+                const float *bs = xbs.data() + (size_t)b * nb;
                 float acc = 0.0f;
+#if defined(__AVX2__)
+                const __m128i m0f = _mm_set1_epi8(0x0F);
+                const __m256i ones = _mm256_set1_epi16(1);
+                for (int blk = 0; blk < nb; blk++) {
+                    const __m128i raw = _mm_loadu_si128(
+                        reinterpret_cast<const __m128i *>(w + blk * 16));
+                    const __m128i lo = _mm_and_si128(raw, m0f);
+                    const __m128i hi = _mm_and_si128(_mm_srli_epi16(raw, 4), m0f);
+                    const __m256i w8 = _mm256_set_m128i(hi, lo);  // u8 in 0..15
+                    const __m256i x8 = _mm256_loadu_si256(
+                        reinterpret_cast<const __m256i *>(xr + blk * QB));
+                    // u8*i8 pair-sums (|pair| <= 2*15*127 < 2^15: no saturation)
+                    const __m256i p16 = _mm256_maddubs_epi16(w8, x8);
+                    const __m256i p32 = _mm256_madd_epi16(p16, ones);
+                    __m128i h = _mm_add_epi32(_mm256_castsi256_si128(p32),
+                                              _mm256_extracti128_si256(p32, 1));
+                    h = _mm_add_epi32(h, _mm_shuffle_epi32(h, 0x4E));
+                    h = _mm_add_epi32(h, _mm_shuffle_epi32(h, 0xB1));
+                    const int idot = _mm_cvtsi128_si32(h);
+                    acc += (float)sc[blk] * sx[blk]
+                           * ((float)idot - 8.0f * bs[blk]);
+                }
+#else
                 for (int blk = 0; blk < nb; blk++) {
                     const uint8_t *wb = w + blk * 16;
-                    const float *xb = xr + blk * QB;
-                    float a = 0.0f;
+                    const int8_t *xb = xr + blk * QB;
+                    int idot = 0;
                     for (int k = 0; k < 16; k++) {
-                        a += (float)((int)(wb[k] & 15) - 8) * xb[k]
-                           + (float)((int)(wb[k] >> 4) - 8) * xb[k + 16];
+                        idot += (int)(wb[k] & 15) * xb[k]
+                              + (int)(wb[k] >> 4) * xb[k + 16];
                     }
-                    acc += (float)sc[blk] * a;
+                    acc += (float)sc[blk] * sx[blk]
+                           * ((float)idot - 8.0f * bs[blk]);
                 }
+#endif
                 Y[(int64_t)b * d + row] = acc;
             }
         }
-    });
+    };
+    if (nt <= 1 || d < 256) {
+        worker(0, d);
+    } else {
+#if defined(_OPENMP)
+        #pragma omp parallel num_threads(nt)
+        {
+            const int t = omp_get_thread_num();
+            const int tn = omp_get_num_threads();
+            const int64_t chunk = (d + tn - 1) / tn;
+            const int64_t r0 = (int64_t)t * chunk;
+            if (r0 < d) worker(r0, std::min<int64_t>(d, r0 + chunk));
+        }
+#else
+        std::vector<std::thread> threads;
+        const int64_t chunk = (d + nt - 1) / nt;
+        for (int t = 0; t < nt; t++) {
+            const int64_t r0 = t * chunk;
+            if (r0 >= d) break;
+            threads.emplace_back(worker, r0, std::min<int64_t>(d, r0 + chunk));
+        }
+        for (auto &th : threads) th.join();
+#endif
+    }
 }
 
 void sync_pack(torch::Tensor q, torch::Tensor s, torch::Tensor buf) {
