@@ -3387,8 +3387,16 @@ void q40_matmul_cpu(torch::Tensor qs, torch::Tensor scales, torch::Tensor x,
     }
     // OpenMP over rows: at::parallel_for silently serialized here (measured
     // 1-thread == 8-thread) and per-call std::thread spawn costs ~0.3 ms;
-    // omp's persistent pool respects torch's thread count via num_threads
-    const int nt = std::max(1, (int)at::get_num_threads());
+    // omp's persistent pool respects torch's thread count via num_threads.
+    // blocktime MUST be short: libomp's default 200 ms post-region spin
+    // starves torch's own intra-op pool between our regions (measured
+    // 61 s/token on a 256-core box); 1 ms still covers back-to-back
+    // matmuls. Capped at 64 threads — the op is memory-bound.
+#if defined(_OPENMP)
+    static const bool _bt0 = [] { kmp_set_blocktime(1); return true; }();
+    (void)_bt0;
+#endif
+    const int nt = std::min(64, std::max(1, (int)at::get_num_threads()));
     auto worker = [&](int64_t r0, int64_t r1) {
         for (int64_t row = r0; row < r1; row++) {
             const uint8_t *w = W + row * (n >> 1);
