@@ -418,3 +418,40 @@ def test_api_completion_on_gpu(assets):
              "max_tokens": 4, "temperature": 0}
     text2, _, n_gen2 = state.complete(body2, lambda d: None)
     assert n_gen2 >= 1
+
+
+def test_engine_stops_at_context_end(assets):
+    """Generation must stop when the context window is exhausted instead of
+    raising from the model's seq-len guard (reference clamps via seqLen)."""
+    from dllama_amd.apps.main import build_parser, load_engine
+    mp, tp = assets
+    args = build_parser().parse_args(
+        ["inference", "--model", mp, "--tokenizer", tp, "--gpu-index", "-1",
+         "--temperature", "0"])
+    engine, m, comm = load_engine(args)
+    seq = m.header.seq_len
+    prompt = list(range(3, 3 + 8))
+    out, stats = engine.generate(prompt, max_tokens=seq * 2)
+    # engine never advances past seq_len and returns what it produced
+    assert engine.pos <= seq
+    assert 1 <= len(out) <= seq
+    assert stats.decode_tokens == len(out)
+
+
+def test_api_sampler_defaults_reset_between_requests(assets):
+    """A request that overrides temperature must not leak it into the next
+    request that omits it (advisor finding; reference re-parses defaults,
+    dllama-api.cpp:491-520)."""
+    from dllama_amd.apps.api import ApiState
+    from dllama_amd.apps.main import build_parser
+    mp, tp = assets
+    args = build_parser().parse_args(
+        ["inference", "--model", mp, "--tokenizer", tp, "--gpu-index", "-1",
+         "--temperature", "0", "--seed", "5"])
+    state = ApiState(args)
+    base = {"messages": [{"role": "user", "content": "ab"}], "max_tokens": 6}
+    t_default, _, _ = state.complete(dict(base), lambda d: None)
+    state.complete({**base, "temperature": 1.7, "seed": 99}, lambda d: None)
+    t_again, _, _ = state.complete(dict(base), lambda d: None)
+    assert state.engine.sampler.temperature == 0.0
+    assert t_again == t_default  # greedy determinism restored
