@@ -21,7 +21,9 @@ import torch
 
 from ..model_file import HIDDEN_ACT_GELU, ModelFile, ROPE_FALCON
 from ..ops import reference as R
-from ..quants import Q40
+from ..parallel.comm import Comm, SingleComm
+from ..quants import Q40, Q80
+from .config import ModelConfig
 
 
 class Q40W:
@@ -35,9 +37,6 @@ class Q40W:
 
     def __getitem__(self, e: int) -> "Q40W":  # stacked MoE experts
         return Q40W(self.qs[e], self.scales[e])
-from ..parallel.comm import Comm, SingleComm
-from ..quants import Q80
-from .config import ModelConfig
 
 
 class CpuTransformer:
