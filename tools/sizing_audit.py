@@ -38,7 +38,7 @@ def audit(name: str, world: int, seq_len: int):
                + c.vocab_size * c.dim * 4           # f32 embedding table
                + q40_plane_bytes(c.vocab0, c.dim)   # logits shard
                + c.dim * 4)
-    kv = c.n_layers * seq_len * c.kv_dim0 * 4 * 2
+    kv = c.n_layers * seq_len * c.kv_dim0 * 2 * 2  # f16 KV (round-2 default)
     act = NB * (3 * c.dim + c.q_dim0 + 2 * c.kv_dim0 + 2 * c.q_dim0
                 + 4 * c.ff_dim0 + c.vocab0) * 4
     act += NB * (c.dim + c.q_dim0 + c.ff_dim0) * 2  # int8+scale quant bufs
@@ -48,7 +48,7 @@ def audit(name: str, world: int, seq_len: int):
     total = weights + kv + act + sync
     print(f"{name:16s} TP={world}  seq={seq_len}")
     print(f"  weights/rank: {weights/1e9:8.2f} GB")
-    print(f"  kv cache    : {kv/1e9:8.2f} GB (f32, {seq_len} x {c.kv_dim0} x {c.n_layers}L x 2)")
+    print(f"  kv cache    : {kv/1e9:8.2f} GB (f16, {seq_len} x {c.kv_dim0} x {c.n_layers}L x 2)")
     print(f"  activations : {act/1e9:8.2f} GB   sync: {sync/1e6:.0f} MB")
     fits = "✅ fits" if total < HBM * 0.97 else "❌ DOES NOT FIT"
     print(f"  total/rank  : {total/1e9:8.2f} GB of 288 GB  -> {fits}\n")
