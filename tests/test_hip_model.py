@@ -378,3 +378,35 @@ def test_tp_path_f32_sync_world1(tiny_path):
     got = tp.forward(torch.tensor(prompt), torch.arange(3)).cpu().clone()
     assert _rel_err(got, want) < 1e-4, _rel_err(got, want)
     assert torch.equal(got.argmax(-1), want.argmax(-1))
+
+
+def test_experimental_paths_match_defaults(tmp_path, monkeypatch):
+    """Env-gated experiment paths (read per model construction) must stay
+    correct while not default: bf16 prefill shadow, K-split
+    down-projections, fused FFN. (DLLAMA_GQA_ATTN and the kernel-level v1
+    toggles are cached in C++ statics at first use, so they are validated
+    by their dedicated kernel tests instead.) dim=256 so every variant's
+    shape constraints are met."""
+    from dllama_amd.models import hip_model as hm
+    p = str(tmp_path / "l256e.m")
+    make_tiny_llama(p, vocab_size=256, dim=256)
+    m = mf.ModelFile(p)
+    cfg = ModelConfig.from_header(m.header)
+    ref = hm.HipTransformer.from_file(m, cfg)
+    tokens = torch.tensor(list(range(3, 15)))  # 12 -> padded 32: prefill path
+    want = ref.forward(tokens, torch.arange(12)).cpu().clone()
+    dt = torch.tensor([5])
+    dp = torch.tensor([12])
+    want_d = ref.forward(dt, dp).cpu().clone()
+
+    for env in ("DLLAMA_PREFILL_BF16", "DLLAMA_KSPLIT_RESID",
+                "DLLAMA_FUSED_FFN"):
+        monkeypatch.setenv(env, "2" if env == "DLLAMA_KSPLIT_RESID" else "1")
+        alt = hm.HipTransformer.from_file(m, cfg)
+        got = alt.forward(tokens, torch.arange(12)).cpu().clone()
+        tol = 0.03 if env == "DLLAMA_PREFILL_BF16" else 2e-3
+        assert _rel_err(got, want) < tol, (env, _rel_err(got, want))
+        assert torch.equal(got.argmax(-1), want.argmax(-1)), env
+        got_d = alt.forward(dt, dp).cpu().clone()
+        assert _rel_err(got_d[0], want_d[0]) < tol, (env, "decode")
+        monkeypatch.delenv(env)
