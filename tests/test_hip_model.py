@@ -412,3 +412,16 @@ def test_experimental_paths_match_defaults(tmp_path, monkeypatch):
         got_d = alt.forward(dt, dp).cpu().clone()
         assert _rel_err(got_d[0], want_d[0]) < tol, (env, "decode")
         monkeypatch.delenv(env)
+
+    # bf16 prefill through the qwen3 branch (qk-norm inside the chunk path)
+    q3 = str(tmp_path / "q3e.m")
+    make_tiny_qwen3(q3, moe=False, dim=256)
+    mq = mf.ModelFile(q3)
+    cq = ModelConfig.from_header(mq.header)
+    refq = hm.HipTransformer.from_file(mq, cq)
+    wantq = refq.forward(tokens, torch.arange(12)).cpu().clone()
+    monkeypatch.setenv("DLLAMA_PREFILL_BF16", "1")
+    altq = hm.HipTransformer.from_file(mq, cq)
+    gotq = altq.forward(tokens, torch.arange(12)).cpu().clone()
+    assert _rel_err(gotq, wantq) < 0.03, _rel_err(gotq, wantq)
+    monkeypatch.delenv("DLLAMA_PREFILL_BF16")
